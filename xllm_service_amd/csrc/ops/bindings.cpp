@@ -1,0 +1,270 @@
+// PyTorch bindings for the MI355X (gfx950) HIP kernels.
+//
+// All tensor-shape/dtype validation lives here so the kernels stay lean.
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <vector>
+
+namespace xllm {
+void launch_rmsnorm(unsigned short*, const unsigned short*,
+                    const unsigned short*, float, int, int, hipStream_t);
+void launch_fused_add_rmsnorm(unsigned short*, unsigned short*,
+                              const unsigned short*, float, int, int,
+                              hipStream_t);
+void launch_rope(unsigned short*, unsigned short*, const long*, const float*,
+                 int, int, int, int, int, hipStream_t);
+void launch_silu_and_mul(unsigned short*, const unsigned short*, int, int,
+                         hipStream_t);
+void launch_gelu_and_mul(unsigned short*, const unsigned short*, int, int,
+                         hipStream_t);
+void launch_reshape_and_cache(unsigned short*, unsigned short*,
+                              const unsigned short*, const unsigned short*,
+                              const long*, int, int, int, int, hipStream_t);
+void launch_copy_blocks(unsigned short*, unsigned short*, const long*, int,
+                        long, hipStream_t);
+void launch_gather_blocks(unsigned short*, unsigned short*, const long*, int,
+                          long, bool, hipStream_t);
+void launch_paged_attn_decode(unsigned short*, const unsigned short*,
+                              const unsigned short*, const unsigned short*,
+                              const int*, const int*, float, int, int, int,
+                              int, int, hipStream_t);
+void launch_paged_attn_prefill(unsigned short*, const unsigned short*,
+                               const unsigned short*, const unsigned short*,
+                               const int*, const int*, const int*, const int*,
+                               const int*, int, float, int, int, int,
+                               hipStream_t);
+void launch_greedy_sample(long*, const unsigned short*, int, int, hipStream_t);
+void launch_mfma_probe(float*, const unsigned short*, const unsigned short*,
+                       hipStream_t);
+}  // namespace xllm
+
+namespace {
+
+#define CHECK_BF16_CUDA(t)                                      \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU");             \
+  TORCH_CHECK((t).dtype() == torch::kBFloat16, #t " must be bf16"); \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+inline unsigned short* u16(torch::Tensor& t) {
+  return reinterpret_cast<unsigned short*>(t.data_ptr());
+}
+inline const unsigned short* u16c(const torch::Tensor& t) {
+  return reinterpret_cast<const unsigned short*>(t.data_ptr());
+}
+inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps) {
+  CHECK_BF16_CUDA(out); CHECK_BF16_CUDA(x); CHECK_BF16_CUDA(w);
+  const int H = x.size(-1);
+  const int T = x.numel() / H;
+  TORCH_CHECK(H % 8 == 0, "hidden size must be a multiple of 8");
+  xllm::launch_rmsnorm(u16(out), u16c(x), u16c(w), (float)eps, T, H,
+                       cur_stream());
+}
+
+void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
+                       torch::Tensor w, double eps) {
+  CHECK_BF16_CUDA(x); CHECK_BF16_CUDA(residual); CHECK_BF16_CUDA(w);
+  const int H = x.size(-1);
+  const int T = x.numel() / H;
+  TORCH_CHECK(H % 8 == 0, "hidden size must be a multiple of 8");
+  xllm::launch_fused_add_rmsnorm(u16(x), u16(residual), u16c(w), (float)eps, T,
+                                 H, cur_stream());
+}
+
+void rope(torch::Tensor positions, torch::Tensor q, torch::Tensor k,
+          torch::Tensor cos_sin, long head_dim, long rot_dim) {
+  CHECK_BF16_CUDA(q); CHECK_BF16_CUDA(k);
+  TORCH_CHECK(positions.dtype() == torch::kLong && positions.is_cuda());
+  TORCH_CHECK(cos_sin.dtype() == torch::kFloat && cos_sin.is_cuda());
+  const int T = positions.size(0);
+  const int n_q = q.numel() / T / head_dim;
+  const int n_k = k.numel() / T / head_dim;
+  xllm::launch_rope(u16(q), u16(k), positions.data_ptr<long>(),
+                    cos_sin.data_ptr<float>(), T, n_q, n_k, (int)head_dim,
+                    (int)rot_dim, cur_stream());
+}
+
+void silu_and_mul(torch::Tensor out, torch::Tensor x) {
+  CHECK_BF16_CUDA(out); CHECK_BF16_CUDA(x);
+  const int I = out.size(-1);
+  const int T = out.numel() / I;
+  TORCH_CHECK(x.size(-1) == 2 * I && I % 8 == 0);
+  xllm::launch_silu_and_mul(u16(out), u16c(x), T, I, cur_stream());
+}
+
+void gelu_and_mul(torch::Tensor out, torch::Tensor x) {
+  CHECK_BF16_CUDA(out); CHECK_BF16_CUDA(x);
+  const int I = out.size(-1);
+  const int T = out.numel() / I;
+  TORCH_CHECK(x.size(-1) == 2 * I && I % 8 == 0);
+  xllm::launch_gelu_and_mul(u16(out), u16c(x), T, I, cur_stream());
+}
+
+void reshape_and_cache(torch::Tensor k, torch::Tensor v,
+                       torch::Tensor k_cache, torch::Tensor v_cache,
+                       torch::Tensor slot_mapping) {
+  CHECK_BF16_CUDA(k); CHECK_BF16_CUDA(v);
+  CHECK_BF16_CUDA(k_cache); CHECK_BF16_CUDA(v_cache);
+  TORCH_CHECK(slot_mapping.dtype() == torch::kLong && slot_mapping.is_cuda());
+  const int T = k.size(0);
+  const int n_kv = k_cache.size(1);
+  const int bs = k_cache.size(2);
+  const int D = k_cache.size(3);
+  TORCH_CHECK(D % 8 == 0);
+  xllm::launch_reshape_and_cache(u16(k_cache), u16(v_cache), u16c(k), u16c(v),
+                                 slot_mapping.data_ptr<long>(), T, n_kv, D, bs,
+                                 cur_stream());
+}
+
+void copy_blocks(torch::Tensor k_cache, torch::Tensor v_cache,
+                 torch::Tensor pairs) {
+  CHECK_BF16_CUDA(k_cache); CHECK_BF16_CUDA(v_cache);
+  TORCH_CHECK(pairs.dtype() == torch::kLong && pairs.is_cuda() &&
+              pairs.is_contiguous());
+  const long numel = (long)k_cache.size(1) * k_cache.size(2) * k_cache.size(3);
+  xllm::launch_copy_blocks(u16(k_cache), u16(v_cache), pairs.data_ptr<long>(),
+                           pairs.size(0), numel, cur_stream());
+}
+
+void gather_blocks(torch::Tensor staging, torch::Tensor cache,
+                   torch::Tensor block_ids, bool gather) {
+  CHECK_BF16_CUDA(staging); CHECK_BF16_CUDA(cache);
+  TORCH_CHECK(block_ids.dtype() == torch::kLong && block_ids.is_cuda());
+  const long numel = (long)cache.size(1) * cache.size(2) * cache.size(3);
+  xllm::launch_gather_blocks(u16(staging), u16(cache),
+                             block_ids.data_ptr<long>(), block_ids.size(0),
+                             numel, gather, cur_stream());
+}
+
+void paged_attn_decode(torch::Tensor out, torch::Tensor q,
+                       torch::Tensor k_cache, torch::Tensor v_cache,
+                       torch::Tensor block_tables, torch::Tensor seq_lens,
+                       double scale) {
+  CHECK_BF16_CUDA(out); CHECK_BF16_CUDA(q);
+  CHECK_BF16_CUDA(k_cache); CHECK_BF16_CUDA(v_cache);
+  TORCH_CHECK(block_tables.dtype() == torch::kInt && block_tables.is_cuda());
+  TORCH_CHECK(seq_lens.dtype() == torch::kInt && seq_lens.is_cuda());
+  const int num_seqs = q.size(0);
+  const int n_qheads = q.size(1);
+  const int D = q.size(2);
+  const int n_kv = k_cache.size(1);
+  TORCH_CHECK(k_cache.size(2) == 16, "block_size must be 16");
+  TORCH_CHECK(D <= 128 && D % 8 == 0, "head_dim must be <=128, mult of 8");
+  TORCH_CHECK(n_qheads % n_kv == 0 && n_qheads / n_kv <= 8,
+              "GQA group must be <= 8");
+  xllm::launch_paged_attn_decode(
+      u16(out), u16c(q), u16c(k_cache), u16c(v_cache),
+      block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), (float)scale,
+      num_seqs, n_qheads, n_kv, D, block_tables.size(1), cur_stream());
+}
+
+void paged_attn_prefill(torch::Tensor out, torch::Tensor q,
+                        torch::Tensor k_cache, torch::Tensor v_cache,
+                        torch::Tensor block_tables, torch::Tensor cu_q,
+                        torch::Tensor seq_lens, torch::Tensor tile_seq,
+                        torch::Tensor tile_q0, double scale) {
+  CHECK_BF16_CUDA(out); CHECK_BF16_CUDA(q);
+  CHECK_BF16_CUDA(k_cache); CHECK_BF16_CUDA(v_cache);
+  for (auto* t : {&block_tables, &cu_q, &seq_lens, &tile_seq, &tile_q0}) {
+    TORCH_CHECK(t->dtype() == torch::kInt && t->is_cuda() && t->is_contiguous());
+  }
+  const int n_qheads = q.size(1);
+  const int D = q.size(2);
+  const int n_kv = k_cache.size(1);
+  TORCH_CHECK(D == 128, "prefill kernel requires head_dim == 128");
+  TORCH_CHECK(k_cache.size(2) == 16, "block_size must be 16");
+  xllm::launch_paged_attn_prefill(
+      u16(out), u16c(q), u16c(k_cache), u16c(v_cache),
+      block_tables.data_ptr<int>(), cu_q.data_ptr<int>(),
+      seq_lens.data_ptr<int>(), tile_seq.data_ptr<int>(),
+      tile_q0.data_ptr<int>(), tile_seq.size(0), (float)scale, n_qheads, n_kv,
+      block_tables.size(1), cur_stream());
+}
+
+torch::Tensor greedy_sample(torch::Tensor logits) {
+  CHECK_BF16_CUDA(logits);
+  const int B = logits.size(0);
+  const int V = logits.size(1);
+  auto out = torch::empty({B}, logits.options().dtype(torch::kLong));
+  xllm::launch_greedy_sample(out.data_ptr<long>(), u16c(logits), B, V,
+                             cur_stream());
+  return out;
+}
+
+torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b) {
+  CHECK_BF16_CUDA(a); CHECK_BF16_CUDA(b);
+  TORCH_CHECK(a.size(0) == 16 && a.size(1) == 32);
+  TORCH_CHECK(b.size(0) == 32 && b.size(1) == 16);
+  auto c = torch::empty({16, 16}, a.options().dtype(torch::kFloat));
+  xllm::launch_mfma_probe(c.data_ptr<float>(), u16c(a), u16c(b), cur_stream());
+  return c;
+}
+
+// ------ cross-GPU KV-block migration over xGMI ----------------------------------
+// Direct peer copy of whole blocks: both caches must have identical geometry.
+// Contiguous runs of (src,dst) pairs collapse into single hipMemcpyPeerAsync
+// calls on the CURRENT stream (callers use a dedicated side stream to overlap
+// with decode; see engine/kv_migration.py).
+void migrate_blocks_peer(torch::Tensor dst_cache, long dst_device,
+                         torch::Tensor src_cache, long src_device,
+                         std::vector<long> src_blocks,
+                         std::vector<long> dst_blocks) {
+  TORCH_CHECK(src_blocks.size() == dst_blocks.size());
+  const long numel =
+      (long)src_cache.size(1) * src_cache.size(2) * src_cache.size(3);
+  const long bytes = numel * 2;  // bf16
+  char* dst = reinterpret_cast<char*>(dst_cache.data_ptr());
+  const char* src = reinterpret_cast<const char*>(src_cache.data_ptr());
+  hipStream_t stream = cur_stream();
+  size_t i = 0;
+  while (i < src_blocks.size()) {
+    // collapse contiguous runs
+    size_t j = i + 1;
+    while (j < src_blocks.size() && src_blocks[j] == src_blocks[j - 1] + 1 &&
+           dst_blocks[j] == dst_blocks[j - 1] + 1)
+      j++;
+    const long n = (long)(j - i);
+    hipError_t err = hipMemcpyPeerAsync(
+        dst + dst_blocks[i] * bytes, (int)dst_device,
+        src + src_blocks[i] * bytes, (int)src_device, (size_t)(n * bytes),
+        stream);
+    TORCH_CHECK(err == hipSuccess, "hipMemcpyPeerAsync failed: ",
+                hipGetErrorString(err));
+    i = j;
+  }
+}
+
+void enable_peer_access(long device, long peer) {
+  int can = 0;
+  hipError_t err = hipDeviceCanAccessPeer(&can, (int)device, (int)peer);
+  TORCH_CHECK(err == hipSuccess && can, "no P2P path between GPUs ", device,
+              " and ", peer);
+  hipSetDevice((int)device);
+  err = hipDeviceEnablePeerAccess((int)peer, 0);
+  TORCH_CHECK(err == hipSuccess || err == hipErrorPeerAccessAlreadyEnabled,
+              "hipDeviceEnablePeerAccess failed: ", hipGetErrorString(err));
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm);
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
+  m.def("rope", &rope);
+  m.def("silu_and_mul", &silu_and_mul);
+  m.def("gelu_and_mul", &gelu_and_mul);
+  m.def("reshape_and_cache", &reshape_and_cache);
+  m.def("copy_blocks", &copy_blocks);
+  m.def("gather_blocks", &gather_blocks);
+  m.def("paged_attn_decode", &paged_attn_decode);
+  m.def("paged_attn_prefill", &paged_attn_prefill);
+  m.def("greedy_sample", &greedy_sample);
+  m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
+  m.def("migrate_blocks_peer", &migrate_blocks_peer);
+  m.def("enable_peer_access", &enable_peer_access);
+}

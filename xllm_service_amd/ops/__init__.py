@@ -1,0 +1,156 @@
+"""Op dispatch: hand-written CDNA4 HIP kernels on GPU, torch references on CPU.
+
+Policy (enforced, not silent): when running on a GPU, the HIP extension MUST
+be present and is the only path taken — a missing extension raises instead of
+falling back to eager torch, so a GPU run can never silently measure the
+reference implementation.
+"""
+from __future__ import annotations
+
+import torch
+
+from . import ref
+
+try:
+    from xllm_service_amd import _ops  # built in-tree by setup.py
+    HAS_EXT = True
+except ImportError:  # CPU-only environments without a built extension
+    _ops = None
+    HAS_EXT = False
+
+
+def _require_ext():
+    if not HAS_EXT:
+        raise RuntimeError(
+            "xllm_service_amd._ops (gfx950 HIP extension) is not built but a "
+            "GPU op was requested. Build it with: "
+            "PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace"
+        )
+
+
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
+    if x.is_cuda:
+        _require_ext()
+        out = torch.empty_like(x)
+        _ops.rmsnorm(out, x, w, eps)
+        return out
+    return ref.rmsnorm(x, w, eps)
+
+
+def fused_add_rmsnorm(x, residual, w, eps):
+    """In-place on GPU: residual += x; x = rmsnorm(residual)*w. Returns (x, residual)."""
+    if x.is_cuda:
+        _require_ext()
+        _ops.fused_add_rmsnorm(x, residual, w, eps)
+        return x, residual
+    out, new_res = ref.fused_add_rmsnorm(x, residual, w, eps)
+    return out, new_res
+
+
+def rope(positions, q, k, cos_sin, head_dim, rot_dim):
+    """In-place on GPU; out-of-place on CPU. Returns (q, k)."""
+    if q.is_cuda:
+        _require_ext()
+        _ops.rope(positions, q, k, cos_sin, head_dim, rot_dim)
+        return q, k
+    return ref.rope(positions, q, k, cos_sin, head_dim, rot_dim)
+
+
+def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        _require_ext()
+        d = x.shape[-1] // 2
+        out = torch.empty(x.shape[:-1] + (d,), dtype=x.dtype, device=x.device)
+        _ops.silu_and_mul(out, x)
+        return out
+    return ref.silu_and_mul(x)
+
+
+def gelu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        _require_ext()
+        d = x.shape[-1] // 2
+        out = torch.empty(x.shape[:-1] + (d,), dtype=x.dtype, device=x.device)
+        _ops.gelu_and_mul(out, x)
+        return out
+    return ref.gelu_and_mul(x)
+
+
+def reshape_and_cache(k, v, k_cache, v_cache, slot_mapping):
+    if k.is_cuda:
+        _require_ext()
+        _ops.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+        return
+    ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+
+
+def copy_blocks(k_cache, v_cache, pairs):
+    if k_cache.is_cuda:
+        _require_ext()
+        _ops.copy_blocks(k_cache, v_cache, pairs)
+        return
+    for s, d in pairs.tolist():
+        k_cache[d].copy_(k_cache[s])
+        v_cache[d].copy_(v_cache[s])
+
+
+def paged_attn_decode(q, k_cache, v_cache, block_tables, seq_lens, scale):
+    if q.is_cuda:
+        _require_ext()
+        out = torch.empty_like(q)
+        _ops.paged_attn_decode(out, q, k_cache, v_cache, block_tables,
+                               seq_lens, scale)
+        return out
+    return ref.paged_attn_decode(q, k_cache, v_cache, block_tables, seq_lens,
+                                 scale)
+
+
+def _prefill_tiles(cu_q):
+    """Host-side tile decomposition for the prefill kernel (64 q rows/WG)."""
+    tile_seq, tile_q0 = [], []
+    for s in range(len(cu_q) - 1):
+        qlen = int(cu_q[s + 1]) - int(cu_q[s])
+        for q0 in range(0, qlen, 64):
+            tile_seq.append(s)
+            tile_q0.append(q0)
+    return tile_seq, tile_q0
+
+
+def paged_attn_prefill(q, k_cache, v_cache, block_tables, cu_q, seq_lens,
+                       scale):
+    if q.is_cuda:
+        _require_ext()
+        out = torch.empty_like(q)
+        tile_seq, tile_q0 = _prefill_tiles(cu_q.cpu())
+        dev = q.device
+        _ops.paged_attn_prefill(
+            out, q, k_cache, v_cache, block_tables, cu_q, seq_lens,
+            torch.tensor(tile_seq, dtype=torch.int32, device=dev),
+            torch.tensor(tile_q0, dtype=torch.int32, device=dev), scale)
+        return out
+    return ref.paged_attn_prefill(q, k_cache, v_cache, block_tables, cu_q,
+                                  seq_lens, scale)
+
+
+def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
+    if logits.is_cuda:
+        _require_ext()
+        return _ops.greedy_sample(logits)
+    return ref.greedy_sample(logits)
+
+
+def migrate_blocks_peer(dst_cache, dst_device, src_cache, src_device,
+                        src_blocks, dst_blocks):
+    _require_ext()
+    _ops.migrate_blocks_peer(dst_cache, dst_device, src_cache, src_device,
+                             list(src_blocks), list(dst_blocks))
+
+
+def enable_peer_access(device: int, peer: int):
+    _require_ext()
+    _ops.enable_peer_access(device, peer)
+
+
+def mfma_probe_16x16x32(a, b):
+    _require_ext()
+    return _ops.mfma_probe_16x16x32(a, b)
