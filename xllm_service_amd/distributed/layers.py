@@ -1,0 +1,74 @@
+"""Tensor-parallel linear layers (Megatron-style sharding over RCCL/xGMI).
+
+At tp_size()==1 these are plain linears with zero overhead. Sharding follows
+the standard column(QKV, gate/up) -> row(o_proj, down) pattern so each
+transformer block needs exactly ONE all-reduce per sublayer — sized for the
+xGMI point-to-point topology (few, large collectives).
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from . import parallel_state as ps
+
+
+class ColumnParallelLinear(nn.Module):
+    """Y = X W^T with W sharded along the output dim."""
+
+    def __init__(self, in_features: int, out_features: int, bias: bool = False,
+                 dtype=None):
+        super().__init__()
+        tp = ps.tp_size()
+        assert out_features % tp == 0, (out_features, tp)
+        self.in_features = in_features
+        self.out_features = out_features
+        self.shard_out = out_features // tp
+        self.weight = nn.Parameter(
+            torch.empty(self.shard_out, in_features, dtype=dtype))
+        self.bias = nn.Parameter(torch.zeros(self.shard_out, dtype=dtype)) if bias else None
+
+    def forward(self, x):
+        return torch.nn.functional.linear(x, self.weight, self.bias)
+
+
+class MergedColumnParallelLinear(ColumnParallelLinear):
+    """Several column-parallel projections fused into one GEMM (e.g. QKV,
+    gate+up). Each constituent is sharded independently so the shard layout
+    is [sum of per-part shards, in]."""
+
+    def __init__(self, in_features: int, out_sizes: list[int], bias: bool = False,
+                 dtype=None):
+        tp = ps.tp_size()
+        for s in out_sizes:
+            assert s % tp == 0, (s, tp)
+        self.out_sizes = out_sizes
+        super().__init__(in_features, sum(out_sizes), bias=bias, dtype=dtype)
+
+    def shard_split(self, y: torch.Tensor):
+        tp = ps.tp_size()
+        return torch.split(y, [s // tp for s in self.out_sizes], dim=-1)
+
+
+class RowParallelLinear(nn.Module):
+    """Y = X W^T with W sharded along the input dim; output all-reduced."""
+
+    def __init__(self, in_features: int, out_features: int, bias: bool = False,
+                 dtype=None):
+        super().__init__()
+        tp = ps.tp_size()
+        assert in_features % tp == 0, (in_features, tp)
+        self.in_features = in_features
+        self.out_features = out_features
+        self.shard_in = in_features // tp
+        self.weight = nn.Parameter(
+            torch.empty(out_features, self.shard_in, dtype=dtype))
+        # bias added once (rank 0's contribution) to keep the sum correct
+        self.bias = nn.Parameter(torch.zeros(out_features, dtype=dtype)) if bias else None
+
+    def forward(self, x):
+        y = torch.nn.functional.linear(x, self.weight)
+        y = ps.tp_all_reduce(y)
+        if self.bias is not None:
+            y = y + self.bias
+        return y

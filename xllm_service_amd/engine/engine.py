@@ -1,0 +1,159 @@
+"""LLMEngine: the per-worker inference engine (colocated prefill+decode).
+
+One engine per GPU process. The serving master talks to it through
+engine/worker.py (RPC); bench.py and tests drive it directly.
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+
+from xllm_service_amd.models.config import ModelConfig, get_config
+from xllm_service_amd.models.registry import create_model
+
+from .block_manager import BlockManager
+from .model_runner import BLOCK_SIZE, ModelRunner
+from .sampling import SamplingParams
+from .scheduler import EngineScheduler
+from .sequence import Sequence, SeqStatus
+
+
+@dataclass
+class StepOutput:
+    request_id: str
+    new_token_ids: List[int]
+    finished: bool
+    finish_reason: Optional[str] = None   # "stop" | "length" | "abort"
+    num_prompt_tokens: int = 0
+    num_output_tokens: int = 0
+    first_token: bool = False
+
+
+@dataclass
+class EngineStats:
+    num_waiting: int = 0
+    num_running: int = 0
+    kv_usage: float = 0.0
+    steps: int = 0
+    generated_tokens: int = 0
+
+
+class LLMEngine:
+    def __init__(self, model_name: str = "llama-3-8b",
+                 device: Optional[str] = None,
+                 dtype=torch.bfloat16,
+                 max_num_seqs: int = 256,
+                 max_batched_tokens: int = 8192,
+                 gpu_memory_utilization: float = 0.85,
+                 max_kv_blocks: Optional[int] = None,
+                 enable_prefix_caching: bool = True,
+                 seed: int = 0):
+        self.cfg: ModelConfig = get_config(model_name)
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = torch.device(device)
+        if self.device.type == "cpu" and dtype == torch.bfloat16:
+            dtype = torch.float32  # CPU reference path runs fp32
+        self.dtype = dtype
+
+        self.model = create_model(self.cfg, dtype=dtype)
+        self.model.random_init(seed)
+        self.model = self.model.to(self.device).eval()
+
+        n_kv_local = getattr(self.model, "local_kv_heads", self.cfg.num_kv_heads)
+        num_blocks = ModelRunner.kv_cache_blocks_for(
+            self.cfg, self.device, n_kv_local,
+            gpu_memory_utilization=gpu_memory_utilization,
+            dtype_bytes=2 if dtype == torch.bfloat16 else 4,
+            max_blocks=max_kv_blocks)
+        self.block_manager = BlockManager(num_blocks, BLOCK_SIZE,
+                                          enable_prefix_caching)
+        self.scheduler = EngineScheduler(self.block_manager,
+                                         max_num_seqs=max_num_seqs,
+                                         max_batched_tokens=max_batched_tokens)
+        self.runner = ModelRunner(self.model, self.cfg, self.device,
+                                  num_blocks, dtype=dtype)
+        self.seqs: Dict[str, Sequence] = {}
+        self.stats = EngineStats()
+        self.eos_token_id: Optional[int] = None  # set by tokenizer owner
+
+    # ---- request API --------------------------------------------------------
+    def add_request(self, request_id: str, prompt_token_ids: List[int],
+                    params: Optional[SamplingParams] = None,
+                    priority: int = 0,
+                    eos_token_id: Optional[int] = None) -> None:
+        if request_id in self.seqs:
+            raise ValueError(f"duplicate request_id {request_id}")
+        seq = Sequence(request_id=request_id,
+                       prompt_token_ids=list(prompt_token_ids),
+                       params=params or SamplingParams(),
+                       eos_token_id=eos_token_id if eos_token_id is not None
+                       else self.eos_token_id,
+                       priority=priority)
+        self.seqs[request_id] = seq
+        self.scheduler.add(seq)
+
+    def abort_request(self, request_id: str) -> bool:
+        seq = self.scheduler.abort(request_id)
+        self.seqs.pop(request_id, None)
+        return seq is not None
+
+    def has_work(self) -> bool:
+        return self.scheduler.has_work()
+
+    # ---- main loop ----------------------------------------------------------
+    def step(self) -> List[StepOutput]:
+        plan = self.scheduler.schedule()
+        if plan.empty:
+            return []
+        new_tokens = self.runner.execute(plan, self.block_manager)
+
+        outputs: List[StepOutput] = []
+        for rid, tok in new_tokens.items():
+            seq = self.seqs.get(rid)
+            if seq is None:
+                continue
+            first = not seq.output_token_ids
+            seq.append_token(tok)
+            finished = seq.check_finish()
+            outputs.append(StepOutput(
+                request_id=rid,
+                new_token_ids=[tok],
+                finished=finished,
+                finish_reason=(
+                    "stop" if seq.status == SeqStatus.FINISHED_STOP else
+                    "length" if seq.status == SeqStatus.FINISHED_LENGTH else
+                    None),
+                num_prompt_tokens=seq.prompt_len,
+                num_output_tokens=len(seq.output_token_ids),
+                first_token=first,
+            ))
+            self.stats.generated_tokens += 1
+        # advance prefill progress + retire finished sequences
+        self.scheduler.on_step_done(plan)
+        for out in outputs:
+            if out.finished:
+                self.seqs.pop(out.request_id, None)
+        self.stats.steps += 1
+        self.stats.num_waiting = self.scheduler.num_waiting
+        self.stats.num_running = len(self.scheduler.running)
+        self.stats.kv_usage = self.block_manager.usage()
+        return outputs
+
+    # ---- convenience (tests, smoke) ----------------------------------------
+    def generate(self, prompts: List[List[int]],
+                 params: Optional[SamplingParams] = None,
+                 timeout_s: float = 600.0) -> List[List[int]]:
+        for i, p in enumerate(prompts):
+            self.add_request(f"gen-{i}", p, params)
+        results: Dict[str, List[int]] = {}
+        t0 = time.monotonic()
+        while self.has_work():
+            if time.monotonic() - t0 > timeout_s:
+                raise TimeoutError("generate() exceeded timeout")
+            for out in self.step():
+                results.setdefault(out.request_id, []).extend(out.new_token_ids)
+        return [results.get(f"gen-{i}", []) for i in range(len(prompts))]

@@ -1,0 +1,180 @@
+"""Model runner: turns a StepPlan into one fused model forward + sampling.
+
+Builds the mixed prefill+decode batch (prefill tokens first, then one token
+per decoding sequence), drives the HIP kernels through the model, and samples
+next tokens (greedy via the HIP argmax kernel; stochastic via torch GPU ops).
+"""
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from xllm_service_amd import ops
+from xllm_service_amd.models.config import ModelConfig
+
+from .metadata import AttnMetadata
+from .scheduler import StepPlan
+from .sequence import Sequence
+
+BLOCK_SIZE = 16
+
+
+class ModelRunner:
+    def __init__(self, model, cfg: ModelConfig, device: torch.device,
+                 num_blocks: int, dtype=torch.bfloat16):
+        self.model = model
+        self.cfg = cfg
+        self.device = device
+        self.dtype = dtype
+        self.num_blocks = num_blocks
+        n_kv = model.local_kv_heads if hasattr(model, "local_kv_heads") else cfg.num_kv_heads
+        self.kv_caches: List[Tuple[torch.Tensor, torch.Tensor]] = [
+            (torch.zeros(num_blocks, n_kv, BLOCK_SIZE, cfg.head_dim,
+                         dtype=dtype, device=device),
+             torch.zeros(num_blocks, n_kv, BLOCK_SIZE, cfg.head_dim,
+                         dtype=dtype, device=device))
+            for _ in range(cfg.num_layers)
+        ]
+
+    @staticmethod
+    def kv_cache_blocks_for(cfg: ModelConfig, device: torch.device,
+                            n_kv_local: int,
+                            gpu_memory_utilization: float = 0.85,
+                            dtype_bytes: int = 2,
+                            max_blocks: Optional[int] = None) -> int:
+        """Size the block pool from actually-free HBM after weights load."""
+        if device.type == "cuda":
+            free, _total = torch.cuda.mem_get_info(device)
+            budget = int(free * gpu_memory_utilization)
+        else:
+            budget = 2 << 30  # CPU tests: 2 GiB worth of blocks
+        per_block = (2 * n_kv_local * BLOCK_SIZE * cfg.head_dim *
+                     dtype_bytes * cfg.num_layers)
+        n = max(budget // per_block, 16)
+        if max_blocks:
+            n = min(n, max_blocks)
+        return int(n)
+
+    # ---- batch construction -------------------------------------------------
+    def _build_batch(self, plan: StepPlan, bm):
+        tokens: List[int] = []
+        positions: List[int] = []
+        slots: List[int] = []
+
+        cu_q = [0]
+        p_seq_lens: List[int] = []
+        p_tables: List[List[int]] = []
+        for sp in plan.prefills:
+            seq = sp.seq
+            chunk_toks = seq.prompt_token_ids[sp.chunk_start:
+                                              sp.chunk_start + sp.chunk_len]
+            tokens.extend(chunk_toks)
+            positions.extend(range(sp.chunk_start, sp.chunk_start + sp.chunk_len))
+            for pos in range(sp.chunk_start, sp.chunk_start + sp.chunk_len):
+                blk = seq.block_table[pos // BLOCK_SIZE]
+                slots.append(blk * BLOCK_SIZE + pos % BLOCK_SIZE)
+            cu_q.append(cu_q[-1] + sp.chunk_len)
+            p_seq_lens.append(sp.chunk_start + sp.chunk_len)
+            p_tables.append(seq.block_table)
+
+        d_seq_lens: List[int] = []
+        d_tables: List[List[int]] = []
+        for seq in plan.decodes:
+            last_tok = (seq.output_token_ids[-1] if seq.output_token_ids
+                        else seq.prompt_token_ids[-1])
+            pos = seq.total_len - 1
+            tokens.append(last_tok)
+            positions.append(pos)
+            slots.append(bm.append_slot(seq))
+            d_seq_lens.append(seq.total_len)
+            d_tables.append(seq.block_table)
+
+        def pad_tables(tables: List[List[int]]) -> torch.Tensor:
+            if not tables:
+                return torch.zeros(0, 1, dtype=torch.int32, device=self.device)
+            w = max(len(t) for t in tables)
+            out = torch.zeros(len(tables), w, dtype=torch.int32)
+            for i, t in enumerate(tables):
+                out[i, :len(t)] = torch.tensor(t, dtype=torch.int32)
+            return out.to(self.device)
+
+        dev = self.device
+        np_ = cu_q[-1]
+        meta = AttnMetadata(
+            num_prefill_tokens=np_,
+            num_decode_tokens=len(plan.decodes),
+            slot_mapping=torch.tensor(slots, dtype=torch.long, device=dev),
+            cu_q=(torch.tensor(cu_q, dtype=torch.int32, device=dev)
+                  if plan.prefills else None),
+            prefill_seq_lens=(torch.tensor(p_seq_lens, dtype=torch.int32,
+                                           device=dev) if plan.prefills else None),
+            prefill_block_tables=pad_tables(p_tables) if plan.prefills else None,
+            decode_seq_lens=(torch.tensor(d_seq_lens, dtype=torch.int32,
+                                          device=dev) if plan.decodes else None),
+            decode_block_tables=pad_tables(d_tables) if plan.decodes else None,
+        )
+        input_ids = torch.tensor(tokens, dtype=torch.long, device=dev)
+        pos_t = torch.tensor(positions, dtype=torch.long, device=dev)
+        return input_ids, pos_t, meta
+
+    # ---- sampling -----------------------------------------------------------
+    def _sample(self, logits: torch.Tensor, seqs: List[Sequence]) -> List[int]:
+        greedy = all(s.params.greedy for s in seqs)
+        if greedy:
+            if logits.is_cuda and logits.dtype == torch.bfloat16:
+                return ops.greedy_sample(logits.contiguous()).tolist()
+            return logits.float().argmax(-1).tolist()
+        out: List[int] = []
+        probs_all = None
+        for i, seq in enumerate(seqs):
+            p = seq.params
+            if p.greedy:
+                out.append(int(logits[i].float().argmax()))
+                continue
+            lg = logits[i].float() / max(p.temperature, 1e-5)
+            if p.top_k > 0:
+                kth = torch.topk(lg, min(p.top_k, lg.numel())).values[-1]
+                lg[lg < kth] = -float("inf")
+            if p.top_p < 1.0:
+                sorted_lg, idx = lg.sort(descending=True)
+                probs = sorted_lg.softmax(-1)
+                cum = probs.cumsum(-1)
+                cut = (cum - probs) > p.top_p
+                sorted_lg[cut] = -float("inf")
+                lg = torch.full_like(lg, -float("inf")).scatter(0, idx, sorted_lg)
+            probs = lg.softmax(-1)
+            gen = None
+            if p.seed is not None:
+                gen = torch.Generator(device=probs.device).manual_seed(
+                    p.seed + seq.total_len)
+            out.append(int(torch.multinomial(probs, 1, generator=gen)))
+        return out
+
+    # ---- one step -----------------------------------------------------------
+    @torch.inference_mode()
+    def execute(self, plan: StepPlan, bm) -> Dict[str, int]:
+        """Run one step; returns {request_id: sampled_token} for sequences
+        that produced a token this step (completed prefills + decodes)."""
+        input_ids, positions, meta = self._build_batch(plan, bm)
+        hidden = self.model(input_ids, positions, self.kv_caches, meta)
+
+        # rows that need logits: last token of each COMPLETED prefill chunk,
+        # plus every decode row
+        rows: List[int] = []
+        seqs: List[Sequence] = []
+        for i, sp in enumerate(plan.prefills):
+            if sp.chunk_start + sp.chunk_len >= sp.seq.prompt_len:
+                rows.append(int(meta.cu_q[i + 1]) - 1)
+                seqs.append(sp.seq)
+        np_ = meta.num_prefill_tokens
+        for j, seq in enumerate(plan.decodes):
+            rows.append(np_ + j)
+            seqs.append(seq)
+        if not rows:
+            return {}
+        sel = hidden[torch.tensor(rows, dtype=torch.long, device=hidden.device)]
+        logits = self.model.compute_logits(sel)
+        tokens = self._sample(logits, seqs)
+        return {seq.request_id: tok for seq, tok in zip(seqs, tokens)}

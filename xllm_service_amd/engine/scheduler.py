@@ -1,0 +1,212 @@
+"""Worker-level continuous-batching scheduler.
+
+Each call to schedule() builds one engine step:
+  * running sequences decode one token each (preempting — offline first,
+    then newest-online-first — when the block pool is exhausted)
+  * waiting sequences are admitted with CHUNKED prefill under a per-step
+    token budget, online (priority 0) ahead of offline (priority 1)
+
+This is the per-instance half of the hybrid online/offline scheduling the
+service layer drives (reference: Request::offline request/request.h:41 and
+the SLO policy, SURVEY.md 2.5/2.7 — implemented for real here).
+"""
+from __future__ import annotations
+
+from collections import deque
+from dataclasses import dataclass, field
+from typing import Deque, List, Optional, Tuple
+
+from .block_manager import BlockManager
+from .sequence import Sequence, SeqStatus
+
+
+@dataclass
+class ScheduledPrefill:
+    seq: Sequence
+    chunk_start: int     # first prompt token this step
+    chunk_len: int
+
+
+@dataclass
+class StepPlan:
+    prefills: List[ScheduledPrefill] = field(default_factory=list)
+    decodes: List[Sequence] = field(default_factory=list)
+    preempted: List[Sequence] = field(default_factory=list)
+
+    @property
+    def empty(self) -> bool:
+        return not self.prefills and not self.decodes
+
+
+class EngineScheduler:
+    def __init__(self, block_manager: BlockManager,
+                 max_num_seqs: int = 256,
+                 max_batched_tokens: int = 8192,
+                 enable_chunked_prefill: bool = True):
+        self.bm = block_manager
+        self.max_num_seqs = max_num_seqs
+        self.max_batched_tokens = max_batched_tokens
+        self.enable_chunked_prefill = enable_chunked_prefill
+        self.waiting: Deque[Sequence] = deque()
+        self.running: List[Sequence] = []
+
+    # ---- API ----------------------------------------------------------------
+    def add(self, seq: Sequence):
+        if seq.priority == 0:
+            # online requests queue ahead of every offline request
+            idx = len(self.waiting)
+            for i, s in enumerate(self.waiting):
+                if s.priority > 0:
+                    idx = i
+                    break
+            self.waiting.insert(idx, seq)
+        else:
+            self.waiting.append(seq)
+
+    def abort(self, request_id: str) -> Optional[Sequence]:
+        for i, s in enumerate(self.running):
+            if s.request_id == request_id:
+                s.status = SeqStatus.FINISHED_ABORT
+                self.bm.free(s)
+                self.running.pop(i)
+                return s
+        for i, s in enumerate(self.waiting):
+            if s.request_id == request_id:
+                s.status = SeqStatus.FINISHED_ABORT
+                if s.block_table:
+                    self.bm.free(s)
+                del self.waiting[i]
+                return s
+        return None
+
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    @property
+    def num_waiting(self) -> int:
+        return len(self.waiting)
+
+    # ---- core ---------------------------------------------------------------
+    def _preempt_one(self) -> bool:
+        """Preempt the lowest-priority, most recent running sequence."""
+        if not self.running:
+            return False
+        victim_idx = None
+        # offline victims first, most recent first
+        for i in range(len(self.running) - 1, -1, -1):
+            if self.running[i].priority > 0:
+                victim_idx = i
+                break
+        if victim_idx is None:
+            victim_idx = len(self.running) - 1
+        victim = self.running.pop(victim_idx)
+        self.bm.free(victim)
+        victim.status = SeqStatus.PREEMPTED
+        victim.num_computed_tokens = 0
+        victim.preempt_count += 1
+        # recompute path: prompt grows by generated tokens so far
+        victim.prompt_token_ids = victim.all_token_ids()
+        victim.output_token_ids = []
+        self.waiting.appendleft(victim)
+        return True
+
+    def schedule(self) -> StepPlan:
+        plan = StepPlan()
+        budget = self.max_batched_tokens
+
+        # 1. decodes for all running seqs (preempt on OOM)
+        for seq in list(self.running):
+            if seq not in self.running:      # became a preemption victim
+                continue
+            if not seq.prefill_done:
+                continue                     # mid-chunked-prefill: step 2
+            while (not self.bm.can_append(seq)
+                   and self._preempt_victim_excluding(seq, plan)):
+                pass
+            if self.bm.can_append(seq):
+                plan.decodes.append(seq)
+                budget -= 1
+            else:                            # pool exhausted: preempt self
+                self.running.remove(seq)
+                self.bm.free(seq)
+                seq.status = SeqStatus.PREEMPTED
+                seq.num_computed_tokens = 0
+                seq.preempt_count += 1
+                seq.prompt_token_ids = seq.all_token_ids()
+                seq.output_token_ids = []
+                self.waiting.appendleft(seq)
+                plan.preempted.append(seq)
+
+        # 2. continue chunked prefills already running
+        for seq in self.running:
+            if seq.prefill_done or budget <= 0:
+                continue
+            chunk = min(seq.prompt_len - seq.num_computed_tokens, budget)
+            if chunk > 0:
+                plan.prefills.append(
+                    ScheduledPrefill(seq, seq.num_computed_tokens, chunk))
+                budget -= chunk
+
+        # 3. admit waiting sequences
+        while self.waiting and budget > 0 and len(self.running) < self.max_num_seqs:
+            seq = self.waiting[0]
+            first_alloc = not seq.block_table
+            if first_alloc:
+                if not self.bm.can_allocate(seq, seq.prompt_len):
+                    break
+                self.bm.allocate_prefill(seq)
+            remaining = seq.prompt_len - seq.num_computed_tokens
+            chunk = min(remaining, budget)
+            if not self.enable_chunked_prefill and chunk < remaining:
+                if first_alloc:
+                    self.bm.free(seq)
+                    seq.num_computed_tokens = 0
+                break
+            if chunk <= 0:
+                break
+            self.waiting.popleft()
+            seq.status = SeqStatus.RUNNING
+            self.running.append(seq)
+            plan.prefills.append(
+                ScheduledPrefill(seq, seq.num_computed_tokens, chunk))
+            budget -= chunk
+
+        return plan
+
+    def _preempt_victim_excluding(self, protected: Sequence,
+                                  plan: StepPlan) -> bool:
+        candidates = [s for s in self.running
+                      if s is not protected and s not in plan.decodes]
+        if not candidates:
+            return False
+        # offline first, then most recent online
+        victim = None
+        for s in reversed(candidates):
+            if s.priority > 0:
+                victim = s
+                break
+        if victim is None:
+            victim = candidates[-1]
+        self.running.remove(victim)
+        self.bm.free(victim)
+        victim.status = SeqStatus.PREEMPTED
+        victim.num_computed_tokens = 0
+        victim.preempt_count += 1
+        victim.prompt_token_ids = victim.all_token_ids()
+        victim.output_token_ids = []
+        self.waiting.appendleft(victim)
+        plan.preempted.append(victim)
+        return True
+
+    # ---- bookkeeping after a step -------------------------------------------
+    def on_step_done(self, plan: StepPlan):
+        """Advance prefill progress; register full prefix blocks; retire
+        finished sequences."""
+        for sp in plan.prefills:
+            sp.seq.num_computed_tokens += sp.chunk_len
+            if sp.seq.prefill_done:
+                self.bm.register_full_blocks(sp.seq)
+        done = [s for s in self.running if s.status.finished]
+        for s in done:
+            self.bm.free(s)
+            self.running.remove(s)

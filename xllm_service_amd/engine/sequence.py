@@ -1,0 +1,79 @@
+"""Sequence (request) state inside a worker engine."""
+from __future__ import annotations
+
+import enum
+import time
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+from .sampling import SamplingParams
+
+
+class SeqStatus(enum.Enum):
+    WAITING = "waiting"
+    RUNNING = "running"
+    PREEMPTED = "preempted"
+    FINISHED_STOP = "finished_stop"      # stop/eos token
+    FINISHED_LENGTH = "finished_length"  # max_tokens reached
+    FINISHED_ABORT = "finished_abort"    # cancelled (client disconnect etc.)
+
+    @property
+    def finished(self) -> bool:
+        return self in (SeqStatus.FINISHED_STOP, SeqStatus.FINISHED_LENGTH,
+                        SeqStatus.FINISHED_ABORT)
+
+
+@dataclass
+class Sequence:
+    request_id: str
+    prompt_token_ids: List[int]
+    params: SamplingParams
+    eos_token_id: Optional[int] = None
+    priority: int = 0           # 0 = online, 1 = offline (online preempts offline)
+    arrival_time: float = field(default_factory=time.monotonic)
+
+    status: SeqStatus = SeqStatus.WAITING
+    output_token_ids: List[int] = field(default_factory=list)
+    block_table: List[int] = field(default_factory=list)
+    num_computed_tokens: int = 0          # prompt tokens already prefilled
+    preempt_count: int = 0
+    # PD-disaggregation: set on a decode instance receiving a migrated prefill
+    migrated_in: bool = False
+    first_token_time: Optional[float] = None
+    cumulative_logprob: float = 0.0
+
+    @property
+    def prompt_len(self) -> int:
+        return len(self.prompt_token_ids)
+
+    @property
+    def total_len(self) -> int:
+        return self.prompt_len + len(self.output_token_ids)
+
+    @property
+    def prefill_done(self) -> bool:
+        return self.num_computed_tokens >= self.prompt_len
+
+    def all_token_ids(self) -> List[int]:
+        return self.prompt_token_ids + self.output_token_ids
+
+    def append_token(self, token_id: int):
+        self.output_token_ids.append(token_id)
+        if self.first_token_time is None:
+            self.first_token_time = time.monotonic()
+
+    def check_finish(self) -> bool:
+        out = self.output_token_ids
+        if len(out) >= self.params.max_tokens:
+            self.status = SeqStatus.FINISHED_LENGTH
+            return True
+        if len(out) >= self.params.min_tokens and out:
+            last = out[-1]
+            if (not self.params.ignore_eos and self.eos_token_id is not None
+                    and last == self.eos_token_id):
+                self.status = SeqStatus.FINISHED_STOP
+                return True
+            if last in self.params.stop_token_ids:
+                self.status = SeqStatus.FINISHED_STOP
+                return True
+        return False

@@ -1,0 +1,69 @@
+"""Model configuration + named presets (random-init; no network access)."""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+
+
+@dataclass
+class ModelConfig:
+    name: str = "llama"
+    architecture: str = "llama"          # llama | opt | qwen2_vl
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    head_dim: int = 128
+    rope_theta: float = 500000.0
+    max_position: int = 8192
+    rms_eps: float = 1e-5
+    tie_word_embeddings: bool = False
+    hidden_act: str = "silu"
+    # vision tower (qwen2_vl only)
+    vision: dict = field(default_factory=dict)
+
+    @property
+    def q_size(self) -> int:
+        return self.num_heads * self.head_dim
+
+    @property
+    def kv_size(self) -> int:
+        return self.num_kv_heads * self.head_dim
+
+
+PRESETS = {
+    # flagship bench model
+    "llama-3-8b": ModelConfig(
+        name="llama-3-8b", vocab_size=128256, hidden_size=4096,
+        intermediate_size=14336, num_layers=32, num_heads=32, num_kv_heads=8,
+        head_dim=128, rope_theta=500000.0, max_position=8192),
+    # small CPU-path / test configs
+    "llama-tiny": ModelConfig(
+        name="llama-tiny", vocab_size=1024, hidden_size=256,
+        intermediate_size=512, num_layers=2, num_heads=4, num_kv_heads=2,
+        head_dim=64, rope_theta=10000.0, max_position=2048),
+    "llama-debug-128": ModelConfig(
+        name="llama-debug-128", vocab_size=512, hidden_size=512,
+        intermediate_size=1024, num_layers=2, num_heads=4, num_kv_heads=2,
+        head_dim=128, rope_theta=10000.0, max_position=2048),
+    # OPT-125m (CPU control-plane check)
+    "opt-125m": ModelConfig(
+        name="opt-125m", architecture="opt", vocab_size=50272,
+        hidden_size=768, intermediate_size=3072, num_layers=12, num_heads=12,
+        num_kv_heads=12, head_dim=64, max_position=2048, hidden_act="gelu"),
+    # Qwen2-VL-7B — language tower dims
+    "qwen2-vl-7b": ModelConfig(
+        name="qwen2-vl-7b", architecture="qwen2_vl", vocab_size=152064,
+        hidden_size=3584, intermediate_size=18944, num_layers=28,
+        num_heads=28, num_kv_heads=4, head_dim=128, rope_theta=1000000.0,
+        max_position=8192,
+        vision=dict(depth=32, embed_dim=1280, num_heads=16, patch_size=14,
+                    spatial_merge_size=2, out_hidden_size=3584)),
+}
+
+
+def get_config(name: str) -> ModelConfig:
+    if name not in PRESETS:
+        raise KeyError(f"unknown model preset {name!r}; have {list(PRESETS)}")
+    return PRESETS[name]

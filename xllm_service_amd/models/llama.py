@@ -1,0 +1,153 @@
+"""Llama-family decoder (Llama-3 / Qwen2 text tower) on the paged-KV engine.
+
+GEMMs go through hipBLASLt (torch linear); everything else on the hot path is
+a hand-written gfx950 HIP op: fused(add+)RMSNorm, fused RoPE, paged-attention
+prefill/decode, fused SwiGLU. Covers BASELINE.md configs 2-4.
+"""
+from __future__ import annotations
+
+import math
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from xllm_service_amd import ops
+from xllm_service_amd.distributed import parallel_state as ps
+from xllm_service_amd.distributed.layers import (ColumnParallelLinear,
+                                                 MergedColumnParallelLinear,
+                                                 RowParallelLinear)
+from xllm_service_amd.engine.metadata import AttnMetadata
+from xllm_service_amd.models.config import ModelConfig
+from xllm_service_amd.ops import ref as op_ref
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, cfg: ModelConfig, dtype):
+        super().__init__()
+        tp = ps.tp_size()
+        self.cfg = cfg
+        self.n_heads = cfg.num_heads // tp
+        self.n_kv_heads = max(cfg.num_kv_heads // tp, 1)
+        self.head_dim = cfg.head_dim
+        self.scale = 1.0 / math.sqrt(cfg.head_dim)
+        self.qkv_proj = MergedColumnParallelLinear(
+            cfg.hidden_size,
+            [cfg.q_size, cfg.kv_size, cfg.kv_size],
+            dtype=dtype)
+        self.o_proj = RowParallelLinear(cfg.q_size, cfg.hidden_size, dtype=dtype)
+
+    def forward(self, x, positions, kv_cache, meta: AttnMetadata, cos_sin):
+        qkv = self.qkv_proj(qkv_in := x)
+        q_sz = self.n_heads * self.head_dim
+        kv_sz = self.n_kv_heads * self.head_dim
+        q, k, v = torch.split(qkv, [q_sz, kv_sz, kv_sz], dim=-1)
+        q = q.contiguous()
+        k = k.contiguous()
+        q, k = ops.rope(positions, q, k, cos_sin, self.head_dim, self.head_dim)
+        T = x.shape[0]
+        qh = q.view(T, self.n_heads, self.head_dim)
+        kh = k.view(T, self.n_kv_heads, self.head_dim)
+        vh = v.view(T, self.n_kv_heads, self.head_dim).contiguous()
+        k_cache, v_cache = kv_cache
+        ops.reshape_and_cache(kh, vh, k_cache, v_cache, meta.slot_mapping)
+
+        np_, nd = meta.num_prefill_tokens, meta.num_decode_tokens
+        out = torch.empty_like(qh)
+        if np_:
+            out[:np_] = ops.paged_attn_prefill(
+                qh[:np_].contiguous(), k_cache, v_cache,
+                meta.prefill_block_tables, meta.cu_q, meta.prefill_seq_lens,
+                self.scale)
+        if nd:
+            out[np_:] = ops.paged_attn_decode(
+                qh[np_:].contiguous(), k_cache, v_cache,
+                meta.decode_block_tables, meta.decode_seq_lens, self.scale)
+        return self.o_proj(out.view(T, q_sz))
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, cfg: ModelConfig, dtype):
+        super().__init__()
+        self.gate_up = MergedColumnParallelLinear(
+            cfg.hidden_size, [cfg.intermediate_size, cfg.intermediate_size],
+            dtype=dtype)
+        self.down = RowParallelLinear(cfg.intermediate_size, cfg.hidden_size,
+                                      dtype=dtype)
+
+    def forward(self, x):
+        return self.down(ops.silu_and_mul(self.gate_up(x)))
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig, dtype):
+        super().__init__()
+        self.input_norm = nn.Parameter(torch.ones(cfg.hidden_size, dtype=dtype))
+        self.post_norm = nn.Parameter(torch.ones(cfg.hidden_size, dtype=dtype))
+        self.attn = LlamaAttention(cfg, dtype)
+        self.mlp = LlamaMLP(cfg, dtype)
+        self.eps = cfg.rms_eps
+
+    def forward(self, x, residual, positions, kv_cache, meta, cos_sin):
+        if residual is None:
+            residual = x
+            x = ops.rmsnorm(x, self.input_norm, self.eps)
+        else:
+            x, residual = ops.fused_add_rmsnorm(x, residual, self.input_norm,
+                                                self.eps)
+        x = self.attn(x, positions, kv_cache, meta, cos_sin)
+        x, residual = ops.fused_add_rmsnorm(x, residual, self.post_norm, self.eps)
+        x = self.mlp(x)
+        return x, residual
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16):
+        super().__init__()
+        self.cfg = cfg
+        self.dtype = dtype
+        self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size, dtype=dtype)
+        self.layers = nn.ModuleList(
+            [LlamaDecoderLayer(cfg, dtype) for _ in range(cfg.num_layers)])
+        self.final_norm = nn.Parameter(torch.ones(cfg.hidden_size, dtype=dtype))
+        if cfg.tie_word_embeddings:
+            self.lm_head = None
+        else:
+            self.lm_head = ColumnParallelLinear(cfg.hidden_size, cfg.vocab_size,
+                                                dtype=dtype)
+        cs = op_ref.rope_table(cfg.head_dim, cfg.max_position, cfg.rope_theta)
+        self.register_buffer("cos_sin", cs, persistent=False)
+
+    # number of kv heads actually stored on this rank (TP-sharded)
+    @property
+    def local_kv_heads(self) -> int:
+        return max(self.cfg.num_kv_heads // ps.tp_size(), 1)
+
+    def random_init(self, seed: int = 0):
+        """Deterministic random init (no checkpoints available offline)."""
+        gen = torch.Generator(device="cpu").manual_seed(seed)
+        std = 1.0 / math.sqrt(self.cfg.hidden_size)
+        for name, p in self.named_parameters():
+            if "norm" in name:
+                continue
+            with torch.no_grad():
+                vals = torch.randn(p.shape, generator=gen, dtype=torch.float32)
+                p.copy_((vals * std).to(p.dtype))
+
+    def forward(self, input_ids, positions, kv_caches: List[Tuple],
+                meta: AttnMetadata,
+                inputs_embeds: Optional[torch.Tensor] = None):
+        x = inputs_embeds if inputs_embeds is not None else self.embed(input_ids)
+        residual = None
+        for i, layer in enumerate(self.layers):
+            x, residual = layer(x, residual, positions, kv_caches[i], meta,
+                                self.cos_sin)
+        x, _ = ops.fused_add_rmsnorm(x, residual, self.final_norm, self.cfg.rms_eps)
+        return x
+
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        if self.lm_head is not None:
+            logits = self.lm_head(hidden)
+        else:
+            logits = hidden @ self.embed.weight.t()
+        return ps.tp_all_gather(logits, dim=-1)
