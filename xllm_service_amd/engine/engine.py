@@ -59,9 +59,9 @@ class LLMEngine:
             dtype = torch.float32  # CPU reference path runs fp32
         self.dtype = dtype
 
-        self.model = create_model(self.cfg, dtype=dtype)
+        self.model = create_model(self.cfg, dtype=dtype).to(self.device)
         self.model.random_init(seed)
-        self.model = self.model.to(self.device).eval()
+        self.model = self.model.eval()
 
         n_kv_local = getattr(self.model, "local_kv_heads", self.cfg.num_kv_heads)
         num_blocks = ModelRunner.kv_cache_blocks_for(

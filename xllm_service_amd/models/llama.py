@@ -124,14 +124,17 @@ class LlamaForCausalLM(nn.Module):
         return max(self.cfg.num_kv_heads // ps.tp_size(), 1)
 
     def random_init(self, seed: int = 0):
-        """Deterministic random init (no checkpoints available offline)."""
-        gen = torch.Generator(device="cpu").manual_seed(seed)
+        """Deterministic random init (no checkpoints available offline).
+        Generates on the parameters' device (fast path for the 8B model)."""
+        dev = next(self.parameters()).device
+        gen = torch.Generator(device=dev).manual_seed(seed)
         std = 1.0 / math.sqrt(self.cfg.hidden_size)
         for name, p in self.named_parameters():
             if "norm" in name:
                 continue
             with torch.no_grad():
-                vals = torch.randn(p.shape, generator=gen, dtype=torch.float32)
+                vals = torch.randn(p.shape, generator=gen, device=dev,
+                                   dtype=torch.float32)
                 p.copy_((vals * std).to(p.dtype))
 
     def forward(self, input_ids, positions, kv_caches: List[Tuple],
