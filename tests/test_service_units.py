@@ -1,0 +1,208 @@
+"""Unit tests for service-layer components: hashing, global KV index,
+policies, parsers, chat template, tokenizers, predictors."""
+import pytest
+
+from xllm_service_amd.chat_template import JinjaChatTemplate
+from xllm_service_amd.service.instance_mgr import Instance
+from xllm_service_amd.service.kvcache_mgr import GlobalKVCacheMgr
+from xllm_service_amd.service.parsers import (infer_model_family,
+                                              make_parsers,
+                                              make_stream_parsers)
+from xllm_service_amd.service.time_predictor import (TPOTPredictor,
+                                                     TTFTPredictor)
+from xllm_service_amd.service.types import InstanceMetaInfo, LoadMetrics
+from xllm_service_amd.tokenizer import (ByteTokenizer, IncrementalDecoder,
+                                        TokenizerFactory)
+from xllm_service_amd.utils.hashing import chain_block_hashes
+
+
+def test_chained_hash_prefix_property():
+    toks = list(range(64))
+    h1 = chain_block_hashes(toks, 16)
+    h2 = chain_block_hashes(toks[:32] + [999] * 32, 16)
+    assert len(h1) == 4
+    assert h1[:2] == h2[:2]          # shared prefix -> same chain
+    assert h1[2:] != h2[2:]          # divergence propagates
+    # chain pins the WHOLE prefix: same block content, different prefix
+    h3 = chain_block_hashes([7] * 16 + toks[16:32], 16)
+    assert h3[1] != h1[1]
+
+
+def test_global_kvcache_match_and_tiers():
+    kv = GlobalKVCacheMgr(block_size=16)
+    toks = list(range(48))
+    hashes = chain_block_hashes(toks, 16)
+    kv.record_updated_kvcaches("i1", stored=hashes[:2], removed=[])
+    kv.record_updated_kvcaches("i2", stored=hashes[:3], removed=[])
+    ov = kv.match(toks)
+    assert ov.matched_blocks == 3
+    assert ov.scores["i2"] > ov.scores["i1"]
+    # removal shrinks the walk
+    kv.record_updated_kvcaches("i2", stored=[], removed=[hashes[2]])
+    ov = kv.match(toks)
+    assert ov.matched_blocks == 2
+    # offload moves tier and lowers weight
+    kv.record_updated_kvcaches("i1", stored=[], removed=[],
+                               offloaded=[hashes[0]])
+    ov = kv.match(toks)
+    assert 0 < ov.scores["i1"] < ov.scores["i2"]
+
+
+class _FakeMgr:
+    def __init__(self, prefills, decodes):
+        self._p, self._d = prefills, decodes
+        self.instances = {i.name: i for i in prefills + decodes}
+        self.prefill_index = [i.name for i in prefills]
+        self.decode_index = [i.name for i in decodes]
+
+    def schedulable_prefills(self):
+        return self._p
+
+    def schedulable_decodes(self):
+        return self._d
+
+    def flip_instance_role(self, name, side):
+        return False
+
+
+def _inst(name, itype="PREFILL", waiting=0, cache=0.0):
+    i = Instance(InstanceMetaInfo(name=name, itype=itype))
+    i.load = LoadMetrics(waiting_requests_num=waiting,
+                         gpu_cache_usage_perc=cache)
+    return i
+
+
+def test_cache_aware_policy_prefers_prefix_holder():
+    from xllm_service_amd.service.policies import CacheAwarePolicy
+    kv = GlobalKVCacheMgr(block_size=16)
+    toks = list(range(64))
+    hashes = chain_block_hashes(toks, 16)
+    kv.record_updated_kvcaches("p1", stored=hashes, removed=[])
+    p0, p1 = _inst("p0"), _inst("p1")
+    d0 = _inst("d0", "DECODE")
+    pol = CacheAwarePolicy(_FakeMgr([p0, p1], [d0]), kv)
+    pair = pol.select_instances_pair(toks)
+    assert pair.prefill.name == "p1"
+    # but heavy load on the cache holder flips the choice
+    p1.load = LoadMetrics(waiting_requests_num=64, gpu_cache_usage_perc=0.99)
+    pair = pol.select_instances_pair(toks)
+    assert pair.prefill.name == "p0"
+
+
+def test_slo_policy_picks_fast_decode():
+    from xllm_service_amd.service.policies import SloAwarePolicy
+    kv = GlobalKVCacheMgr(block_size=16)
+    p0 = _inst("p0")
+    d_slow, d_fast = _inst("ds", "DECODE"), _inst("df", "DECODE")
+    pol = SloAwarePolicy(_FakeMgr([p0], [d_slow, d_fast]), kv,
+                         target_tpot_ms=50.0)
+    for b in range(1, 20):
+        pol.observe_tpot("ds", b, 128, 40.0 + 12.0 * b)   # slow instance
+        pol.observe_tpot("df", b, 128, 10.0 + 1.0 * b)    # fast instance
+    d_slow.num_decoding = 4
+    d_fast.num_decoding = 4
+    pair = pol.select_instances_pair(list(range(128)))
+    assert pair.decode.name == "df"
+
+
+def test_ttft_tpot_predictors_fit():
+    tt = TTFTPredictor()
+    for n in range(10, 200, 10):
+        tt.add_sample(n, 5 + 0.1 * n + 0.001 * n * n)
+    assert abs(tt.predict(100) - (5 + 10 + 10)) < 1.5
+    tp = TPOTPredictor()
+    for b in range(1, 20):
+        tp.add_sample(b, 100, 8 + 2 * b + 0.01 * 100)
+    assert abs(tp.predict(10, 100) - (8 + 20 + 1)) < 1.0
+
+
+def test_model_family_inference():
+    assert infer_model_family("Qwen3-32B") == "qwen3"
+    assert infer_model_family("qwen2-vl-7b") == "qwen2"
+    assert infer_model_family("DeepSeek-V3") == "deepseek_v3"
+    assert infer_model_family("Kimi-K2-Instruct") == "kimi_k2"
+    assert infer_model_family("unknown-model-7b") is None
+    rp, tp = make_parsers("unknown-model-7b")
+    assert rp is None and tp is None  # auto silently disables
+
+
+def test_reasoning_and_toolcall_nonstream():
+    rp, tp = make_parsers("Qwen3-8B")
+    text = ("<think>let me think\nhard</think>\nThe answer.\n"
+            '<tool_call>{"name": "get_weather", "arguments": {"city": "SF"}}'
+            "</tool_call>")
+    reasoning, rest = rp.extract(text)
+    assert reasoning == "let me think\nhard"
+    content, calls = tp.extract(rest)
+    assert content == "The answer."
+    assert len(calls) == 1 and calls[0].name == "get_weather"
+    assert '"city"' in calls[0].arguments
+
+
+def test_streaming_parsers_chunked():
+    rp, tp = make_stream_parsers("Qwen3-8B")
+    full = ('<think>abc</think>hello <tool_call>{"name": "f", '
+            '"arguments": {}}</tool_call> bye')
+    reason = content = ""
+    calls = []
+    for i in range(0, len(full), 3):   # ragged chunks
+        r, c = rp.feed(full[i:i + 3])
+        reason += r
+        c2, cl = tp.feed(c)
+        content += c2
+        calls.extend(cl)
+    calls.extend(tp.flush())
+    assert reason == "abc"
+    assert content == "hello  bye"
+    assert len(calls) == 1 and calls[0].name == "f"
+
+
+def test_chat_template_default_and_custom():
+    ct = JinjaChatTemplate()
+    out = ct.apply([{"role": "user", "content": "hi"}])
+    assert "<|im_start|>user\nhi<|im_end|>" in out
+    assert out.endswith("<|im_start|>assistant\n")
+    # custom template with kwargs + tools suppression
+    ct2 = JinjaChatTemplate(
+        "{% if tools %}TOOLS:{{ tools|length }} {% endif %}"
+        "{% for m in messages %}{{ m['content'] }}{% endfor %}"
+        "{% if extra %}E={{ extra }}{% endif %}")
+    msgs = [{"role": "user", "content": "x"}]
+    tools = [{"type": "function", "function": {"name": "f"}}]
+    assert ct2.apply(msgs, tools=tools) == "TOOLS:1 x"
+    assert ct2.apply(msgs, tools=tools, tool_choice="none") == "x"
+    assert ct2.apply(msgs, chat_template_kwargs={"extra": 1}) == "xE=1"
+
+
+def test_byte_tokenizer_roundtrip_and_incremental():
+    tk = ByteTokenizer()
+    ids = tk.encode("hello é world")
+    assert tk.decode(ids) == "hello é world"
+    dec = IncrementalDecoder(tk)
+    text = ""
+    for i in ids:
+        text += dec.push([i])
+    assert text == "hello é world"
+
+
+def test_tiktoken_tokenizer(tmp_path):
+    import base64
+    vocab = {}
+    rank = 0
+    for b in range(256):
+        vocab[bytes([b])] = rank
+        rank += 1
+    for merged in [b"he", b"ll", b"llo", b"hello", b" wo", b"rld"]:
+        vocab[merged] = rank
+        rank += 1
+    path = tmp_path / "test.tiktoken"
+    with open(path, "wb") as f:
+        for tok, r in vocab.items():
+            f.write(base64.b64encode(tok) + b" " + str(r).encode() + b"\n")
+    from xllm_service_amd.tokenizer.tiktoken_tok import TiktokenTokenizer
+    tk = TiktokenTokenizer(str(path))
+    ids = tk.encode("hello world")
+    assert tk.decode(ids) == "hello world"
+    assert len(ids) < len("hello world")  # merges happened
+    factory_tk = TokenizerFactory.create(str(tmp_path))
+    assert type(factory_tk).__name__ == "TiktokenTokenizer"

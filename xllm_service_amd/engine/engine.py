@@ -77,6 +77,7 @@ class LLMEngine:
         self.runner = ModelRunner(self.model, self.cfg, self.device,
                                   num_blocks, dtype=dtype)
         self.seqs: Dict[str, Sequence] = {}
+        self.held: Dict[str, Sequence] = {}   # finished, blocks kept (PD)
         self.stats = EngineStats()
         self.eos_token_id: Optional[int] = None  # set by tokenizer owner
 
@@ -84,7 +85,8 @@ class LLMEngine:
     def add_request(self, request_id: str, prompt_token_ids: List[int],
                     params: Optional[SamplingParams] = None,
                     priority: int = 0,
-                    eos_token_id: Optional[int] = None) -> None:
+                    eos_token_id: Optional[int] = None,
+                    hold_blocks: bool = False) -> None:
         if request_id in self.seqs:
             raise ValueError(f"duplicate request_id {request_id}")
         seq = Sequence(request_id=request_id,
@@ -92,9 +94,57 @@ class LLMEngine:
                        params=params or SamplingParams(),
                        eos_token_id=eos_token_id if eos_token_id is not None
                        else self.eos_token_id,
-                       priority=priority)
+                       priority=priority,
+                       hold_blocks=hold_blocks)
         self.seqs[request_id] = seq
         self.scheduler.add(seq)
+
+    # ---- PD-disaggregation support -----------------------------------------
+    def held_block_table(self, request_id: str) -> List[int]:
+        """Block ids of a finished-but-held sequence (prefill side)."""
+        return list(self.held[request_id].block_table)
+
+    def release_held(self, request_id: str) -> None:
+        seq = self.held.pop(request_id, None)
+        if seq is not None:
+            self.block_manager.free(seq)
+
+    def alloc_migration_blocks(self, n: int) -> List[int]:
+        """Allocate raw destination blocks for an incoming KV migration."""
+        return self.block_manager.allocate_raw(n)
+
+    def free_blocks(self, block_ids: List[int]) -> None:
+        from .sequence import Sequence as _S
+        tmp = _S("_tmp", [], SamplingParams())
+        tmp.block_table = list(block_ids)
+        self.block_manager.free(tmp)
+
+    def activate_migrated_request(self, request_id: str,
+                                  prompt_token_ids: List[int],
+                                  first_token_ids: List[int],
+                                  block_ids: List[int],
+                                  params: Optional[SamplingParams] = None,
+                                  eos_token_id: Optional[int] = None,
+                                  priority: int = 0) -> None:
+        """Resume a request whose prompt KV was migrated into block_ids
+        (decode side). No recompute: decode continues from the first
+        prefill-produced token."""
+        seq = Sequence(request_id=request_id,
+                       prompt_token_ids=list(prompt_token_ids),
+                       params=params or SamplingParams(),
+                       eos_token_id=eos_token_id if eos_token_id is not None
+                       else self.eos_token_id,
+                       priority=priority)
+        seq.block_table = list(block_ids)
+        seq.num_computed_tokens = seq.prompt_len
+        seq.output_token_ids = list(first_token_ids)
+        seq.migrated_in = True
+        seq.status = SeqStatus.RUNNING
+        if seq.first_token_time is None:
+            import time as _t
+            seq.first_token_time = _t.monotonic()
+        self.seqs[request_id] = seq
+        self.scheduler.running.append(seq)
 
     def abort_request(self, request_id: str) -> bool:
         seq = self.scheduler.abort(request_id)
@@ -136,12 +186,40 @@ class LLMEngine:
         self.scheduler.on_step_done(plan)
         for out in outputs:
             if out.finished:
-                self.seqs.pop(out.request_id, None)
+                seq = self.seqs.pop(out.request_id, None)
+                if seq is not None and seq.hold_blocks:
+                    self.held[out.request_id] = seq
         self.stats.steps += 1
         self.stats.num_waiting = self.scheduler.num_waiting
         self.stats.num_running = len(self.scheduler.running)
         self.stats.kv_usage = self.block_manager.usage()
         return outputs
+
+    def export_block_bytes(self, block_ids: List[int]) -> bytes:
+        """Serialize KV blocks (all layers, k then v) — the DRAM/RPC
+        migration transport (CPU path + cross-node fallback; the same-node
+        GPU path is xGMI P2P in engine/kv_migration.py)."""
+        idx = torch.tensor(block_ids, dtype=torch.long, device=self.device)
+        parts = []
+        for (kc, vc) in self.runner.kv_caches:
+            for c in (kc, vc):
+                t = c.index_select(0, idx).contiguous()
+                parts.append(t.view(torch.uint8).cpu().numpy().tobytes())
+        return b"".join(parts)
+
+    def import_block_bytes(self, block_ids: List[int], data: bytes) -> None:
+        import numpy as np
+        idx = torch.tensor(block_ids, dtype=torch.long, device=self.device)
+        kc0 = self.runner.kv_caches[0][0]
+        per = len(block_ids) * kc0[0].numel() * kc0.element_size()
+        off = 0
+        for (kc, vc) in self.runner.kv_caches:
+            for c in (kc, vc):
+                raw = np.frombuffer(data, dtype=np.uint8, count=per, offset=off)
+                t = torch.from_numpy(raw.copy()).view(c.dtype).reshape(
+                    len(block_ids), *c.shape[1:])
+                c[idx] = t.to(self.device)
+                off += per
 
     # ---- convenience (tests, smoke) ----------------------------------------
     def generate(self, prompts: List[List[int]],

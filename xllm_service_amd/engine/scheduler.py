@@ -208,5 +208,6 @@ class EngineScheduler:
                 self.bm.register_full_blocks(sp.seq)
         done = [s for s in self.running if s.status.finished]
         for s in done:
-            self.bm.free(s)
+            if not s.hold_blocks:
+                self.bm.free(s)
             self.running.remove(s)

@@ -39,6 +39,9 @@ class Sequence:
     preempt_count: int = 0
     # PD-disaggregation: set on a decode instance receiving a migrated prefill
     migrated_in: bool = False
+    # PD-disaggregation: keep KV blocks alive after finish (prefill side
+    # holds them until the decode instance has pulled the blocks)
+    hold_blocks: bool = False
     first_token_time: Optional[float] = None
     cumulative_logprob: float = 0.0
 

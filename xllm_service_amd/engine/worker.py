@@ -1,0 +1,481 @@
+"""Worker process: one engine per GPU, speaking the instance-side contract.
+
+Implements everything SURVEY.md §2.9 requires of an instance:
+  1. serves execute_request (the XllmAPIService Completions equivalent)
+  2. serves link_instance / unlink_instance with peer cluster info
+  3. registers itself in the registry under XLLM:<TYPE>:<name> with a TTL
+     lease + incarnation id
+  4. heartbeats the master every ~3 s with LoadMetrics + KvCacheEvents +
+     LatencyMetrics
+  5. pushes generated tokens to the master ("generations", batched), with
+     finished_on_prefill marking the TTFT token
+  6. serves health probes
+plus PD-disaggregation: prefill role computes the prompt + first token,
+holds the KV blocks, migrates them to the decode peer (xGMI P2P on GPU via
+kv_migration.py; serialized-bytes RPC transport elsewhere), then releases.
+
+Threading model: asyncio loop for RPC/registry; the engine runs on its own
+thread, fed through a command queue; outputs hop back via a thread-safe
+queue drained by the push task.
+"""
+from __future__ import annotations
+
+import asyncio
+import concurrent.futures
+import logging
+import queue
+import socket
+import threading
+import time
+from typing import Any, Dict, List, Optional
+
+import torch
+
+from xllm_service_amd.registry.server import RegistryClient
+from xllm_service_amd.service.types import (KEY_INSTANCE, KEY_MASTER,
+                                            InstanceMetaInfo, InstanceType)
+from xllm_service_amd.utils import msgrpc
+
+from .engine import LLMEngine
+from .sampling import SamplingParams
+
+log = logging.getLogger("xllm.worker")
+
+
+def params_from_dict(d: Dict[str, Any]) -> SamplingParams:
+    d = d or {}
+    return SamplingParams(
+        temperature=float(d.get("temperature", 0.0) or 0.0),
+        top_p=float(d.get("top_p", 1.0) or 1.0),
+        top_k=int(d.get("top_k", -1) or -1),
+        max_tokens=int(d.get("max_tokens", 128) or 128),
+        min_tokens=int(d.get("min_tokens", 0) or 0),
+        stop_token_ids=list(d.get("stop_token_ids") or []),
+        ignore_eos=bool(d.get("ignore_eos", False)),
+        seed=d.get("seed"),
+        logprobs=d.get("logprobs"),
+    )
+
+
+class Worker:
+    def __init__(self, name: str, itype: str = "DEFAULT",
+                 model: str = "llama-tiny", device: Optional[str] = None,
+                 registry_host: str = "127.0.0.1", registry_port: int = 0,
+                 rpc_host: str = "127.0.0.1", rpc_port: int = 0,
+                 heartbeat_s: float = 1.0, lease_ttl_s: float = 3.0,
+                 eos_token_id: Optional[int] = None,
+                 max_kv_blocks: Optional[int] = None,
+                 engine_kwargs: Optional[Dict[str, Any]] = None):
+        self.name = name
+        self.itype = InstanceType(itype)
+        self.model = model
+        self.device = device
+        self.registry_addr = (registry_host, registry_port)
+        self.rpc_host = rpc_host
+        self.rpc_port = rpc_port
+        self.heartbeat_s = heartbeat_s
+        self.lease_ttl_s = lease_ttl_s
+        self.eos_token_id = eos_token_id
+        self.max_kv_blocks = max_kv_blocks
+        self.engine_kwargs = engine_kwargs or {}
+        self.incarnation = int(time.time() * 1000)
+
+        self.engine: Optional[LLMEngine] = None
+        self.registry: Optional[RegistryClient] = None
+        self.master_conn: Optional[msgrpc.Connection] = None
+        self.rpc_server: Optional[msgrpc.Server] = None
+        self.peers: Dict[str, InstanceMetaInfo] = {}
+        self.peer_conns: Dict[str, msgrpc.Connection] = {}
+        # requests this PREFILL instance must migrate after the first token:
+        # rid -> dict(routing/params)
+        self.pending_migration: Dict[str, Dict[str, Any]] = {}
+        # remember request params for usage accounting
+        self.req_meta: Dict[str, Dict[str, Any]] = {}
+
+        self._cmd_q: "queue.Queue" = queue.Queue()
+        self._out_q: "queue.Queue" = queue.Queue()
+        self._stop = threading.Event()
+        self._engine_thread: Optional[threading.Thread] = None
+        self._tasks: List[asyncio.Task] = []
+        self._loop: Optional[asyncio.AbstractEventLoop] = None
+        self._lease_id: Optional[int] = None
+        self._ttft_samples: List[float] = []
+        self._tbt_samples: List[float] = []
+
+    # ------------------------------------------------------------------ setup
+    async def start(self):
+        self._loop = asyncio.get_running_loop()
+        # engine init is slow (weights); do it off-loop
+        self.engine = await self._loop.run_in_executor(None, self._make_engine)
+        self._engine_thread = threading.Thread(target=self._engine_loop,
+                                               daemon=True,
+                                               name=f"engine-{self.name}")
+        self._engine_thread.start()
+
+        self.rpc_server = msgrpc.Server(lambda conn: self, self.rpc_host,
+                                        self.rpc_port)
+        self.rpc_port = await self.rpc_server.start()
+
+        self.registry = await RegistryClient().connect(*self.registry_addr)
+        self._lease_id = await self.registry.grant_lease(self.lease_ttl_s)
+        await self.registry.put_json(self._regkey(), self.meta().to_dict(),
+                                     lease_id=self._lease_id)
+        await self._connect_master()
+        self._tasks = [
+            asyncio.create_task(self._keepalive_loop()),
+            asyncio.create_task(self._heartbeat_loop()),
+            asyncio.create_task(self._push_loop()),
+        ]
+        log.info("worker %s (%s) up: rpc=%s:%d", self.name, self.itype.value,
+                 self.rpc_host, self.rpc_port)
+
+    def _make_engine(self) -> LLMEngine:
+        eng = LLMEngine(self.model, device=self.device,
+                        max_kv_blocks=self.max_kv_blocks,
+                        **self.engine_kwargs)
+        eng.eos_token_id = self.eos_token_id
+        return eng
+
+    def _regkey(self) -> str:
+        return KEY_INSTANCE[self.itype] + self.name
+
+    def meta(self) -> InstanceMetaInfo:
+        dev = -1
+        if self.engine is not None and self.engine.device.type == "cuda":
+            dev = self.engine.device.index or 0
+        return InstanceMetaInfo(
+            name=self.name, itype=self.itype.value,
+            rpc_host=self.rpc_host, rpc_port=self.rpc_port,
+            device_index=dev,
+            cluster_ids=[dev] if dev >= 0 else [],
+            num_kv_blocks=self.engine.block_manager.num_blocks
+            if self.engine else 0,
+            block_size=16, model=self.model,
+            incarnation_id=self.incarnation,
+            k_cache_ids=list(range(self.engine.cfg.num_layers))
+            if self.engine else [],
+            v_cache_ids=list(range(self.engine.cfg.num_layers))
+            if self.engine else [],
+        )
+
+    async def _connect_master(self):
+        async def on_master_change(ev):
+            if ev.type == "put":
+                await self._dial_master()
+
+        await self.registry.watch(KEY_MASTER, on_master_change)
+        await self._dial_master()
+
+    async def _dial_master(self):
+        info = await self.registry.get_json(KEY_MASTER)
+        if not info:
+            return
+        host, port = info["rpc_host"], info["rpc_port"]
+        if self.master_conn and not self.master_conn.closed.is_set():
+            return
+        try:
+            self.master_conn = await msgrpc.connect(host, port, handler=self)
+            await self.master_conn.call("hello", name=self.name)
+        except OSError as e:
+            log.warning("worker %s: master dial failed: %s", self.name, e)
+
+    async def stop(self):
+        self._stop.set()
+        for t in self._tasks:
+            t.cancel()
+        if self._engine_thread:
+            self._engine_thread.join(timeout=5)
+        if self.rpc_server:
+            await self.rpc_server.stop()
+        for c in self.peer_conns.values():
+            await c.close()
+        if self.master_conn:
+            await self.master_conn.close()
+        if self.registry:
+            if self._lease_id:
+                try:
+                    await self.registry.revoke_lease(self._lease_id)
+                except Exception:
+                    pass
+            await self.registry.close()
+
+    # ---------------------------------------------------------- engine thread
+    def _engine_loop(self):
+        while not self._stop.is_set():
+            did = False
+            try:
+                while True:
+                    fn, fut = self._cmd_q.get_nowait()
+                    try:
+                        res = fn()
+                        if fut:
+                            fut.set_result(res)
+                    except Exception as e:  # noqa: BLE001
+                        if fut:
+                            fut.set_exception(e)
+                        else:
+                            log.exception("engine cmd failed")
+                    did = True
+            except queue.Empty:
+                pass
+            if self.engine.has_work():
+                outs = self.engine.step()
+                if outs:
+                    self._out_q.put(outs)
+                did = True
+            if not did:
+                time.sleep(0.002)
+
+    async def _run_on_engine(self, fn):
+        fut = concurrent.futures.Future()
+        self._cmd_q.put((fn, fut))
+        return await asyncio.wrap_future(fut)
+
+    def _post_to_engine(self, fn):
+        self._cmd_q.put((fn, None))
+
+    # --------------------------------------------------------------- RPC API
+    def rpc_health(self, conn) -> bool:
+        return True
+
+    def rpc_get_info(self, conn) -> dict:
+        return self.meta().to_dict()
+
+    async def rpc_link_instance(self, conn, peer: dict) -> bool:
+        meta = InstanceMetaInfo.from_dict(peer)
+        self.peers[meta.name] = meta
+        log.info("worker %s: linked peer %s (%s)", self.name, meta.name,
+                 meta.itype)
+        return True
+
+    async def rpc_unlink_instance(self, conn, peer_name: str) -> bool:
+        self.peers.pop(peer_name, None)
+        c = self.peer_conns.pop(peer_name, None)
+        if c:
+            await c.close()
+        return True
+
+    async def _peer_conn(self, name: str) -> Optional[msgrpc.Connection]:
+        c = self.peer_conns.get(name)
+        if c and not c.closed.is_set():
+            return c
+        meta = self.peers.get(name)
+        if meta is None:
+            return None
+        try:
+            c = await msgrpc.connect(meta.rpc_host, meta.rpc_port)
+            self.peer_conns[name] = c
+            return c
+        except OSError:
+            return None
+
+    # execute_request arrives as a notification (fire-and-forget dispatch)
+    async def on_execute_request(self, conn, service_request_id: str,
+                                 token_ids: List[int], params: dict,
+                                 routing: dict, offline: bool = False,
+                                 multimodal: Optional[dict] = None):
+        routing = routing or {}
+        sp = params_from_dict(params)
+        self.req_meta[service_request_id] = dict(
+            params=params, routing=routing, prompt_len=len(token_ids),
+            offline=offline, multimodal=multimodal)
+        decode_name = routing.get("decode_name")
+        do_migrate = (self.itype == InstanceType.PREFILL
+                      and decode_name and decode_name != self.name)
+        if do_migrate:
+            # prefill role: produce exactly the first token, hold blocks
+            first_sp = params_from_dict(params)
+            first_sp.max_tokens = 1
+            first_sp.ignore_eos = True
+            self.pending_migration[service_request_id] = dict(
+                routing=routing, params=params, token_ids=list(token_ids),
+                offline=offline)
+            self._post_to_engine(
+                lambda: self.engine.add_request(
+                    service_request_id, token_ids, first_sp,
+                    priority=1 if offline else 0, hold_blocks=True))
+        else:
+            self._post_to_engine(
+                lambda: self.engine.add_request(
+                    service_request_id, token_ids, sp,
+                    priority=1 if offline else 0))
+
+    def on_abort_request(self, conn, service_request_id: str):
+        self.pending_migration.pop(service_request_id, None)
+        self._post_to_engine(
+            lambda: self.engine.abort_request(service_request_id))
+
+    # decode side of a migration (called by the prefill worker)
+    async def rpc_migrate_in(self, conn, service_request_id: str,
+                             prompt_token_ids: List[int],
+                             first_token_ids: List[int], params: dict,
+                             n_blocks: int, transport: str,
+                             data: Optional[bytes] = None,
+                             src_name: Optional[str] = None,
+                             src_blocks: Optional[List[int]] = None,
+                             offline: bool = False) -> bool:
+        sp = params_from_dict(params)
+        # account the token(s) the prefill already produced
+        sp.max_tokens = max(sp.max_tokens, 1)
+        meta = self.req_meta.setdefault(service_request_id, {})
+        meta.update(prompt_len=len(prompt_token_ids), params=params)
+
+        def _recv():
+            blocks = self.engine.alloc_migration_blocks(n_blocks)
+            try:
+                if transport == "bytes":
+                    self.engine.import_block_bytes(blocks, data)
+                elif transport == "xgmi":
+                    from .kv_migration import migrate_in_xgmi
+                    migrate_in_xgmi(self.engine, self.peers.get(src_name),
+                                    src_blocks, blocks)
+                else:
+                    raise ValueError(f"unknown transport {transport}")
+            except Exception:
+                self.engine.free_blocks(blocks)
+                raise
+            self.engine.activate_migrated_request(
+                service_request_id, prompt_token_ids, first_token_ids,
+                blocks, sp, priority=1 if offline else 0)
+
+        await self._run_on_engine(_recv)
+        return True
+
+    # ------------------------------------------------------------- push loop
+    async def _push_loop(self):
+        while not self._stop.is_set():
+            outs = await self._loop.run_in_executor(None, self._out_q_get)
+            if outs is None:
+                continue
+            gens = []
+            migrations = []
+            for o in outs:
+                rid = o.request_id
+                mig = self.pending_migration.get(rid)
+                if mig is not None and o.finished:
+                    # first token produced by prefill: announce + migrate
+                    gens.append(dict(
+                        service_request_id=rid,
+                        token_ids=o.new_token_ids,
+                        finished=False, finished_on_prefill=True,
+                        prompt_tokens=o.num_prompt_tokens,
+                        completion_tokens=o.num_output_tokens))
+                    migrations.append((rid, self.pending_migration.pop(rid),
+                                       o.new_token_ids))
+                    continue
+                gens.append(dict(
+                    service_request_id=rid,
+                    token_ids=o.new_token_ids,
+                    finished=o.finished,
+                    finish_reason=o.finish_reason,
+                    finished_on_prefill=(
+                        o.first_token and
+                        self.itype in (InstanceType.DEFAULT, InstanceType.MIX,
+                                       InstanceType.PREFILL)),
+                    prompt_tokens=o.num_prompt_tokens,
+                    completion_tokens=o.num_output_tokens))
+                if o.finished:
+                    self.req_meta.pop(rid, None)
+            if gens and self.master_conn and not self.master_conn.closed.is_set():
+                try:
+                    await self.master_conn.notify("generations", gens=gens)
+                except Exception:
+                    log.warning("worker %s: generations push failed", self.name)
+            for rid, mig, first_toks in migrations:
+                asyncio.create_task(self._do_migration(rid, mig, first_toks))
+
+    def _out_q_get(self):
+        try:
+            return self._out_q.get(timeout=0.1)
+        except queue.Empty:
+            return None
+
+    async def _do_migration(self, rid: str, mig: dict, first_toks: List[int]):
+        decode_name = mig["routing"]["decode_name"]
+        conn = await self._peer_conn(decode_name)
+        try:
+            if conn is None:
+                raise RuntimeError(f"decode peer {decode_name} unreachable")
+            blocks = await self._run_on_engine(
+                lambda: self.engine.held_block_table(rid))
+            peer = self.peers.get(decode_name)
+            transport = self._pick_transport(peer)
+            kwargs: Dict[str, Any] = dict(
+                service_request_id=rid,
+                prompt_token_ids=mig["token_ids"],
+                first_token_ids=first_toks,
+                params=mig["params"], n_blocks=len(blocks),
+                transport=transport, src_name=self.name,
+                offline=mig.get("offline", False))
+            if transport == "bytes":
+                kwargs["data"] = await self._run_on_engine(
+                    lambda: self.engine.export_block_bytes(blocks))
+            else:
+                kwargs["src_blocks"] = blocks
+            await conn.call("migrate_in", timeout=60.0, **kwargs)
+        except Exception as e:
+            log.warning("migration of %s to %s failed: %s", rid, decode_name, e)
+            # tell the master the request died (client will retry)
+            if self.master_conn:
+                try:
+                    await self.master_conn.notify("generations", gens=[dict(
+                        service_request_id=rid, token_ids=[], finished=True,
+                        finish_reason="abort",
+                        error=f"kv migration failed: {e}")])
+                except Exception:
+                    pass
+        finally:
+            self._post_to_engine(lambda: self.engine.release_held(rid))
+
+    def _pick_transport(self, peer: Optional[InstanceMetaInfo]) -> str:
+        if (peer is not None and peer.device_index >= 0
+                and self.engine.device.type == "cuda"):
+            try:
+                from . import kv_migration
+                if kv_migration.available():
+                    return "xgmi"
+            except Exception:
+                pass
+        return "bytes"
+
+    # ---------------------------------------------------------- housekeeping
+    async def _keepalive_loop(self):
+        while not self._stop.is_set():
+            await asyncio.sleep(self.lease_ttl_s / 3)
+            try:
+                ok = await self.registry.keepalive(self._lease_id)
+                if not ok:  # lease expired (e.g. long GC pause): re-register
+                    self._lease_id = await self.registry.grant_lease(
+                        self.lease_ttl_s)
+                    await self.registry.put_json(self._regkey(),
+                                                 self.meta().to_dict(),
+                                                 lease_id=self._lease_id)
+            except Exception:
+                pass
+
+    async def _heartbeat_loop(self):
+        while not self._stop.is_set():
+            await asyncio.sleep(self.heartbeat_s)
+            if self.master_conn is None or self.master_conn.closed.is_set():
+                await self._dial_master()
+                if self.master_conn is None or self.master_conn.closed.is_set():
+                    continue
+            try:
+                ev = self.engine.block_manager.events.drain()
+                st = self.engine.stats
+                await self.master_conn.call(
+                    "heartbeat", timeout=5.0,
+                    name=self.name, incarnation=self.incarnation,
+                    load=dict(
+                        waiting_requests_num=st.num_waiting,
+                        running_requests_num=st.num_running,
+                        gpu_cache_usage_perc=st.kv_usage),
+                    latency=dict(
+                        recent_max_ttft_ms=max(self._ttft_samples or [0.0]),
+                        recent_max_tbt_ms=max(self._tbt_samples or [0.0])),
+                    kv_stored=[bytes(h) for h in ev.stored],
+                    kv_removed=[bytes(h) for h in ev.removed])
+                self._ttft_samples.clear()
+                self._tbt_samples.clear()
+            except Exception:
+                pass

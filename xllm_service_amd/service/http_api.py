@@ -1,0 +1,206 @@
+"""OpenAI-compatible HTTP front end (reference: http_service/).
+
+Routes: /hello, /v1/completions, /v1/chat/completions, /v1/models,
+/v1/embeddings (501, like the reference), /metrics. Readiness gating: when
+no viable instance group exists the API answers 503 (the reference stops
+its HTTP listener; answering 503 is the same contract to a load balancer).
+"""
+from __future__ import annotations
+
+import logging
+from typing import Any, Dict, List, Optional, Union
+
+from fastapi import FastAPI, Request
+from fastapi.responses import JSONResponse, Response, StreamingResponse
+from pydantic import BaseModel, Field
+
+from . import metrics
+from .request import ServiceRequest, make_request_id
+from .scheduler import SchedulerError, ServiceScheduler
+
+log = logging.getLogger("xllm.http")
+
+
+class CompletionBody(BaseModel):
+    model: str = ""
+    prompt: Union[str, List[int], List[str], None] = ""
+    max_tokens: int = 16
+    temperature: float = 1.0
+    top_p: float = 1.0
+    top_k: int = -1
+    n: int = 1
+    stream: bool = False
+    stream_options: Optional[Dict[str, Any]] = None
+    stop: Union[str, List[str], None] = None
+    seed: Optional[int] = None
+    logprobs: Optional[int] = None
+    echo: bool = False
+    ignore_eos: bool = False
+    offline: bool = False
+    min_tokens: int = 0
+
+
+class ChatBody(BaseModel):
+    model: str = ""
+    messages: List[Dict[str, Any]] = Field(default_factory=list)
+    tools: Optional[List[Dict[str, Any]]] = None
+    tool_choice: Union[str, Dict[str, Any], None] = None
+    chat_template_kwargs: Optional[Dict[str, Any]] = None
+    max_tokens: Optional[int] = None
+    max_completion_tokens: Optional[int] = None
+    temperature: float = 1.0
+    top_p: float = 1.0
+    top_k: int = -1
+    n: int = 1
+    stream: bool = False
+    stream_options: Optional[Dict[str, Any]] = None
+    stop: Union[str, List[str], None] = None
+    seed: Optional[int] = None
+    ignore_eos: bool = False
+    offline: bool = False
+    min_tokens: int = 0
+
+
+def build_app(master) -> FastAPI:
+    """master: service.master.Master (owns scheduler + response handler)."""
+    app = FastAPI(title="xllm-service-amd", docs_url=None, redoc_url=None)
+
+    def scheduler() -> ServiceScheduler:
+        return master.scheduler
+
+    @app.get("/hello")
+    @app.post("/hello")
+    async def hello(request: Request):
+        try:
+            body = await request.json()
+        except Exception:
+            body = None
+        return {"message": "hello", "echo": body}
+
+    @app.get("/health")
+    async def health():
+        return {"status": "ok"}
+
+    @app.get("/metrics")
+    async def metrics_route():
+        return Response(content=metrics.render(),
+                        media_type="text/plain; version=0.0.4")
+
+    @app.get("/v1/models")
+    async def models():
+        return {"object": "list",
+                "data": [{"id": m, "object": "model",
+                          "owned_by": "xllm-service-amd"}
+                         for m in master.served_models()]}
+
+    @app.post("/v1/embeddings")
+    async def embeddings():
+        return JSONResponse({"error": {"message": "not support embeddings",
+                                       "type": "invalid_request_error"}},
+                            status_code=501)
+
+    def _not_ready() -> Optional[JSONResponse]:
+        if not scheduler().has_available_instances():
+            metrics.REQUEST_ERROR_TOTAL.labels(
+                method="any", reason="no_instances").inc()
+            return JSONResponse(
+                {"error": {"message": "no available instances",
+                           "type": "server_error"}}, status_code=503)
+        return None
+
+    @app.post("/v1/completions")
+    async def completions(body: CompletionBody, request: Request):
+        metrics.REQUEST_IN_TOTAL.labels(method="completion").inc()
+        na = _not_ready()
+        if na is not None:
+            return na
+        if isinstance(body.prompt, list) and body.prompt \
+                and isinstance(body.prompt[0], int):
+            token_ids = list(body.prompt)
+            prompt_text = ""
+        else:
+            prompt_text = body.prompt if isinstance(body.prompt, str) \
+                else "".join(body.prompt or [])
+            token_ids = scheduler().tokenizer.encode(prompt_text)
+        req = ServiceRequest(
+            service_request_id=make_request_id("cmpl"),
+            kind="completion", model=body.model or master.model_id,
+            stream=body.stream, token_ids=token_ids, prompt_text=prompt_text,
+            params=_sampling_dict(body), offline=body.offline)
+        return await _run(req, request, body.stream, chat=False)
+
+    @app.post("/v1/chat/completions")
+    async def chat_completions(body: ChatBody, request: Request):
+        metrics.REQUEST_IN_TOTAL.labels(method="chat").inc()
+        na = _not_ready()
+        if na is not None:
+            return na
+        try:
+            prompt_text, token_ids = scheduler().tokenize_chat(
+                body.messages, tools=body.tools, tool_choice=body.tool_choice,
+                chat_template_kwargs=body.chat_template_kwargs)
+        except Exception as e:
+            return JSONResponse(
+                {"error": {"message": f"chat template error: {e}",
+                           "type": "invalid_request_error"}}, status_code=400)
+        params = _sampling_dict(body)
+        if body.max_completion_tokens is not None:
+            params["max_tokens"] = body.max_completion_tokens
+        elif body.max_tokens is None:
+            params["max_tokens"] = 512
+        req = ServiceRequest(
+            service_request_id=make_request_id("chatcmpl"),
+            kind="chat", model=body.model or master.model_id,
+            stream=body.stream, token_ids=token_ids, prompt_text=prompt_text,
+            params=params, offline=body.offline)
+        return await _run(req, request, body.stream, chat=True)
+
+    def _sampling_dict(body) -> Dict[str, Any]:
+        d = dict(temperature=body.temperature, top_p=body.top_p,
+                 top_k=body.top_k, max_tokens=body.max_tokens or 16,
+                 min_tokens=body.min_tokens, seed=body.seed,
+                 ignore_eos=body.ignore_eos)
+        if body.stream_options:
+            d["include_usage"] = bool(
+                body.stream_options.get("include_usage"))
+        stops = body.stop if isinstance(body.stop, list) else (
+            [body.stop] if body.stop else [])
+        stop_ids = []
+        for s in stops:
+            ids = scheduler().tokenizer.encode(s)
+            if len(ids) == 1:
+                stop_ids.append(ids[0])
+        d["stop_token_ids"] = stop_ids
+        return d
+
+    async def _run(req: ServiceRequest, http_request: Request, stream: bool,
+                   chat: bool):
+        sch = scheduler()
+        req.is_disconnected = lambda: False  # updated below for streams
+        master.tracer.trace(req.service_request_id, "request_in",
+                            {"kind": req.kind, "tokens": len(req.token_ids)})
+        try:
+            sch.schedule(req)
+            await sch.dispatch(req)
+        except SchedulerError as e:
+            metrics.REQUEST_ERROR_TOTAL.labels(
+                method=req.kind, reason="schedule").inc()
+            return JSONResponse({"error": {"message": str(e),
+                                           "type": "server_error"}},
+                                status_code=e.status_code)
+
+        async def on_cancel(r):
+            await sch.cancel_request(r, reason="client disconnected")
+
+        rh = master.response_handler
+        if stream:
+            gen = (rh.stream_chat(req, on_cancel) if chat
+                   else rh.stream_completion(req, on_cancel))
+            return StreamingResponse(gen, media_type="text/event-stream")
+        result = await (rh.collect_chat(req) if chat
+                        else rh.collect_completion(req))
+        status = 500 if "error" in result else 200
+        master.tracer.trace(req.service_request_id, "response_out", result)
+        return JSONResponse(result, status_code=status)
+
+    return app

@@ -1,0 +1,402 @@
+"""InstanceMgr: cluster membership, failure detection, link fan-out,
+scheduling primitives.
+
+Re-design of the reference's largest component (instance_mgr.cpp, 1678 LoC;
+SURVEY.md 2.5) on asyncio + the embedded registry:
+
+  * registration is watch-driven: instances PUT XLLM:<TYPE>:<name> with a
+    TTL lease; the manager bootstraps with a prefix scan then follows
+    PUT/DELETE events
+  * failure state machine ACTIVE -> LEASE_LOST -> SUSPECT -> deregistered:
+    lease DELETE triggers a health probe (grace on success), a reconcile
+    task demotes silent LEASE_LOST instances and evicts old SUSPECTs,
+    heartbeats restore
+  * incarnation ids make restart-with-the-same-name safe: stale events for
+    an older incarnation are ignored; a new incarnation replaces the old
+  * P<->D link fan-out on (de)registration: link_instance/unlink_instance
+    RPCs carry peer cluster info; on one 8xMI355X node this configures
+    hipDeviceEnablePeerAccess + block-table registration for xGMI KV
+    migration (the reference's device-network LinkInstance degenerates to
+    this — SURVEY.md 5.8)
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import logging
+import time
+from dataclasses import dataclass, field
+from typing import Any, Callable, Dict, List, Optional, Tuple
+
+from xllm_service_amd.registry.server import RegistryClient
+from xllm_service_amd.registry.store import WatchEvent
+from xllm_service_amd.utils import msgrpc
+
+from .types import (KEY_INSTANCE, KEY_LOADMETRICS, InstanceMetaInfo,
+                    InstanceStatus, InstanceType, LatencyMetrics, LoadMetrics)
+
+log = logging.getLogger("xllm.instance_mgr")
+
+
+@dataclass
+class Instance:
+    meta: InstanceMetaInfo
+    status: InstanceStatus = InstanceStatus.ACTIVE
+    conn: Optional[msgrpc.Connection] = None
+    load: LoadMetrics = field(default_factory=LoadMetrics)
+    latency: LatencyMetrics = field(default_factory=LatencyMetrics)
+    last_heartbeat: float = field(default_factory=time.monotonic)
+    last_status_change: float = field(default_factory=time.monotonic)
+    # SLO bookkeeping
+    num_scheduled: int = 0
+    num_prefill_unfinished: int = 0
+    num_decoding: int = 0
+    pending_prefill_tokens: int = 0
+
+    @property
+    def name(self) -> str:
+        return self.meta.name
+
+    @property
+    def itype(self) -> InstanceType:
+        return InstanceType(self.meta.itype)
+
+    @property
+    def schedulable(self) -> bool:
+        return self.status != InstanceStatus.SUSPECT
+
+    def set_status(self, st: InstanceStatus):
+        if st != self.status:
+            self.status = st
+            self.last_status_change = time.monotonic()
+
+
+class InstanceMgr:
+    def __init__(self, registry: RegistryClient,
+                 on_instance_failed: Optional[Callable[[str, int], Any]] = None,
+                 probe_timeout_s: float = 1.0,
+                 probe_attempts: int = 2,
+                 lease_lost_heartbeat_timeout_s: float = 3.0,
+                 suspect_eviction_s: float = 15.0,
+                 is_master: Callable[[], bool] = lambda: True):
+        self.registry = registry
+        self.on_instance_failed = on_instance_failed
+        self.probe_timeout_s = probe_timeout_s
+        self.probe_attempts = probe_attempts
+        self.lease_lost_heartbeat_timeout_s = lease_lost_heartbeat_timeout_s
+        self.suspect_eviction_s = suspect_eviction_s
+        self.is_master = is_master
+
+        self.instances: Dict[str, Instance] = {}
+        self.prefill_index: List[str] = []   # PREFILL + DEFAULT (+ MIX overflow)
+        self.decode_index: List[str] = []    # DECODE (+ MIX)
+        self.encode_index: List[str] = []    # ENCODE (multimodal E-stage)
+        self._rr_pos = 0
+        self._reconcile_task: Optional[asyncio.Task] = None
+        self._lock = asyncio.Lock()
+        self._updated_loadmetrics: Dict[str, LoadMetrics] = {}
+
+    # ---- lifecycle ----------------------------------------------------------
+    async def start(self):
+        for t in InstanceType:
+            await self.registry.watch(KEY_INSTANCE[t], self._on_registry_event)
+        await self.registry.watch(KEY_LOADMETRICS, self._on_loadmetrics_event)
+        # bootstrap from existing keys
+        for t in InstanceType:
+            for key, value in await self.registry.range(KEY_INSTANCE[t]):
+                await self._register(InstanceMetaInfo.from_dict(
+                    json.loads(value)))
+        self._reconcile_task = asyncio.create_task(self._reconcile_loop())
+
+    async def stop(self):
+        if self._reconcile_task:
+            self._reconcile_task.cancel()
+        for inst in self.instances.values():
+            if inst.conn:
+                await inst.conn.close()
+
+    # ---- registry watch -----------------------------------------------------
+    async def _on_registry_event(self, ev: WatchEvent):
+        if ev.type == "put":
+            meta = InstanceMetaInfo.from_dict(json.loads(ev.value))
+            existing = self.instances.get(meta.name)
+            if existing is None:
+                await self._register(meta)
+            elif meta.incarnation_id > existing.meta.incarnation_id:
+                # instance replacement: new incarnation under the same name
+                await self._deregister(meta.name,
+                                       existing.meta.incarnation_id,
+                                       reason="replaced")
+                await self._register(meta)
+            elif meta.incarnation_id == existing.meta.incarnation_id:
+                # re-PUT refreshes a degraded instance back to ACTIVE
+                existing.set_status(InstanceStatus.ACTIVE)
+                existing.last_heartbeat = time.monotonic()
+            # stale older incarnation: ignore
+        elif ev.type == "delete":
+            prev = json.loads(ev.prev_value) if ev.prev_value else {}
+            name = prev.get("name") or ev.key.rsplit(":", 1)[-1]
+            inst = self.instances.get(name)
+            if inst is None:
+                return
+            if prev.get("incarnation_id", inst.meta.incarnation_id) < \
+                    inst.meta.incarnation_id:
+                return  # stale delete for an old incarnation
+            await self._probe_after_lease_loss(inst)
+
+    async def _on_loadmetrics_event(self, ev: WatchEvent):
+        # non-master replicas learn load metrics via registry watch
+        if ev.type != "put" or self.is_master():
+            return
+        name = ev.key[len(KEY_LOADMETRICS):]
+        inst = self.instances.get(name)
+        if inst:
+            inst.load = LoadMetrics.from_dict(json.loads(ev.value))
+
+    # ---- register / deregister ----------------------------------------------
+    async def _register(self, meta: InstanceMetaInfo):
+        async with self._lock:
+            if meta.name in self.instances:
+                return
+            inst = Instance(meta)
+            try:
+                inst.conn = await msgrpc.connect(meta.rpc_host, meta.rpc_port)
+            except OSError as e:
+                log.warning("register %s: cannot connect rpc (%s)", meta.name, e)
+                inst.conn = None
+            self.instances[meta.name] = inst
+            # link fan-out: P<->D peers exchange cluster info
+            links: List[Tuple[Instance, InstanceMetaInfo]] = []
+            it = inst.itype
+            if it in (InstanceType.PREFILL, InstanceType.MIX,
+                      InstanceType.ENCODE):
+                for other in self.instances.values():
+                    if other.name != meta.name and other.itype in (
+                            InstanceType.DECODE, InstanceType.MIX,
+                            InstanceType.PREFILL):
+                        links.append((other, meta))
+                        links.append((inst, other.meta))
+            elif it == InstanceType.DECODE:
+                for other in self.instances.values():
+                    if other.name != meta.name and other.itype in (
+                            InstanceType.PREFILL, InstanceType.MIX):
+                        links.append((inst, other.meta))
+                        links.append((other, meta))
+            done: List[Tuple[Instance, InstanceMetaInfo]] = []
+            ok = True
+            for target, peer in links:
+                if await self._call_link(target, peer):
+                    done.append((target, peer))
+                else:
+                    ok = False
+                    break
+            if not ok:  # roll back partial links
+                for target, peer in done:
+                    await self._call_unlink(target, peer.name)
+                log.warning("register %s: link fan-out failed, rolled back",
+                            meta.name)
+            self._add_to_index(inst)
+            log.info("registered instance %s type=%s inc=%d", meta.name,
+                     meta.itype, meta.incarnation_id)
+
+    def _add_to_index(self, inst: Instance):
+        it = inst.itype
+        if it in (InstanceType.DEFAULT, InstanceType.PREFILL):
+            if inst.name not in self.prefill_index:
+                self.prefill_index.append(inst.name)
+        elif it == InstanceType.DECODE:
+            if inst.name not in self.decode_index:
+                self.decode_index.append(inst.name)
+        elif it == InstanceType.ENCODE:
+            if inst.name not in self.encode_index:
+                self.encode_index.append(inst.name)
+        elif it == InstanceType.MIX:
+            # MIX joins whichever side is empty (reference behaviour)
+            if not self.prefill_index:
+                self.prefill_index.append(inst.name)
+            else:
+                self.decode_index.append(inst.name)
+
+    def _remove_from_index(self, name: str):
+        for idx in (self.prefill_index, self.decode_index, self.encode_index):
+            if name in idx:
+                idx.remove(name)
+
+    async def _deregister(self, name: str, incarnation: int, reason: str):
+        async with self._lock:
+            inst = self.instances.get(name)
+            if inst is None or inst.meta.incarnation_id != incarnation:
+                return
+            # unlink fan-out
+            for other in self.instances.values():
+                if other.name == name:
+                    continue
+                await self._call_unlink(other, name)
+            self._remove_from_index(name)
+            del self.instances[name]
+            self._updated_loadmetrics.pop(name, None)
+        log.warning("deregistered instance %s (%s)", name, reason)
+        if self.on_instance_failed:
+            r = self.on_instance_failed(name, incarnation)
+            if asyncio.iscoroutine(r):
+                await r
+        if inst.conn:
+            await inst.conn.close()
+
+    async def _call_link(self, target: Instance, peer: InstanceMetaInfo) -> bool:
+        if target.conn is None:
+            return True  # no rpc channel (e.g. registry-only test instance)
+        try:
+            await target.conn.call("link_instance", timeout=5.0,
+                                   peer=peer.to_dict())
+            return True
+        except Exception as e:
+            log.warning("link_instance(%s <- %s) failed: %s", target.name,
+                        peer.name, e)
+            return False
+
+    async def _call_unlink(self, target: Instance, peer_name: str):
+        if target.conn is None:
+            return
+        try:
+            await target.conn.call("unlink_instance", timeout=5.0,
+                                   peer_name=peer_name)
+        except Exception:
+            pass
+
+    # ---- failure detection --------------------------------------------------
+    async def _probe_after_lease_loss(self, inst: Instance):
+        ok = False
+        for _ in range(self.probe_attempts):
+            if await self._probe_health(inst):
+                ok = True
+                break
+        if ok:
+            inst.set_status(InstanceStatus.LEASE_LOST)   # grace: schedulable
+            log.info("instance %s lease lost but healthy (grace)", inst.name)
+        else:
+            inst.set_status(InstanceStatus.SUSPECT)
+            log.warning("instance %s lease lost and probe failed -> SUSPECT",
+                        inst.name)
+
+    async def _probe_health(self, inst: Instance) -> bool:
+        if inst.conn is None or inst.conn.closed.is_set():
+            return False
+        try:
+            r = await inst.conn.call("health", timeout=self.probe_timeout_s)
+            return bool(r)
+        except Exception:
+            return False
+
+    async def _reconcile_loop(self):
+        while True:
+            await asyncio.sleep(1.0)
+            try:
+                await self._reconcile_once()
+            except asyncio.CancelledError:
+                raise
+            except Exception:
+                log.exception("reconcile error")
+
+    async def _reconcile_once(self):
+        now = time.monotonic()
+        for inst in list(self.instances.values()):
+            if inst.status == InstanceStatus.LEASE_LOST:
+                if now - inst.last_heartbeat > self.lease_lost_heartbeat_timeout_s:
+                    inst.set_status(InstanceStatus.SUSPECT)
+                    log.warning("instance %s silent in LEASE_LOST -> SUSPECT",
+                                inst.name)
+            elif inst.status == InstanceStatus.SUSPECT:
+                if now - inst.last_status_change > self.suspect_eviction_s:
+                    await self._deregister(inst.name,
+                                           inst.meta.incarnation_id,
+                                           reason="suspect eviction")
+
+    # ---- heartbeats ---------------------------------------------------------
+    def record_heartbeat(self, name: str, incarnation: int,
+                         load: Optional[dict] = None,
+                         latency: Optional[dict] = None) -> bool:
+        inst = self.instances.get(name)
+        if inst is None or incarnation != inst.meta.incarnation_id:
+            return False
+        inst.last_heartbeat = time.monotonic()
+        if inst.status == InstanceStatus.SUSPECT:
+            inst.set_status(InstanceStatus.LEASE_LOST)
+        if load is not None:
+            inst.load = LoadMetrics.from_dict(load)
+            self._updated_loadmetrics[name] = inst.load
+        if latency is not None:
+            inst.latency = LatencyMetrics.from_dict(latency)
+        return True
+
+    async def upload_load_metrics(self):
+        """Master-only: publish dirty load metrics so replicas share the
+        cluster view (reference: upload_load_metrics, 3 s cadence)."""
+        dirty, self._updated_loadmetrics = self._updated_loadmetrics, {}
+        for name, lm in dirty.items():
+            await self.registry.put_json(KEY_LOADMETRICS + name, lm.to_dict())
+
+    # ---- scheduling primitives ----------------------------------------------
+    def get(self, name: str) -> Optional[Instance]:
+        return self.instances.get(name)
+
+    def _schedulable(self, names: List[str]) -> List[Instance]:
+        return [self.instances[n] for n in names
+                if n in self.instances and self.instances[n].schedulable]
+
+    def schedulable_prefills(self) -> List[Instance]:
+        return self._schedulable(self.prefill_index)
+
+    def schedulable_decodes(self) -> List[Instance]:
+        return self._schedulable(self.decode_index)
+
+    def schedulable_encodes(self) -> List[Instance]:
+        return self._schedulable(self.encode_index)
+
+    def has_available_instances(self) -> bool:
+        """A viable serving group exists (reference: has_available_instances).
+        DEFAULT alone suffices; otherwise a prefill+decode pair is needed."""
+        prefills = self.schedulable_prefills()
+        decodes = self.schedulable_decodes()
+        for p in prefills:
+            if p.itype == InstanceType.DEFAULT:
+                return True
+        return bool(prefills and decodes)
+
+    def next_rr_pair(self) -> Tuple[Optional[Instance], Optional[Instance]]:
+        """Round-robin (prefill, decode) pair skipping unschedulable
+        instances; DEFAULT instances may serve alone."""
+        prefills = self.schedulable_prefills()
+        if not prefills:
+            return None, None
+        self._rr_pos = (self._rr_pos + 1) % len(prefills)
+        prefill = prefills[self._rr_pos]
+        decodes = self.schedulable_decodes()
+        if not decodes:
+            if prefill.itype == InstanceType.DEFAULT:
+                return prefill, None
+            return None, None
+        decode = decodes[self._rr_pos % len(decodes)]
+        return prefill, decode
+
+    # ---- role flipping (SLO-aware adaptive P:D) ------------------------------
+    def flip_instance_role(self, name: str, new_side: str) -> bool:
+        """Move an instance between the prefill and decode scheduling sides
+        (MIX instances only are eligible; memory re-partitioning is handled
+        worker-side on its next idle point)."""
+        inst = self.instances.get(name)
+        if inst is None:
+            return False
+        if new_side == "decode" and name in self.prefill_index:
+            if len(self.prefill_index) <= 1:
+                return False
+            self.prefill_index.remove(name)
+            self.decode_index.append(name)
+            return True
+        if new_side == "prefill" and name in self.decode_index:
+            if len(self.decode_index) <= 1:
+                return False
+            self.decode_index.remove(name)
+            self.prefill_index.append(name)
+            return True
+        return False

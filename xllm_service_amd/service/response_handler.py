@@ -1,0 +1,210 @@
+"""ResponseHandler: OpenAI wire serialization for the four delivery paths
+(chat/completion x stream/non-stream), with reasoning + tool-call parsing.
+(reference: scheduler/response_handler.{h,cpp})
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+from typing import AsyncIterator, Optional
+
+from xllm_service_amd.tokenizer import IncrementalDecoder, Tokenizer
+
+from .parsers import make_parsers, make_stream_parsers
+from .request import GenerationDelta, ServiceRequest
+
+STREAM_TIMEOUT_S = 600.0
+
+
+def _sse(obj) -> str:
+    return f"data: {json.dumps(obj, ensure_ascii=False)}\n\n"
+
+
+class ResponseHandler:
+    def __init__(self, tokenizer: Tokenizer, parser_mode: str = "auto"):
+        self.tokenizer = tokenizer
+        self.parser_mode = parser_mode
+
+    # ------------------------------------------------------------ chat stream
+    async def stream_chat(self, req: ServiceRequest,
+                          on_cancel) -> AsyncIterator[str]:
+        rid = req.service_request_id
+        created = int(req.created)
+        model = req.model
+        include_usage = bool(req.params.get("include_usage"))
+        dec = IncrementalDecoder(self.tokenizer)
+        rp, tp = make_stream_parsers(model, self.parser_mode)
+        sent_role = False
+        tool_idx = 0
+        emitted_tool = False
+        usage = None
+
+        def chunk(delta: dict, finish: Optional[str] = None):
+            return _sse({
+                "id": rid, "object": "chat.completion.chunk",
+                "created": created, "model": model,
+                "choices": [{"index": 0, "delta": delta,
+                             "finish_reason": finish}]})
+
+        try:
+            while True:
+                gen: GenerationDelta = await asyncio.wait_for(
+                    req.output_queue.get(), STREAM_TIMEOUT_S)
+                if gen.error:
+                    yield _sse({"error": {"message": gen.error,
+                                          "type": "server_error"}})
+                    break
+                text = dec.push(gen.token_ids) if gen.token_ids else ""
+                if not sent_role and (text or gen.finished):
+                    yield chunk({"role": "assistant", "content": ""})
+                    sent_role = True
+                reasoning_delta, content = (None, text)
+                if rp is not None and text:
+                    reasoning_delta, content = rp.feed(text)
+                    if reasoning_delta:
+                        yield chunk({"reasoning_content": reasoning_delta})
+                if tp is not None and content:
+                    content, calls = tp.feed(content)
+                    for tc in calls:
+                        yield chunk({"tool_calls": [{
+                            "index": tool_idx, "id": tc.id, "type": "function",
+                            "function": {"name": tc.name,
+                                         "arguments": tc.arguments}}]})
+                        tool_idx += 1
+                        emitted_tool = True
+                if content:
+                    yield chunk({"content": content})
+                if gen.finished:
+                    if tp is not None:
+                        for tc in tp.flush():
+                            yield chunk({"tool_calls": [{
+                                "index": tool_idx, "id": tc.id,
+                                "type": "function",
+                                "function": {"name": tc.name,
+                                             "arguments": tc.arguments}}]})
+                            tool_idx += 1
+                            emitted_tool = True
+                    finish = gen.finish_reason or "stop"
+                    if emitted_tool and finish == "stop":
+                        finish = "tool_calls"
+                    yield chunk({}, finish=finish)
+                    usage = {
+                        "prompt_tokens": gen.usage_prompt_tokens,
+                        "completion_tokens": gen.usage_completion_tokens,
+                        "total_tokens": gen.usage_prompt_tokens +
+                        gen.usage_completion_tokens}
+                    break
+            if include_usage and usage:
+                yield _sse({"id": rid, "object": "chat.completion.chunk",
+                            "created": created, "model": model,
+                            "choices": [], "usage": usage})
+            yield "data: [DONE]\n\n"
+        except (asyncio.CancelledError, GeneratorExit):
+            await on_cancel(req)
+            raise
+        except asyncio.TimeoutError:
+            yield _sse({"error": {"message": "stream timeout",
+                                  "type": "server_error"}})
+            await on_cancel(req)
+
+    # -------------------------------------------------------- chat non-stream
+    async def collect_chat(self, req: ServiceRequest) -> dict:
+        token_ids, usage, finish, err = await self._collect(req)
+        if err:
+            return {"error": {"message": err, "type": "server_error"}}
+        text = self.tokenizer.decode(token_ids)
+        rp, tp = make_parsers(req.model, self.parser_mode)
+        reasoning = None
+        tool_calls = []
+        if rp is not None:
+            reasoning, text = rp.extract(text)
+        if tp is not None:
+            text, tool_calls = tp.extract(text)
+        if tool_calls and finish == "stop":
+            finish = "tool_calls"
+        msg = {"role": "assistant", "content": text}
+        if reasoning:
+            msg["reasoning_content"] = reasoning
+        if tool_calls:
+            msg["tool_calls"] = [{
+                "id": tc.id, "type": "function",
+                "function": {"name": tc.name, "arguments": tc.arguments}}
+                for tc in tool_calls]
+        return {
+            "id": req.service_request_id, "object": "chat.completion",
+            "created": int(req.created), "model": req.model,
+            "choices": [{"index": 0, "message": msg, "finish_reason": finish}],
+            "usage": usage}
+
+    # ------------------------------------------------------ completion paths
+    async def stream_completion(self, req: ServiceRequest,
+                                on_cancel) -> AsyncIterator[str]:
+        rid = req.service_request_id
+        created = int(req.created)
+        dec = IncrementalDecoder(self.tokenizer)
+        try:
+            while True:
+                gen: GenerationDelta = await asyncio.wait_for(
+                    req.output_queue.get(), STREAM_TIMEOUT_S)
+                if gen.error:
+                    yield _sse({"error": {"message": gen.error,
+                                          "type": "server_error"}})
+                    break
+                text = dec.push(gen.token_ids) if gen.token_ids else ""
+                if text:
+                    yield _sse({"id": rid, "object": "text_completion",
+                                "created": created, "model": req.model,
+                                "choices": [{"index": 0, "text": text,
+                                             "finish_reason": None}]})
+                if gen.finished:
+                    yield _sse({"id": rid, "object": "text_completion",
+                                "created": created, "model": req.model,
+                                "choices": [{"index": 0, "text": "",
+                                             "finish_reason":
+                                             gen.finish_reason or "stop"}],
+                                "usage": {
+                                    "prompt_tokens": gen.usage_prompt_tokens,
+                                    "completion_tokens":
+                                    gen.usage_completion_tokens,
+                                    "total_tokens": gen.usage_prompt_tokens +
+                                    gen.usage_completion_tokens}})
+                    break
+            yield "data: [DONE]\n\n"
+        except (asyncio.CancelledError, GeneratorExit):
+            await on_cancel(req)
+            raise
+        except asyncio.TimeoutError:
+            yield _sse({"error": {"message": "stream timeout",
+                                  "type": "server_error"}})
+            await on_cancel(req)
+
+    async def collect_completion(self, req: ServiceRequest) -> dict:
+        token_ids, usage, finish, err = await self._collect(req)
+        if err:
+            return {"error": {"message": err, "type": "server_error"}}
+        return {
+            "id": req.service_request_id, "object": "text_completion",
+            "created": int(req.created), "model": req.model,
+            "choices": [{"index": 0, "text": self.tokenizer.decode(token_ids),
+                         "finish_reason": finish}],
+            "usage": usage}
+
+    # ---------------------------------------------------------------- common
+    async def _collect(self, req: ServiceRequest):
+        token_ids = []
+        usage = {"prompt_tokens": 0, "completion_tokens": 0, "total_tokens": 0}
+        finish = "stop"
+        while True:
+            gen: GenerationDelta = await asyncio.wait_for(
+                req.output_queue.get(), STREAM_TIMEOUT_S)
+            if gen.error:
+                return token_ids, usage, finish, gen.error
+            token_ids.extend(gen.token_ids)
+            if gen.finished:
+                finish = gen.finish_reason or "stop"
+                usage = {"prompt_tokens": gen.usage_prompt_tokens,
+                         "completion_tokens": gen.usage_completion_tokens,
+                         "total_tokens": gen.usage_prompt_tokens +
+                         gen.usage_completion_tokens}
+                return token_ids, usage, finish, None
