@@ -37,6 +37,27 @@ def test_gpu_engine_matches_dense_reference():
     assert matches / total >= 0.8, f"only {matches}/{total} tokens matched"
 
 
+def test_graph_decode_matches_eager():
+    """hipGraph-captured decode must reproduce the eager decode exactly."""
+    cfg = get_config("llama-debug-128")
+    torch.manual_seed(33)
+    prompts = [torch.randint(0, cfg.vocab_size, (n,)).tolist()
+               for n in (5, 21, 40, 64)]
+    outs = {}
+    for graphs in (False, True):
+        eng = LLMEngine("llama-debug-128", device="cuda:0", max_kv_blocks=256,
+                        seed=9, enable_graphs=graphs, max_num_seqs=8,
+                        max_model_len=512)
+        if graphs:
+            assert eng.runner.graph_runner is not None
+            assert len(eng.runner.graph_runner.graphs) > 0
+        outs[graphs] = eng.generate(
+            prompts, SamplingParams(max_tokens=12, ignore_eos=True))
+        del eng
+        torch.cuda.empty_cache()
+    assert outs[True] == outs[False]
+
+
 def test_gpu_prefix_cache_and_chunked_prefill():
     eng = LLMEngine("llama-debug-128", device="cuda:0", max_kv_blocks=256,
                     seed=5, max_batched_tokens=64)

@@ -50,6 +50,8 @@ class LLMEngine:
                  gpu_memory_utilization: float = 0.85,
                  max_kv_blocks: Optional[int] = None,
                  enable_prefix_caching: bool = True,
+                 enable_graphs: bool = True,
+                 max_model_len: int = 4096,
                  seed: int = 0):
         self.cfg: ModelConfig = get_config(model_name)
         if device is None:
@@ -75,7 +77,12 @@ class LLMEngine:
                                          max_num_seqs=max_num_seqs,
                                          max_batched_tokens=max_batched_tokens)
         self.runner = ModelRunner(self.model, self.cfg, self.device,
-                                  num_blocks, dtype=dtype)
+                                  num_blocks, dtype=dtype,
+                                  enable_graphs=enable_graphs,
+                                  max_model_len=max_model_len,
+                                  max_graph_batch=min(max_num_seqs, 256))
+        if self.device.type == "cuda":
+            self.runner.capture_graphs()
         self.seqs: Dict[str, Sequence] = {}
         self.held: Dict[str, Sequence] = {}   # finished, blocks kept (PD)
         self.stats = EngineStats()
@@ -149,6 +156,8 @@ class LLMEngine:
     def abort_request(self, request_id: str) -> bool:
         seq = self.scheduler.abort(request_id)
         self.seqs.pop(request_id, None)
+        if self.runner.graph_runner is not None:
+            self.runner.graph_runner.forget(request_id)
         return seq is not None
 
     def has_work(self) -> bool:
@@ -189,6 +198,8 @@ class LLMEngine:
                 seq = self.seqs.pop(out.request_id, None)
                 if seq is not None and seq.hold_blocks:
                     self.held[out.request_id] = seq
+                if self.runner.graph_runner is not None:
+                    self.runner.graph_runner.forget(out.request_id)
         self.stats.steps += 1
         self.stats.num_waiting = self.scheduler.num_waiting
         self.stats.num_running = len(self.scheduler.running)

@@ -1,0 +1,126 @@
+"""hipGraph capture of the decode step (torch.cuda.CUDAGraph == hipGraph on
+ROCm).
+
+Decode steps are launch-bound: ~400 small kernel launches per step left the
+GPU ~78% idle pre-capture (profiles/kernel_stats_r01_pregraph.txt). The
+whole 32-layer decode forward + logits is captured once per batch-size
+bucket and replayed with inputs copied into static buffers; padding rows
+point at block 0 with seq_len=16 so replays are always well-defined.
+"""
+from __future__ import annotations
+
+import logging
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from .metadata import AttnMetadata
+
+log = logging.getLogger("xllm.graph")
+
+BUCKETS = (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256)
+
+
+class DecodeGraphRunner:
+    def __init__(self, model, kv_caches, device, max_model_len: int = 4096,
+                 max_batch: int = 256):
+        self.model = model
+        self.kv_caches = kv_caches
+        self.device = device
+        self.max_blocks = (max_model_len + 15) // 16
+        self.buckets = [b for b in BUCKETS if b <= max_batch]
+        self.graphs: Dict[int, Tuple[torch.cuda.CUDAGraph, dict,
+                                     torch.Tensor]] = {}
+        self.pool = None
+        # per-request padded block-table rows (numpy, cheap incremental update)
+        self._bt_rows: Dict[str, Tuple[np.ndarray, int]] = {}
+
+    def capture_all(self):
+        for bs in sorted(self.buckets, reverse=True):  # largest first: pool
+            self._capture(bs)
+        log.info("captured %d decode graphs (max_blocks=%d)",
+                 len(self.graphs), self.max_blocks)
+
+    @torch.inference_mode()
+    def _capture(self, bs: int):
+        dev = self.device
+        static = dict(
+            input_ids=torch.zeros(bs, dtype=torch.long, device=dev),
+            positions=torch.zeros(bs, dtype=torch.long, device=dev),
+            slot_mapping=torch.full((bs,), -1, dtype=torch.long, device=dev),
+            seq_lens=torch.full((bs,), 16, dtype=torch.int32, device=dev),
+            block_tables=torch.zeros(bs, self.max_blocks, dtype=torch.int32,
+                                     device=dev),
+        )
+        meta = AttnMetadata(
+            num_prefill_tokens=0, num_decode_tokens=bs,
+            slot_mapping=static["slot_mapping"],
+            decode_seq_lens=static["seq_lens"],
+            decode_block_tables=static["block_tables"])
+
+        def fwd():
+            hidden = self.model(static["input_ids"], static["positions"],
+                                self.kv_caches, meta)
+            return self.model.compute_logits(hidden)
+
+        # warm up twice outside capture (allocator + lazy inits settle)
+        for _ in range(2):
+            fwd()
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        ctx = (torch.cuda.graph(graph, pool=self.pool) if self.pool is not None
+               else torch.cuda.graph(graph))
+        with ctx:
+            logits = fwd()
+        if self.pool is None:
+            self.pool = graph.pool()
+        self.graphs[bs] = (graph, static, logits)
+
+    # ------------------------------------------------------------------ run
+    def bucket_for(self, n: int) -> Optional[int]:
+        for b in self.buckets:
+            if b >= n:
+                return b
+        return None
+
+    def block_row(self, seq) -> np.ndarray:
+        """Cached padded block-table row; invalidated by growth AND by
+        preemption (preempt_count is part of the cache signature, since a
+        re-admitted sequence may land on different blocks at equal length)."""
+        sig = (len(seq.block_table), seq.preempt_count)
+        row, cached_sig = self._bt_rows.get(seq.request_id, (None, None))
+        if row is None:
+            row = np.zeros(self.max_blocks, dtype=np.int32)
+        if cached_sig != sig:
+            bt = np.asarray(seq.block_table, dtype=np.int32)
+            row[: len(bt)] = bt
+            self._bt_rows[seq.request_id] = (row, sig)
+        return row
+
+    def forget(self, request_id: str):
+        self._bt_rows.pop(request_id, None)
+
+    @torch.inference_mode()
+    def run(self, input_ids: np.ndarray, positions: np.ndarray,
+            slots: np.ndarray, seq_lens: np.ndarray,
+            bt_rows: List[np.ndarray]) -> torch.Tensor:
+        n = len(input_ids)
+        bs = self.bucket_for(n)
+        graph, static, logits = self.graphs[bs]
+        dev = self.device
+        static["input_ids"][:n].copy_(
+            torch.from_numpy(input_ids), non_blocking=True)
+        static["positions"][:n].copy_(
+            torch.from_numpy(positions), non_blocking=True)
+        static["slot_mapping"][:n].copy_(
+            torch.from_numpy(slots), non_blocking=True)
+        if n < bs:  # neutralize padding rows
+            static["slot_mapping"][n:bs].fill_(-1)
+            static["seq_lens"][n:bs].fill_(16)
+        static["seq_lens"][:n].copy_(
+            torch.from_numpy(seq_lens), non_blocking=True)
+        bt = torch.from_numpy(np.stack(bt_rows))
+        static["block_tables"][:n].copy_(bt, non_blocking=True)
+        graph.replay()
+        return logits[:n]

@@ -23,12 +23,17 @@ BLOCK_SIZE = 16
 
 class ModelRunner:
     def __init__(self, model, cfg: ModelConfig, device: torch.device,
-                 num_blocks: int, dtype=torch.bfloat16):
+                 num_blocks: int, dtype=torch.bfloat16,
+                 enable_graphs: bool = True, max_model_len: int = 4096,
+                 max_graph_batch: int = 256):
         self.model = model
         self.cfg = cfg
         self.device = device
         self.dtype = dtype
         self.num_blocks = num_blocks
+        self.graph_runner = None
+        self._graph_opts = (enable_graphs and device.type == "cuda",
+                            max_model_len, max_graph_batch)
         n_kv = model.local_kv_heads if hasattr(model, "local_kv_heads") else cfg.num_kv_heads
         self.kv_caches: List[Tuple[torch.Tensor, torch.Tensor]] = [
             (torch.zeros(num_blocks, n_kv, BLOCK_SIZE, cfg.head_dim,
@@ -56,6 +61,16 @@ class ModelRunner:
         if max_blocks:
             n = min(n, max_blocks)
         return int(n)
+
+    def capture_graphs(self):
+        enable, max_len, max_bs = self._graph_opts
+        if not enable:
+            return
+        from .graph_runner import DecodeGraphRunner
+        self.graph_runner = DecodeGraphRunner(
+            self.model, self.kv_caches, self.device,
+            max_model_len=max_len, max_batch=max_bs)
+        self.graph_runner.capture_all()
 
     # ---- batch construction -------------------------------------------------
     def _build_batch(self, plan: StepPlan, bm):
@@ -157,6 +172,12 @@ class ModelRunner:
     def execute(self, plan: StepPlan, bm) -> Dict[str, int]:
         """Run one step; returns {request_id: sampled_token} for sequences
         that produced a token this step (completed prefills + decodes)."""
+        gr = self.graph_runner
+        if (gr is not None and not plan.prefills and plan.decodes
+                and gr.bucket_for(len(plan.decodes)) is not None
+                and all(s.total_len <= gr.max_blocks * BLOCK_SIZE
+                        for s in plan.decodes)):
+            return self._execute_decode_graph(plan, bm)
         input_ids, positions, meta = self._build_batch(plan, bm)
         hidden = self.model(input_ids, positions, self.kv_caches, meta)
 
@@ -178,3 +199,27 @@ class ModelRunner:
         logits = self.model.compute_logits(sel)
         tokens = self._sample(logits, seqs)
         return {seq.request_id: tok for seq, tok in zip(seqs, tokens)}
+
+    def _execute_decode_graph(self, plan: StepPlan, bm) -> Dict[str, int]:
+        import numpy as np
+        seqs = plan.decodes
+        n = len(seqs)
+        input_ids = np.empty(n, dtype=np.int64)
+        positions = np.empty(n, dtype=np.int64)
+        slots = np.empty(n, dtype=np.int64)
+        seq_lens = np.empty(n, dtype=np.int32)
+        gr = self.graph_runner
+        bt_rows = []
+        for i, seq in enumerate(seqs):
+            input_ids[i] = (seq.output_token_ids[-1] if seq.output_token_ids
+                            else seq.prompt_token_ids[-1])
+            positions[i] = seq.total_len - 1
+            slots[i] = bm.append_slot(seq)
+            seq_lens[i] = seq.total_len
+            bt_rows.append(gr.block_row(seq))
+        logits = gr.run(input_ids, positions, slots, seq_lens, bt_rows)
+        tokens = self._sample(logits, seqs)
+        out = {}
+        for seq, tok in zip(seqs, tokens):
+            out[seq.request_id] = tok
+        return out
