@@ -20,6 +20,15 @@ import json
 import os
 import time
 
+# hipBLASLt/rocBLAS algo selection tuned offline on MI355X (TunableOp);
+# must be configured before the first torch import in the process.
+_TUNABLE = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                        "configs", "tunableop_gfx950.csv")
+if os.path.exists(_TUNABLE):
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", _TUNABLE)
+
 import torch
 
 

@@ -14,14 +14,15 @@ void launch_fused_add_rmsnorm(unsigned short*, unsigned short*,
                               const unsigned short*, float, int, int,
                               hipStream_t);
 void launch_rope(unsigned short*, unsigned short*, const long*, const float*,
-                 int, int, int, int, int, hipStream_t);
+                 int, int, int, int, int, long, long, hipStream_t);
 void launch_silu_and_mul(unsigned short*, const unsigned short*, int, int,
                          hipStream_t);
 void launch_gelu_and_mul(unsigned short*, const unsigned short*, int, int,
                          hipStream_t);
 void launch_reshape_and_cache(unsigned short*, unsigned short*,
                               const unsigned short*, const unsigned short*,
-                              const long*, int, int, int, int, hipStream_t);
+                              const long*, int, int, int, int, long,
+                              hipStream_t);
 void launch_copy_blocks(unsigned short*, unsigned short*, const long*, int,
                         long, hipStream_t);
 void launch_gather_blocks(unsigned short*, unsigned short*, const long*, int,
@@ -29,12 +30,12 @@ void launch_gather_blocks(unsigned short*, unsigned short*, const long*, int,
 void launch_paged_attn_decode(unsigned short*, const unsigned short*,
                               const unsigned short*, const unsigned short*,
                               const int*, const int*, float, int, int, int,
-                              int, int, hipStream_t);
+                              int, int, long, long, hipStream_t);
 void launch_paged_attn_prefill(unsigned short*, const unsigned short*,
                                const unsigned short*, const unsigned short*,
                                const int*, const int*, const int*, const int*,
-                               const int*, int, float, int, int, int,
-                               hipStream_t);
+                               const int*, int, float, int, int, int, long,
+                               long, hipStream_t);
 void launch_greedy_sample(long*, const unsigned short*, int, int, hipStream_t);
 void launch_mfma_probe(float*, const unsigned short*, const unsigned short*,
                        hipStream_t);
@@ -46,6 +47,15 @@ namespace {
   TORCH_CHECK((t).is_cuda(), #t " must be on GPU");             \
   TORCH_CHECK((t).dtype() == torch::kBFloat16, #t " must be bf16"); \
   TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+// bf16 [T, H, D] (or [T, H*D]) view: innermost contiguous, heads packed;
+// rows may be strided (views into a fused qkv output)
+#define CHECK_BF16_ROWVIEW(t)                                              \
+  TORCH_CHECK((t).is_cuda() && (t).dtype() == torch::kBFloat16,            \
+              #t " must be bf16 on GPU");                                  \
+  TORCH_CHECK((t).stride(-1) == 1, #t " innermost dim must be contiguous");\
+  TORCH_CHECK((t).dim() < 3 || (t).stride(1) == (t).size(2),               \
+              #t " heads must be packed")
 
 inline unsigned short* u16(torch::Tensor& t) {
   return reinterpret_cast<unsigned short*>(t.data_ptr());
@@ -78,7 +88,7 @@ void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
 
 void rope(torch::Tensor positions, torch::Tensor q, torch::Tensor k,
           torch::Tensor cos_sin, long head_dim, long rot_dim) {
-  CHECK_BF16_CUDA(q); CHECK_BF16_CUDA(k);
+  CHECK_BF16_ROWVIEW(q); CHECK_BF16_ROWVIEW(k);
   TORCH_CHECK(positions.dtype() == torch::kLong && positions.is_cuda());
   TORCH_CHECK(cos_sin.dtype() == torch::kFloat && cos_sin.is_cuda());
   const int T = positions.size(0);
@@ -86,7 +96,7 @@ void rope(torch::Tensor positions, torch::Tensor q, torch::Tensor k,
   const int n_k = k.numel() / T / head_dim;
   xllm::launch_rope(u16(q), u16(k), positions.data_ptr<long>(),
                     cos_sin.data_ptr<float>(), T, n_q, n_k, (int)head_dim,
-                    (int)rot_dim, cur_stream());
+                    (int)rot_dim, q.stride(0), k.stride(0), cur_stream());
 }
 
 void silu_and_mul(torch::Tensor out, torch::Tensor x) {
@@ -108,9 +118,10 @@ void gelu_and_mul(torch::Tensor out, torch::Tensor x) {
 void reshape_and_cache(torch::Tensor k, torch::Tensor v,
                        torch::Tensor k_cache, torch::Tensor v_cache,
                        torch::Tensor slot_mapping) {
-  CHECK_BF16_CUDA(k); CHECK_BF16_CUDA(v);
+  CHECK_BF16_ROWVIEW(k); CHECK_BF16_ROWVIEW(v);
   CHECK_BF16_CUDA(k_cache); CHECK_BF16_CUDA(v_cache);
   TORCH_CHECK(slot_mapping.dtype() == torch::kLong && slot_mapping.is_cuda());
+  TORCH_CHECK(k.stride(0) == v.stride(0), "k/v must share row stride");
   const int T = k.size(0);
   const int n_kv = k_cache.size(1);
   const int bs = k_cache.size(2);
@@ -118,7 +129,7 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v,
   TORCH_CHECK(D % 8 == 0);
   xllm::launch_reshape_and_cache(u16(k_cache), u16(v_cache), u16c(k), u16c(v),
                                  slot_mapping.data_ptr<long>(), T, n_kv, D, bs,
-                                 cur_stream());
+                                 k.stride(0), cur_stream());
 }
 
 void copy_blocks(torch::Tensor k_cache, torch::Tensor v_cache,
@@ -145,7 +156,7 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
                        torch::Tensor k_cache, torch::Tensor v_cache,
                        torch::Tensor block_tables, torch::Tensor seq_lens,
                        double scale) {
-  CHECK_BF16_CUDA(out); CHECK_BF16_CUDA(q);
+  CHECK_BF16_ROWVIEW(out); CHECK_BF16_ROWVIEW(q);
   CHECK_BF16_CUDA(k_cache); CHECK_BF16_CUDA(v_cache);
   TORCH_CHECK(block_tables.dtype() == torch::kInt && block_tables.is_cuda());
   TORCH_CHECK(seq_lens.dtype() == torch::kInt && seq_lens.is_cuda());
@@ -160,7 +171,8 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
   xllm::launch_paged_attn_decode(
       u16(out), u16c(q), u16c(k_cache), u16c(v_cache),
       block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), (float)scale,
-      num_seqs, n_qheads, n_kv, D, block_tables.size(1), cur_stream());
+      num_seqs, n_qheads, n_kv, D, block_tables.size(1), q.stride(0),
+      out.stride(0), cur_stream());
 }
 
 void paged_attn_prefill(torch::Tensor out, torch::Tensor q,
@@ -168,7 +180,7 @@ void paged_attn_prefill(torch::Tensor out, torch::Tensor q,
                         torch::Tensor block_tables, torch::Tensor cu_q,
                         torch::Tensor seq_lens, torch::Tensor tile_seq,
                         torch::Tensor tile_q0, double scale) {
-  CHECK_BF16_CUDA(out); CHECK_BF16_CUDA(q);
+  CHECK_BF16_ROWVIEW(out); CHECK_BF16_ROWVIEW(q);
   CHECK_BF16_CUDA(k_cache); CHECK_BF16_CUDA(v_cache);
   for (auto* t : {&block_tables, &cu_q, &seq_lens, &tile_seq, &tile_q0}) {
     TORCH_CHECK(t->dtype() == torch::kInt && t->is_cuda() && t->is_contiguous());
@@ -183,7 +195,7 @@ void paged_attn_prefill(torch::Tensor out, torch::Tensor q,
       block_tables.data_ptr<int>(), cu_q.data_ptr<int>(),
       seq_lens.data_ptr<int>(), tile_seq.data_ptr<int>(),
       tile_q0.data_ptr<int>(), tile_seq.size(0), (float)scale, n_qheads, n_kv,
-      block_tables.size(1), cur_stream());
+      block_tables.size(1), q.stride(0), out.stride(0), cur_stream());
 }
 
 torch::Tensor greedy_sample(torch::Tensor logits) {

@@ -20,7 +20,8 @@ __global__ void reshape_and_cache_kernel(
     const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v,
     const long* __restrict__ slot_mapping,
-    const int n_kv_heads, const int head_dim, const int block_size) {
+    const int n_kv_heads, const int head_dim, const int block_size,
+    const long src_stride) {  // token stride of k/v (strided qkv views)
   const int token = blockIdx.x;
   const long slot = slot_mapping[token];
   if (slot < 0) return;
@@ -31,7 +32,7 @@ __global__ void reshape_and_cache_kernel(
   const int nwaves = blockDim.x >> 6;
 
   for (int h = wid; h < n_kv_heads; h += nwaves) {
-    const long src = ((long)token * n_kv_heads + h) * head_dim;
+    const long src = (long)token * src_stride + (long)h * head_dim;
     const long dst = (((blk * n_kv_heads + h) * block_size) + off) * head_dim;
     // head_dim multiple of 8: lane moves 8 bf16 = 16 B
     for (int i = lane * 8; i < head_dim; i += 64 * 8) {
@@ -46,11 +47,12 @@ __global__ void reshape_and_cache_kernel(
 void launch_reshape_and_cache(unsigned short* k_cache, unsigned short* v_cache,
                               const unsigned short* k, const unsigned short* v,
                               const long* slot_mapping, int T, int n_kv_heads,
-                              int head_dim, int block_size, hipStream_t stream) {
+                              int head_dim, int block_size, long src_stride,
+                              hipStream_t stream) {
   dim3 grid(T), block(256);
   hipLaunchKernelGGL(reshape_and_cache_kernel, grid, block, 0, stream, k_cache,
                      v_cache, k, v, slot_mapping, n_kv_heads, head_dim,
-                     block_size);
+                     block_size, src_stride);
 }
 
 // Copy whole KV blocks within one device (cache defrag, swap, COW fork).

@@ -1,18 +1,21 @@
 // Fused RMSNorm kernels for MI355X (gfx950).
 //
-// Memory-bound: target HBM BW. bf16 I/O vectorized 16 B/lane (ushort8),
-// fp32 accumulation, one workgroup per row (hidden sizes 1k-16k).
+// Single-pass, register-resident: each row is handled by one workgroup of
+// up to 512 threads; a thread keeps its <=32 bf16 elements in registers
+// between the sum-of-squares reduction and the scale, so the row is read
+// once and written once (the fused-add variant also reads/writes the
+// residual stream once). Sized for decode batches (small T) as well as
+// prefill (large T): block width scales with H, not with T.
 //
-// Capability parity: the reference engine's fused RMSNorm (xLLM engine,
-// absent submodule) — re-designed CDNA4-native.
+// Capability parity: the reference engine's fused RMSNorm.
 #include "common.h"
 
 namespace xllm {
 
-// out[t, :] = x[t, :] / rms(x[t, :]) * w
-template <bool FUSED_ADD>
+// ELEMS = number of ushort8 octets each thread owns (H <= block*8*ELEMS)
+template <bool FUSED_ADD, int ELEMS>
 __global__ void rmsnorm_kernel(
-    unsigned short* __restrict__ out,       // [T, H] bf16 (= input if FUSED_ADD)
+    unsigned short* __restrict__ out,       // [T, H] bf16 (= x if FUSED_ADD)
     unsigned short* __restrict__ residual,  // [T, H] bf16 io (FUSED_ADD only)
     const unsigned short* __restrict__ x,   // [T, H] bf16
     const unsigned short* __restrict__ w,   // [H] bf16
@@ -27,60 +30,82 @@ __global__ void rmsnorm_kernel(
   unsigned short* rr = FUSED_ADD ? residual + (long)row * H : nullptr;
   unsigned short* orow = out + (long)row * H;
 
+  float v[ELEMS][8];
   float sumsq = 0.0f;
-  // pass 1: (optional residual add) + sum of squares; vectorized 8 bf16 = 16 B
-  for (int i = tid * 8; i < H; i += nthread * 8) {
-    ushort8_t v = *reinterpret_cast<const ushort8_t*>(xr + i);
-    if constexpr (FUSED_ADD) {
-      ushort8_t r = *reinterpret_cast<ushort8_t*>(rr + i);
 #pragma unroll
-      for (int j = 0; j < 8; j++) {
-        float f = bf16_to_f32(v.x[j]) + bf16_to_f32(r.x[j]);
-        v.x[j] = f32_to_bf16(f);
+  for (int e = 0; e < ELEMS; e++) {
+    const int i = (tid + e * nthread) * 8;
+    if (i < H) {
+      ushort8_t u = *reinterpret_cast<const ushort8_t*>(xr + i);
+#pragma unroll
+      for (int j = 0; j < 8; j++) v[e][j] = bf16_to_f32(u.x[j]);
+      if constexpr (FUSED_ADD) {
+        ushort8_t r = *reinterpret_cast<const ushort8_t*>(rr + i);
+        ushort8_t rw;
+#pragma unroll
+        for (int j = 0; j < 8; j++) {
+          v[e][j] += bf16_to_f32(r.x[j]);
+          rw.x[j] = f32_to_bf16(v[e][j]);
+        }
+        *reinterpret_cast<ushort8_t*>(rr + i) = rw;
       }
-      // write the new residual back (residual stream carries the sum)
-      *reinterpret_cast<ushort8_t*>(rr + i) = v;
-    }
 #pragma unroll
-    for (int j = 0; j < 8; j++) {
-      float f = bf16_to_f32(v.x[j]);
-      sumsq += f * f;
-    }
-    if constexpr (FUSED_ADD) {
-      // stash the summed row in out so pass 2 reads it from there (L2-hot)
-      *reinterpret_cast<ushort8_t*>(orow + i) = v;
+      for (int j = 0; j < 8; j++) sumsq += v[e][j] * v[e][j];
     }
   }
   sumsq = block_reduce_sum(sumsq, red_tmp);
   const float rrms = rsqrtf(sumsq / (float)H + eps);
 
-  // pass 2: scale (rows are L2-resident after pass 1)
-  for (int i = tid * 8; i < H; i += nthread * 8) {
-    ushort8_t v = *reinterpret_cast<const ushort8_t*>((FUSED_ADD ? orow : xr) + i);
-    ushort8_t wv = *reinterpret_cast<const ushort8_t*>(w + i);
 #pragma unroll
-    for (int j = 0; j < 8; j++) {
-      v.x[j] = f32_to_bf16(bf16_to_f32(v.x[j]) * rrms * bf16_to_f32(wv.x[j]));
+  for (int e = 0; e < ELEMS; e++) {
+    const int i = (tid + e * nthread) * 8;
+    if (i < H) {
+      ushort8_t wv = *reinterpret_cast<const ushort8_t*>(w + i);
+      ushort8_t o;
+#pragma unroll
+      for (int j = 0; j < 8; j++)
+        o.x[j] = f32_to_bf16(v[e][j] * rrms * bf16_to_f32(wv.x[j]));
+      *reinterpret_cast<ushort8_t*>(orow + i) = o;
     }
-    *reinterpret_cast<ushort8_t*>(orow + i) = v;
+  }
+}
+
+static inline int norm_block(int H) {
+  int octets = H / 8;
+  int b = octets < 512 ? octets : 512;
+  return ((b + 63) / 64) * 64;  // multiple of the 64-lane wave
+}
+
+#define RMS_LAUNCH(FUSED, E)                                              \
+  hipLaunchKernelGGL((rmsnorm_kernel<FUSED, E>), dim3(T), dim3(block), 0, \
+                     stream, out, residual, x, w, eps, H)
+
+template <bool FUSED>
+static void rmsnorm_dispatch(unsigned short* out, unsigned short* residual,
+                             const unsigned short* x, const unsigned short* w,
+                             float eps, int T, int H, hipStream_t stream) {
+  const int block = norm_block(H);
+  const int elems = (H + block * 8 - 1) / (block * 8);
+  switch (elems) {
+    case 1: RMS_LAUNCH(FUSED, 1); break;
+    case 2: RMS_LAUNCH(FUSED, 2); break;
+    case 3: RMS_LAUNCH(FUSED, 3); break;
+    case 4: RMS_LAUNCH(FUSED, 4); break;
+    default: RMS_LAUNCH(FUSED, 8); break;  // up to H = 32768
   }
 }
 
 void launch_rmsnorm(unsigned short* out, const unsigned short* x,
                     const unsigned short* w, float eps, int T, int H,
                     hipStream_t stream) {
-  dim3 grid(T), block(256);
-  hipLaunchKernelGGL((rmsnorm_kernel<false>), grid, block, 0, stream, out,
-                     nullptr, x, w, eps, H);
+  rmsnorm_dispatch<false>(out, nullptr, x, w, eps, T, H, stream);
 }
 
 void launch_fused_add_rmsnorm(unsigned short* x, unsigned short* residual,
                               const unsigned short* w, float eps, int T, int H,
                               hipStream_t stream) {
   // in-place: residual += x ; x = rmsnorm(residual) * w
-  dim3 grid(T), block(256);
-  hipLaunchKernelGGL((rmsnorm_kernel<true>), grid, block, 0, stream, x,
-                     residual, x, w, eps, H);
+  rmsnorm_dispatch<true>(x, residual, x, w, eps, T, H, stream);
 }
 
 }  // namespace xllm

@@ -34,7 +34,8 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
     const int* __restrict__ block_tables,    // [num_seqs, max_blocks]
     const int* __restrict__ seq_lens,        // [num_seqs]
     const float scale,
-    const int n_kv_heads, const int D, const int max_blocks_per_seq) {
+    const int n_kv_heads, const int D, const int max_blocks_per_seq,
+    const long q_stride, const long out_stride) {
   const int seq = blockIdx.x / n_kv_heads;
   const int kvh = blockIdx.x % n_kv_heads;
   const int n_qheads = n_kv_heads * G;
@@ -51,7 +52,7 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
   float qf[G][8];
 #pragma unroll
   for (int g = 0; g < G; g++) {
-    const long qoff = ((long)seq * n_qheads + (kvh * G + g)) * D + 8 * d8;
+    const long qoff = (long)seq * q_stride + (long)(kvh * G + g) * D + 8 * d8;
     if (dvalid) {
       ushort8_t v = *reinterpret_cast<const ushort8_t*>(q + qoff);
 #pragma unroll
@@ -156,7 +157,7 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
     }
     const float inv = (l_tot > 0.0f) ? 1.0f / l_tot : 0.0f;
     if (2 * lane < D) {
-      const long ooff = ((long)seq * n_qheads + (kvh * G + g)) * D + 2 * lane;
+      const long ooff = (long)seq * out_stride + (long)(kvh * G + g) * D + 2 * lane;
       out[ooff] = f32_to_bf16(o0 * inv);
       out[ooff + 1] = f32_to_bf16(o1 * inv);
     }
@@ -166,7 +167,7 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
 #define PA_DISPATCH_G(GV)                                                     \
   hipLaunchKernelGGL((paged_attn_decode_kernel<GV>), grid, block, 0, stream,  \
                      out, q, k_cache, v_cache, block_tables, seq_lens, scale, \
-                     n_kv_heads, D, max_blocks_per_seq)
+                     n_kv_heads, D, max_blocks_per_seq, q_stride, out_stride)
 
 void launch_paged_attn_decode(unsigned short* out, const unsigned short* q,
                               const unsigned short* k_cache,
@@ -174,6 +175,7 @@ void launch_paged_attn_decode(unsigned short* out, const unsigned short* q,
                               const int* block_tables, const int* seq_lens,
                               float scale, int num_seqs, int n_qheads,
                               int n_kv_heads, int D, int max_blocks_per_seq,
+                              long q_stride, long out_stride,
                               hipStream_t stream) {
   dim3 grid(num_seqs * n_kv_heads), block(256);
   const int G = n_qheads / n_kv_heads;

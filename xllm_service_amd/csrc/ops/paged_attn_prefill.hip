@@ -49,7 +49,8 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_kernel(
     const int* __restrict__ tile_seq,        // [total_tiles] tile -> seq
     const int* __restrict__ tile_q0,         // [total_tiles] tile -> local q row
     const float scale,
-    const int n_qheads, const int n_kv_heads, const int max_blocks_per_seq) {
+    const int n_qheads, const int n_kv_heads, const int max_blocks_per_seq,
+    const long q_stride, const long out_stride) {
   const int tile = blockIdx.x;
   const int qh = blockIdx.y;
   const int kvh = qh / (n_qheads / n_kv_heads);
@@ -83,10 +84,10 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_kernel(
   // ---- load Q fragments (lane holds row frow, cols ks*32+fcol8..+8) ----------
   bf16x8 qf[4];
   if (wave_active && frow < wq_rows) {
-    const long qrow = (long)(q_start + wq0 + frow) * n_qheads + qh;
+    const long qrow = (long)(q_start + wq0 + frow) * q_stride + (long)qh * PF_D;
 #pragma unroll
     for (int ks = 0; ks < 4; ks++)
-      qf[ks] = *reinterpret_cast<const bf16x8*>(q + qrow * PF_D + ks * 32 + fcol8);
+      qf[ks] = *reinterpret_cast<const bf16x8*>(q + qrow + ks * 32 + fcol8);
   } else {
 #pragma unroll
     for (int ks = 0; ks < 4; ks++) qf[ks] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
@@ -232,7 +233,7 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_kernel(
       const int qrow = wq0 + crow4 + r;
       if (qrow >= q_len) continue;
       const float inv = (l_r[r] > 0.0f) ? 1.0f / l_r[r] : 0.0f;
-      const long obase = ((long)(q_start + qrow) * n_qheads + qh) * PF_D;
+      const long obase = (long)(q_start + qrow) * out_stride + (long)qh * PF_D;
 #pragma unroll
       for (int n = 0; n < 8; n++) {
         out[obase + n * 16 + frow] = f32_to_bf16(o_acc[n][r] * inv);
@@ -248,11 +249,13 @@ void launch_paged_attn_prefill(unsigned short* out, const unsigned short* q,
                                const int* seq_lens, const int* tile_seq,
                                const int* tile_q0, int total_tiles, float scale,
                                int n_qheads, int n_kv_heads,
-                               int max_blocks_per_seq, hipStream_t stream) {
+                               int max_blocks_per_seq, long q_stride,
+                               long out_stride, hipStream_t stream) {
   dim3 grid(total_tiles, n_qheads), block(256);
   hipLaunchKernelGGL(paged_attn_prefill_kernel, grid, block, 0, stream, out, q,
                      k_cache, v_cache, block_tables, cu_q, seq_lens, tile_seq,
-                     tile_q0, scale, n_qheads, n_kv_heads, max_blocks_per_seq);
+                     tile_q0, scale, n_qheads, n_kv_heads, max_blocks_per_seq,
+                     q_stride, out_stride);
 }
 
 // ---------------------------------------------------------------------------------

@@ -1,38 +1,42 @@
 // Fused rotary position embedding (RoPE) for MI355X (gfx950).
 //
-// NeoX / Llama "rotate-half" layout, applied in-place to q and k.
-// cos/sin are precomputed host-side as an f32 table [max_pos, rot_dim]
-// (first half cos(theta_i), second half sin(theta_i)): per the CDNA guide,
-// on-device sinf/cosf turns this memory-bound op VALU-bound.
+// NeoX / Llama "rotate-half" layout, applied in-place to q and k, which may
+// be strided row views into a fused qkv projection output (no host-side
+// .contiguous() copies). One 64-lane wave per (token, head) pair; the grid
+// is sized by total wave-work so decode batches fill the chip.
 //
-// Parity target: the reference engine's fused RoPE.
+// cos/sin precomputed host-side as f32 [max_pos, rot_dim] (first half cos):
+// on-device trig would turn this memory-bound op VALU-bound (guide App. B).
+//
+// Parity: the reference engine's fused RoPE.
 #include "common.h"
 
 namespace xllm {
 
-// q: [T, n_qheads * head_dim], k: [T, n_kheads * head_dim]; rot_dim <= head_dim.
-// One wave handles one (token, head): lane i covers the rotation pair
-// (i, i + rot/2) for i < rot/2 (rot/2 <= 64 for head_dim up to 128).
 __global__ void rope_kernel(
-    unsigned short* __restrict__ q,
-    unsigned short* __restrict__ k,
+    unsigned short* __restrict__ q,      // [T, n_qheads, D] rows stride q_stride
+    unsigned short* __restrict__ k,      // [T, n_kheads, D] rows stride k_stride
     const long* __restrict__ positions,  // [T]
     const float* __restrict__ cos_sin,   // [max_pos, rot_dim]
     const int n_qheads, const int n_kheads,
-    const int head_dim, const int rot_dim) {
-  const int token = blockIdx.x;
-  const int wid = threadIdx.x >> 6;       // wave in block (4 waves)
+    const int head_dim, const int rot_dim,
+    const long q_stride, const long k_stride,
+    const int T) {
   const int lane = threadIdx.x & 63;
-  const int nwaves = blockDim.x >> 6;
-  const int half = rot_dim / 2;
-  const long pos = positions[token];
-  const float* cs = cos_sin + pos * rot_dim;
-
+  const int wid = threadIdx.x >> 6;
+  const int waves_per_block = blockDim.x >> 6;
   const int total_heads = n_qheads + n_kheads;
-  for (int h = wid; h < total_heads; h += nwaves) {
+  const long total_work = (long)T * total_heads;
+  const int half = rot_dim / 2;
+
+  for (long wk = blockIdx.x * (long)waves_per_block + wid; wk < total_work;
+       wk += (long)gridDim.x * waves_per_block) {
+    const int token = (int)(wk / total_heads);
+    const int h = (int)(wk % total_heads);
     unsigned short* base =
-        (h < n_qheads) ? q + ((long)token * n_qheads + h) * head_dim
-                       : k + ((long)token * n_kheads + (h - n_qheads)) * head_dim;
+        (h < n_qheads) ? q + token * q_stride + (long)h * head_dim
+                       : k + token * k_stride + (long)(h - n_qheads) * head_dim;
+    const float* cs = cos_sin + positions[token] * rot_dim;
     for (int i = lane; i < half; i += 64) {
       const float c = cs[i];
       const float s = cs[half + i];
@@ -46,10 +50,15 @@ __global__ void rope_kernel(
 
 void launch_rope(unsigned short* q, unsigned short* k, const long* positions,
                  const float* cos_sin, int T, int n_qheads, int n_kheads,
-                 int head_dim, int rot_dim, hipStream_t stream) {
-  dim3 grid(T), block(256);
-  hipLaunchKernelGGL(rope_kernel, grid, block, 0, stream, q, k, positions,
-                     cos_sin, n_qheads, n_kheads, head_dim, rot_dim);
+                 int head_dim, int rot_dim, long q_stride, long k_stride,
+                 hipStream_t stream) {
+  const long waves = (long)T * (n_qheads + n_kheads);
+  long g = (waves + 3) / 4;  // 4 waves per 256-thread block
+  if (g > 2048) g = 2048;
+  if (g < 1) g = 1;
+  hipLaunchKernelGGL(rope_kernel, dim3((unsigned)g), dim3(256), 0, stream, q,
+                     k, positions, cos_sin, n_qheads, n_kheads, head_dim,
+                     rot_dim, q_stride, k_stride, T);
 }
 
 }  // namespace xllm

@@ -38,32 +38,35 @@ class LlamaAttention(nn.Module):
         self.o_proj = RowParallelLinear(cfg.q_size, cfg.hidden_size, dtype=dtype)
 
     def forward(self, x, positions, kv_cache, meta: AttnMetadata, cos_sin):
-        qkv = self.qkv_proj(qkv_in := x)
+        qkv = self.qkv_proj(x)
         q_sz = self.n_heads * self.head_dim
         kv_sz = self.n_kv_heads * self.head_dim
-        q, k, v = torch.split(qkv, [q_sz, kv_sz, kv_sz], dim=-1)
-        q = q.contiguous()
-        k = k.contiguous()
-        q, k = ops.rope(positions, q, k, cos_sin, self.head_dim, self.head_dim)
         T = x.shape[0]
-        qh = q.view(T, self.n_heads, self.head_dim)
-        kh = k.view(T, self.n_kv_heads, self.head_dim)
-        vh = v.view(T, self.n_kv_heads, self.head_dim).contiguous()
+        # strided head views into the fused qkv output — the HIP ops are
+        # stride-aware, so no .contiguous() copies on the hot path
+        qh = qkv[:, :q_sz].unflatten(-1, (self.n_heads, self.head_dim))
+        kh = qkv[:, q_sz:q_sz + kv_sz].unflatten(
+            -1, (self.n_kv_heads, self.head_dim))
+        vh = qkv[:, q_sz + kv_sz:].unflatten(
+            -1, (self.n_kv_heads, self.head_dim))
+        qh, kh = ops.rope(positions, qh, kh, cos_sin, self.head_dim,
+                          self.head_dim)
         k_cache, v_cache = kv_cache
         ops.reshape_and_cache(kh, vh, k_cache, v_cache, meta.slot_mapping)
 
         np_, nd = meta.num_prefill_tokens, meta.num_decode_tokens
-        out = torch.empty_like(qh)
+        out = torch.empty(T, q_sz, dtype=qkv.dtype, device=qkv.device)
+        out3 = out.unflatten(-1, (self.n_heads, self.head_dim))
         if np_:
-            out[:np_] = ops.paged_attn_prefill(
-                qh[:np_].contiguous(), k_cache, v_cache,
-                meta.prefill_block_tables, meta.cu_q, meta.prefill_seq_lens,
-                self.scale)
+            ops.paged_attn_prefill(
+                qh[:np_], k_cache, v_cache, meta.prefill_block_tables,
+                meta.cu_q, meta.prefill_seq_lens, self.scale,
+                out=out3[:np_])
         if nd:
-            out[np_:] = ops.paged_attn_decode(
-                qh[np_:].contiguous(), k_cache, v_cache,
-                meta.decode_block_tables, meta.decode_seq_lens, self.scale)
-        return self.o_proj(out.view(T, q_sz))
+            ops.paged_attn_decode(
+                qh[np_:], k_cache, v_cache, meta.decode_block_tables,
+                meta.decode_seq_lens, self.scale, out=out3[np_:])
+        return self.o_proj(out)
 
 
 class LlamaMLP(nn.Module):

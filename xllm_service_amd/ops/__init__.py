@@ -94,15 +94,21 @@ def copy_blocks(k_cache, v_cache, pairs):
         v_cache[d].copy_(v_cache[s])
 
 
-def paged_attn_decode(q, k_cache, v_cache, block_tables, seq_lens, scale):
+def paged_attn_decode(q, k_cache, v_cache, block_tables, seq_lens, scale,
+                      out=None):
     if q.is_cuda:
         _require_ext()
-        out = torch.empty_like(q)
+        if out is None:
+            out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         _ops.paged_attn_decode(out, q, k_cache, v_cache, block_tables,
                                seq_lens, scale)
         return out
-    return ref.paged_attn_decode(q, k_cache, v_cache, block_tables, seq_lens,
-                                 scale)
+    res = ref.paged_attn_decode(q, k_cache, v_cache, block_tables, seq_lens,
+                                scale)
+    if out is not None:
+        out.copy_(res)
+        return out
+    return res
 
 
 def _prefill_tiles(cu_q):
@@ -117,10 +123,11 @@ def _prefill_tiles(cu_q):
 
 
 def paged_attn_prefill(q, k_cache, v_cache, block_tables, cu_q, seq_lens,
-                       scale):
+                       scale, out=None):
     if q.is_cuda:
         _require_ext()
-        out = torch.empty_like(q)
+        if out is None:
+            out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         tile_seq, tile_q0 = _prefill_tiles(cu_q.cpu())
         dev = q.device
         _ops.paged_attn_prefill(
@@ -128,8 +135,12 @@ def paged_attn_prefill(q, k_cache, v_cache, block_tables, cu_q, seq_lens,
             torch.tensor(tile_seq, dtype=torch.int32, device=dev),
             torch.tensor(tile_q0, dtype=torch.int32, device=dev), scale)
         return out
-    return ref.paged_attn_prefill(q, k_cache, v_cache, block_tables, cu_q,
-                                  seq_lens, scale)
+    res = ref.paged_attn_prefill(q, k_cache, v_cache, block_tables, cu_q,
+                                 seq_lens, scale)
+    if out is not None:
+        out.copy_(res)
+        return out
+    return res
 
 
 def greedy_sample(logits: torch.Tensor) -> torch.Tensor:

@@ -38,7 +38,7 @@ def rope(positions, q, k, cos_sin, head_dim, rot_dim):
     """NeoX rotate-half RoPE, applied out-of-place (reference)."""
     def _apply(x):
         T = positions.shape[0]
-        xs = x.view(T, -1, head_dim).float()
+        xs = x.reshape(T, -1, head_dim).float()
         cs = cos_sin[positions]  # [T, rot_dim]
         cos = cs[:, : rot_dim // 2].unsqueeze(1)
         sin = cs[:, rot_dim // 2:].unsqueeze(1)
@@ -47,7 +47,7 @@ def rope(positions, q, k, cos_sin, head_dim, rot_dim):
         o1 = x1 * cos - x2 * sin
         o2 = x2 * cos + x1 * sin
         out = torch.cat([o1, o2, xs[..., rot_dim:]], dim=-1)
-        return out.to(x.dtype).view(x.shape)
+        return out.to(x.dtype).reshape(x.shape)
     return _apply(q), _apply(k)
 
 
