@@ -104,10 +104,11 @@ def test_reshape_and_cache(dev):
     assert torch.equal(v_cache.float().cpu(), vc)
 
 
-@pytest.mark.parametrize("G", [1, 4, 7, 8])
-def test_paged_attn_decode(dev, G):
+@pytest.mark.parametrize("G,D", [(1, 128), (4, 128), (7, 128), (8, 128),
+                                 (4, 64)])
+def test_paged_attn_decode(dev, G, D):
     torch.manual_seed(G)
-    n_kv, D, bs = 4, 128, 16
+    n_kv, bs = 4, 16
     Hq = n_kv * G
     seq_lens_list = [1, 16, 100, 1023]
     S = len(seq_lens_list)

@@ -1,29 +1,34 @@
-// Paged-attention DECODE kernel for MI355X (gfx950 / CDNA4).
+// Paged-attention DECODE kernel for MI355X (gfx950 / CDNA4) — MFMA-based.
 //
-// Memory-bound regime: one new query token per sequence reads the whole KV
-// cache. The design streams K/V at wide coalesced granularity and keeps the
-// whole online-softmax state in registers:
+// v2 design: the GQA head group IS the MFMA M dimension. For one
+// (sequence, kv_head), the G query heads (G <= 16, zero-padded to 16) form
+// the A operand rows, so QK^T and P·V are mfma_f32_16x16x32_bf16 chains and
+// the VALU does only softmax bookkeeping — v1's shuffle-reduce design was
+// VALU-bound at ~2.1 TB/s of KV stream; here MFMA issue is ~free and the
+// kernel streams K/V at memory speed.
 //
-//   * one workgroup per (sequence, kv_head); 256 threads = 4 waves
-//   * GQA: all G = n_qheads / n_kv_heads query heads of the kv_head are
-//     processed together, so K/V are read ONCE for the whole group
-//   * a wave processes one 16-token KV block per iteration: the 64 lanes are
-//     split into 4 x 16-lane groups, each group owning one token; a lane
-//     loads 8 bf16 (16 B) of the token's 128-dim row -> full 256-B coalesced
-//     row per group, 1 KiB per wave instruction
-//   * scores reduce within the 16-lane group (4 shfl_xor hops)
-//   * each 16-lane group keeps independent online-softmax state (m, l,
-//     acc[G][8]); the 16 partials (4 waves x 4 groups) merge through LDS at
-//     the end (flash-decoding style merge)
+//   * one workgroup (4 waves) per (seq, kv_head); wave w owns KV tiles
+//     w, w+4, w+8... of 32 keys (2 cache blocks) — sequence-partitioned
+//     flash-decoding, merged through LDS at the end
+//   * K feeds the QK^T B-fragment STRAIGHT from the paged cache (the
+//     [tok][d] cache row layout is exactly the B-fragment gather: 16-B
+//     vector loads per lane, no staging)
+//   * V bounces through a per-wave LDS transpose to build P·V B-fragments
+//   * per-row (=per-head) online softmax identical to the prefill kernel
 //
-// Parity: reference engine's paged-attention decode,
-// re-designed for 64-lane waves + 8 TB/s HBM3E rather than ported.
+// Fragment layouts as verified by tests/test_gpu_ops.py::test_mfma_fragment_layout.
+// Parity: reference engine's paged-attention decode.
 #include "common.h"
 
 namespace xllm {
 
-#define PA_BLOCK_SIZE 16   // tokens per KV block (engine-wide constant)
-#define PA_MAX_D 128       // max head_dim supported by this kernel
+typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bf16x8;
+typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4;
+
+#define PA_BS 16           // cache block size (tokens)
+#define PA_KBLK 32         // keys per tile (2 cache blocks)
+#define PA_D 128           // head_dim handled by the MFMA path
+#define PA_PAD 8           // LDS row padding (elements)
 
 template <int G>
 __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
@@ -34,21 +39,209 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
     const int* __restrict__ block_tables,    // [num_seqs, max_blocks]
     const int* __restrict__ seq_lens,        // [num_seqs]
     const float scale,
-    const int n_kv_heads, const int D, const int max_blocks_per_seq,
+    const int n_kv_heads, const int max_blocks_per_seq,
     const long q_stride, const long out_stride) {
   const int seq = blockIdx.x / n_kv_heads;
   const int kvh = blockIdx.x % n_kv_heads;
   const int n_qheads = n_kv_heads * G;
   const int seq_len = seq_lens[seq];
-  const int n_blocks = (seq_len + PA_BLOCK_SIZE - 1) / PA_BLOCK_SIZE;
+  const int n_tiles = (seq_len + PA_KBLK - 1) / PA_KBLK;
 
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int tg = lane >> 4;    // 16-lane group (token slot)
-  const int d8 = lane & 15;    // dim octet: this lane covers dims [8*d8, 8*d8+8)
-  const int dvalid = (8 * d8) < D;  // lanes beyond head_dim idle (D<128)
+  const int frow = lane & 15;
+  const int fcol8 = (lane >> 4) * 8;
+  const int crow4 = (lane >> 4) * 4;
 
-  // ---- load q fragments (shared across all KV) --------------------------------
+  // per-wave LDS scratch + cross-wave merge buffers
+  __shared__ unsigned short p_lds[4][16][PA_KBLK + PA_PAD];
+  __shared__ unsigned short vt_lds[4][PA_D][PA_KBLK + PA_PAD];
+  __shared__ float m_merge[4][16][2];              // (m, l) per wave per row
+  __shared__ float o_merge[4][16][PA_D];           // rescaled O per wave
+
+  // ---- Q as A-fragment: row = head-in-group (zero-padded to 16) -----------
+  bf16x8 qf[4];
+  if (frow < G) {
+    const long qoff = (long)seq * q_stride + (long)(kvh * G + frow) * PA_D;
+#pragma unroll
+    for (int ks = 0; ks < 4; ks++)
+      qf[ks] = *reinterpret_cast<const bf16x8*>(q + qoff + ks * 32 + fcol8);
+  } else {
+#pragma unroll
+    for (int ks = 0; ks < 4; ks++) qf[ks] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  }
+
+  float m_r[4], l_r[4];
+  f32x4 o_acc[8];
+#pragma unroll
+  for (int r = 0; r < 4; r++) { m_r[r] = -INFINITY; l_r[r] = 0.0f; }
+#pragma unroll
+  for (int n = 0; n < 8; n++) o_acc[n] = f32x4{0, 0, 0, 0};
+
+  const int* btab = block_tables + (long)seq * max_blocks_per_seq;
+  const long head_stride = (long)PA_BS * PA_D;
+
+  for (int tile = wid; tile < n_tiles; tile += 4) {
+    const int t0 = tile * PA_KBLK;
+    const long base0 = ((long)btab[t0 / PA_BS] * n_kv_heads + kvh) * head_stride;
+    const long base1 = (t0 + 16 < seq_len)
+        ? ((long)btab[t0 / PA_BS + 1] * n_kv_heads + kvh) * head_stride
+        : base0;
+
+    // ---- S = Q K^T (B-fragment loads directly from the paged cache) ------
+    f32x4 s[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+#pragma unroll
+    for (int n = 0; n < 2; n++) {
+      const long rbase = (n ? base1 : base0) + (long)frow * PA_D;
+      const bool tok_ok = (t0 + n * 16 + frow) < seq_len;
+#pragma unroll
+      for (int ks = 0; ks < 4; ks++) {
+        bf16x8 bk;
+        if (tok_ok) {
+          bk = *reinterpret_cast<const bf16x8*>(
+              k_cache + rbase + ks * 32 + fcol8);
+        } else {
+          bk = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+        }
+        s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[ks], bk, s[n], 0, 0, 0);
+      }
+    }
+
+    // ---- stage V^T for this tile (per-wave private; vector read, scalar
+    // transpose write) ------------------------------------------------------
+    {
+      // lane covers token tv = lane>>1, d-half dv = (lane&1)*64
+      const int tv = lane >> 1;
+      const int dv = (lane & 1) * 64;
+      const bool vok = (t0 + tv) < seq_len;
+      const long vbase = ((tv < 16) ? base0 + (long)tv * PA_D
+                                    : base1 + (long)(tv - 16) * PA_D) + dv;
+#pragma unroll
+      for (int mseg = 0; mseg < 8; mseg++) {
+        ushort8_t vv;
+        if (vok) {
+          vv = *reinterpret_cast<const ushort8_t*>(v_cache + vbase + mseg * 8);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; j++) vv.x[j] = 0;
+        }
+#pragma unroll
+        for (int j = 0; j < 8; j++)
+          vt_lds[wid][dv + mseg * 8 + j][tv] = vv.x[j];
+      }
+    }
+
+    // ---- mask + online softmax (rows are heads) ---------------------------
+    float p_val[2][4];
+    float alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      float smax = -INFINITY;
+#pragma unroll
+      for (int n = 0; n < 2; n++) {
+        float sv = s[n][r] * scale;
+        if (t0 + n * 16 + frow >= seq_len) sv = -INFINITY;
+        p_val[n][r] = sv;
+        smax = fmaxf(smax, sv);
+      }
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        smax = fmaxf(smax, __shfl_xor(smax, off, 64));
+      const float m_new = fmaxf(m_r[r], smax);
+      alpha[r] = (m_new == -INFINITY) ? 1.0f : __expf(m_r[r] - m_new);
+      float psum = 0.0f;
+#pragma unroll
+      for (int n = 0; n < 2; n++) {
+        const float p = (p_val[n][r] == -INFINITY)
+                            ? 0.0f : __expf(p_val[n][r] - m_new);
+        p_val[n][r] = p;
+        psum += p;
+      }
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) psum += __shfl_xor(psum, off, 64);
+      m_r[r] = m_new;
+      l_r[r] = l_r[r] * alpha[r] + psum;
+    }
+    // P -> LDS -> A-fragment
+#pragma unroll
+    for (int n = 0; n < 2; n++) {
+#pragma unroll
+      for (int r = 0; r < 4; r++)
+        p_lds[wid][crow4 + r][n * 16 + frow] = f32_to_bf16(p_val[n][r]);
+    }
+#pragma unroll
+    for (int n = 0; n < 8; n++) {
+#pragma unroll
+      for (int r = 0; r < 4; r++) o_acc[n][r] *= alpha[r];
+    }
+    bf16x8 pa = *reinterpret_cast<const bf16x8*>(&p_lds[wid][frow][fcol8]);
+    // ---- O += P V ---------------------------------------------------------
+#pragma unroll
+    for (int n = 0; n < 8; n++) {
+      bf16x8 bv = *reinterpret_cast<const bf16x8*>(
+          &vt_lds[wid][n * 16 + frow][fcol8]);
+      o_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bv, o_acc[n], 0, 0, 0);
+    }
+  }
+
+  // ---- cross-wave merge ----------------------------------------------------
+  // each wave writes its (m, l) and O to LDS (C layout: row crow4+r, col frow)
+  if (frow == 0) {
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      m_merge[wid][crow4 + r][0] = m_r[r];
+      m_merge[wid][crow4 + r][1] = l_r[r];
+    }
+  }
+#pragma unroll
+  for (int n = 0; n < 8; n++) {
+#pragma unroll
+    for (int r = 0; r < 4; r++)
+      o_merge[wid][crow4 + r][n * 16 + frow] = o_acc[n][r];
+  }
+  __syncthreads();
+
+  // 256 threads cover (g, d) pairs: G*128 <= 2048 elements
+  for (int idx = threadIdx.x; idx < G * PA_D; idx += 256) {
+    const int g = idx / PA_D;
+    const int d = idx % PA_D;
+    float m_star = -INFINITY;
+#pragma unroll
+    for (int w = 0; w < 4; w++) m_star = fmaxf(m_star, m_merge[w][g][0]);
+    float l_tot = 0.0f, o = 0.0f;
+#pragma unroll
+    for (int w = 0; w < 4; w++) {
+      const float mw = m_merge[w][g][0];
+      const float wgt = (mw == -INFINITY) ? 0.0f : __expf(mw - m_star);
+      l_tot += wgt * m_merge[w][g][1];
+      o += wgt * o_merge[w][g][d];
+    }
+    const float inv = (l_tot > 0.0f) ? 1.0f / l_tot : 0.0f;
+    out[(long)seq * out_stride + (long)(kvh * G + g) * PA_D + d] =
+        f32_to_bf16(o * inv);
+  }
+}
+
+// ---------------- fallback (head_dim != 128): v1 shuffle-reduce design ------
+template <int G>
+__global__ __launch_bounds__(256) void paged_attn_decode_small_kernel(
+    unsigned short* __restrict__ out, const unsigned short* __restrict__ q,
+    const unsigned short* __restrict__ k_cache,
+    const unsigned short* __restrict__ v_cache,
+    const int* __restrict__ block_tables, const int* __restrict__ seq_lens,
+    const float scale, const int n_kv_heads, const int D,
+    const int max_blocks_per_seq, const long q_stride, const long out_stride) {
+  const int seq = blockIdx.x / n_kv_heads;
+  const int kvh = blockIdx.x % n_kv_heads;
+  const int n_qheads = n_kv_heads * G;
+  const int seq_len = seq_lens[seq];
+  const int n_blocks = (seq_len + PA_BS - 1) / PA_BS;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int tg = lane >> 4;
+  const int d8 = lane & 15;
+  const int dvalid = (8 * d8) < D;
+
   float qf[G][8];
 #pragma unroll
   for (int g = 0; g < G; g++) {
@@ -62,75 +255,55 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
       for (int j = 0; j < 8; j++) qf[g][j] = 0.0f;
     }
   }
-
-  // ---- stream KV blocks, online softmax ---------------------------------------
   float m[G], l[G], acc[G][8];
 #pragma unroll
   for (int g = 0; g < G; g++) {
-    m[g] = -INFINITY;
-    l[g] = 0.0f;
+    m[g] = -INFINITY; l[g] = 0.0f;
 #pragma unroll
     for (int j = 0; j < 8; j++) acc[g][j] = 0.0f;
   }
-
   const int* btab = block_tables + (long)seq * max_blocks_per_seq;
-  const long kv_head_stride = (long)PA_BLOCK_SIZE * D;
-
+  const long head_stride = (long)PA_BS * D;
   for (int bi = wid; bi < n_blocks; bi += 4) {
-    const long phys = btab[bi];
-    const long base = (phys * n_kv_heads + kvh) * kv_head_stride;
+    const long base = ((long)btab[bi] * n_kv_heads + kvh) * head_stride;
 #pragma unroll
     for (int j = 0; j < 4; j++) {
-      const int t = j * 4 + tg;  // token within block, one per 16-lane group
-      const int tok = bi * PA_BLOCK_SIZE + t;
-      if (tok >= seq_len) continue;  // group-uniform branch
+      const int t = j * 4 + tg;
+      const int tok = bi * PA_BS + t;
+      if (tok >= seq_len) continue;
       const long roff = base + (long)t * D + 8 * d8;
-      float kf[8];
+      float kf[8], vf[8];
       if (dvalid) {
-        ushort8_t kv8 = *reinterpret_cast<const ushort8_t*>(k_cache + roff);
+        ushort8_t k8 = *reinterpret_cast<const ushort8_t*>(k_cache + roff);
+        ushort8_t v8 = *reinterpret_cast<const ushort8_t*>(v_cache + roff);
 #pragma unroll
-        for (int u = 0; u < 8; u++) kf[u] = bf16_to_f32(kv8.x[u]);
+        for (int u = 0; u < 8; u++) { kf[u] = bf16_to_f32(k8.x[u]); vf[u] = bf16_to_f32(v8.x[u]); }
       } else {
 #pragma unroll
-        for (int u = 0; u < 8; u++) kf[u] = 0.0f;
-      }
-      float vf[8];
-      if (dvalid) {
-        ushort8_t vv8 = *reinterpret_cast<const ushort8_t*>(v_cache + roff);
-#pragma unroll
-        for (int u = 0; u < 8; u++) vf[u] = bf16_to_f32(vv8.x[u]);
-      } else {
-#pragma unroll
-        for (int u = 0; u < 8; u++) vf[u] = 0.0f;
+        for (int u = 0; u < 8; u++) { kf[u] = 0.0f; vf[u] = 0.0f; }
       }
 #pragma unroll
       for (int g = 0; g < G; g++) {
         float s = 0.0f;
 #pragma unroll
         for (int u = 0; u < 8; u++) s += qf[g][u] * kf[u];
-        s = group_reduce_sum<16>(s) * scale;  // uniform across the 16 lanes
+        s = group_reduce_sum<16>(s) * scale;
         const float m_new = fmaxf(m[g], s);
-        const float alpha = __expf(m[g] - m_new);  // exp(-inf)=0 on first hit
+        const float a = __expf(m[g] - m_new);
         const float p = __expf(s - m_new);
         m[g] = m_new;
-        l[g] = l[g] * alpha + p;
+        l[g] = l[g] * a + p;
 #pragma unroll
-        for (int u = 0; u < 8; u++) acc[g][u] = acc[g][u] * alpha + p * vf[u];
+        for (int u = 0; u < 8; u++) acc[g][u] = acc[g][u] * a + p * vf[u];
       }
     }
   }
-
-  // ---- merge the 16 partials through LDS --------------------------------------
-  // layout: ml[16][G][2], accs[16][G][16 lanes][8]
   __shared__ float ml[16][G][2];
   __shared__ float accs[16][G][16][8];
   const int part = wid * 4 + tg;
   if (d8 == 0) {
 #pragma unroll
-    for (int g = 0; g < G; g++) {
-      ml[part][g][0] = m[g];
-      ml[part][g][1] = l[g];
-    }
+    for (int g = 0; g < G; g++) { ml[part][g][0] = m[g]; ml[part][g][1] = l[g]; }
   }
 #pragma unroll
   for (int g = 0; g < G; g++) {
@@ -138,19 +311,17 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
     for (int u = 0; u < 8; u++) accs[part][g][d8][u] = acc[g][u];
   }
   __syncthreads();
-
-  // wave `wid` merges heads g = wid, wid+4, ... ; lanes cover the 128 dims
   for (int g = wid; g < G; g += 4) {
     float m_star = -INFINITY;
 #pragma unroll
     for (int p = 0; p < 16; p++) m_star = fmaxf(m_star, ml[p][g][0]);
-    float l_tot = 0.0f;
-    float o0 = 0.0f, o1 = 0.0f;  // lane covers dims (2*lane, 2*lane+1)
-    const int pd8 = lane >> 2;         // octet holding dim 2*lane (=2*lane/8)
-    const int pu = (2 * lane) & 7;     // position within octet
+    float l_tot = 0.0f, o0 = 0.0f, o1 = 0.0f;
+    const int pd8 = lane >> 2;
+    const int pu = (2 * lane) & 7;
 #pragma unroll
     for (int p = 0; p < 16; p++) {
-      const float w = __expf(ml[p][g][0] - m_star);  // 0 if m=-inf
+      const float mw = ml[p][g][0];
+      const float w = (mw == -INFINITY) ? 0.0f : __expf(mw - m_star);
       l_tot += w * ml[p][g][1];
       o0 += w * accs[p][g][pd8][pu];
       o1 += w * accs[p][g][pd8][pu + 1];
@@ -164,10 +335,19 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
   }
 }
 
-#define PA_DISPATCH_G(GV)                                                     \
-  hipLaunchKernelGGL((paged_attn_decode_kernel<GV>), grid, block, 0, stream,  \
-                     out, q, k_cache, v_cache, block_tables, seq_lens, scale, \
-                     n_kv_heads, D, max_blocks_per_seq, q_stride, out_stride)
+#define PA_DISPATCH_G(GV)                                                      \
+  do {                                                                         \
+    if (D == PA_D)                                                             \
+      hipLaunchKernelGGL((paged_attn_decode_kernel<GV>), grid, block, 0,       \
+                         stream, out, q, k_cache, v_cache, block_tables,       \
+                         seq_lens, scale, n_kv_heads, max_blocks_per_seq,      \
+                         q_stride, out_stride);                                \
+    else                                                                       \
+      hipLaunchKernelGGL((paged_attn_decode_small_kernel<GV>), grid, block, 0, \
+                         stream, out, q, k_cache, v_cache, block_tables,       \
+                         seq_lens, scale, n_kv_heads, D, max_blocks_per_seq,   \
+                         q_stride, out_stride);                                \
+  } while (0)
 
 void launch_paged_attn_decode(unsigned short* out, const unsigned short* q,
                               const unsigned short* k_cache,

@@ -57,6 +57,8 @@ class LLMEngine:
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
         self.device = torch.device(device)
+        if self.device.type == "cuda":
+            self._load_gemm_tuning()
         if self.device.type == "cpu" and dtype == torch.bfloat16:
             dtype = torch.float32  # CPU reference path runs fp32
         self.dtype = dtype
@@ -87,6 +89,26 @@ class LLMEngine:
         self.held: Dict[str, Sequence] = {}   # finished, blocks kept (PD)
         self.stats = EngineStats()
         self.eos_token_id: Optional[int] = None  # set by tokenizer owner
+
+    @staticmethod
+    def _load_gemm_tuning():
+        """Load the offline-tuned hipBLASLt/rocBLAS algo table
+        (configs/tunableop_gfx950.csv, produced by PYTORCH_TUNABLEOP_TUNING
+        on an MI355X) so library GEMMs use the fastest algo per shape."""
+        import os
+        csv = os.path.join(os.path.dirname(os.path.dirname(
+            os.path.dirname(os.path.abspath(__file__)))),
+            "configs", "tunableop_gfx950.csv")
+        if not os.path.exists(csv):
+            return
+        try:
+            import torch.cuda.tunable as tunable
+            tunable.enable(True)
+            tunable.tuning_enable(False)
+            tunable.set_filename(csv)
+            tunable.read_file(csv)
+        except Exception:
+            pass
 
     # ---- request API --------------------------------------------------------
     def add_request(self, request_id: str, prompt_token_ids: List[int],
