@@ -6,6 +6,7 @@
 #include <ATen/hip/HIPContext.h>
 
 #include <vector>
+#include <cstring>
 
 namespace xllm {
 void launch_rmsnorm(unsigned short*, const unsigned short*,
@@ -251,6 +252,72 @@ void migrate_blocks_peer(torch::Tensor dst_cache, long dst_device,
   }
 }
 
+// ---- HIP IPC: cross-process cache sharing for same-node KV migration ----
+// A decode worker opens the prefill worker's cache once at LinkInstance
+// time, then pulls blocks with device-to-device copies over xGMI.
+std::vector<char> ipc_get_handle(torch::Tensor t) {
+  TORCH_CHECK(t.is_cuda());
+  hipIpcMemHandle_t h;
+  hipError_t err = hipIpcGetMemHandle(&h, t.data_ptr());
+  TORCH_CHECK(err == hipSuccess, "hipIpcGetMemHandle: ",
+              hipGetErrorString(err));
+  return std::vector<char>(reinterpret_cast<char*>(&h),
+                           reinterpret_cast<char*>(&h) + sizeof(h));
+}
+
+int64_t ipc_open_handle(std::vector<char> handle_bytes, long device) {
+  TORCH_CHECK(handle_bytes.size() == sizeof(hipIpcMemHandle_t));
+  hipIpcMemHandle_t h;
+  memcpy(&h, handle_bytes.data(), sizeof(h));
+  hipSetDevice((int)device);
+  void* ptr = nullptr;
+  hipError_t err =
+      hipIpcOpenMemHandle(&ptr, h, hipIpcMemLazyEnablePeerAccess);
+  TORCH_CHECK(err == hipSuccess, "hipIpcOpenMemHandle: ",
+              hipGetErrorString(err));
+  return reinterpret_cast<int64_t>(ptr);
+}
+
+void ipc_close_handle(int64_t ptr) {
+  hipIpcCloseMemHandle(reinterpret_cast<void*>(ptr));
+}
+
+// Pull blocks from a peer cache (opened via IPC) into the local cache.
+// Both caches share geometry; contiguous runs collapse into single copies.
+void migrate_blocks_from_ptr(torch::Tensor dst_cache, int64_t src_ptr,
+                             long src_device, long dst_device,
+                             std::vector<long> src_blocks,
+                             std::vector<long> dst_blocks) {
+  TORCH_CHECK(src_blocks.size() == dst_blocks.size());
+  const long numel =
+      (long)dst_cache.size(1) * dst_cache.size(2) * dst_cache.size(3);
+  const long bytes = numel * dst_cache.element_size();
+  char* dst = reinterpret_cast<char*>(dst_cache.data_ptr());
+  const char* src = reinterpret_cast<const char*>(src_ptr);
+  hipStream_t stream = cur_stream();
+  size_t i = 0;
+  while (i < src_blocks.size()) {
+    size_t j = i + 1;
+    while (j < src_blocks.size() && src_blocks[j] == src_blocks[j - 1] + 1 &&
+           dst_blocks[j] == dst_blocks[j - 1] + 1)
+      j++;
+    const long n = (long)(j - i);
+    hipError_t err;
+    if (src_device == dst_device) {
+      err = hipMemcpyAsync(dst + dst_blocks[i] * bytes,
+                           src + src_blocks[i] * bytes, (size_t)(n * bytes),
+                           hipMemcpyDeviceToDevice, stream);
+    } else {
+      err = hipMemcpyPeerAsync(dst + dst_blocks[i] * bytes, (int)dst_device,
+                               src + src_blocks[i] * bytes, (int)src_device,
+                               (size_t)(n * bytes), stream);
+    }
+    TORCH_CHECK(err == hipSuccess, "block migration copy failed: ",
+                hipGetErrorString(err));
+    i = j;
+  }
+}
+
 void enable_peer_access(long device, long peer) {
   int can = 0;
   hipError_t err = hipDeviceCanAccessPeer(&can, (int)device, (int)peer);
@@ -278,5 +345,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("greedy_sample", &greedy_sample);
   m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
   m.def("migrate_blocks_peer", &migrate_blocks_peer);
+  m.def("ipc_get_handle", &ipc_get_handle);
+  m.def("ipc_open_handle", &ipc_open_handle);
+  m.def("ipc_close_handle", &ipc_close_handle);
+  m.def("migrate_blocks_from_ptr", &migrate_blocks_from_ptr);
   m.def("enable_peer_access", &enable_peer_access);
 }

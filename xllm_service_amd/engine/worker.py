@@ -86,6 +86,7 @@ class Worker:
         self.rpc_server: Optional[msgrpc.Server] = None
         self.peers: Dict[str, InstanceMetaInfo] = {}
         self.peer_conns: Dict[str, msgrpc.Connection] = {}
+        self.mig = None  # decode-side xGMI MigrationManager (lazy)
         # requests this PREFILL instance must migrate after the first token:
         # rid -> dict(routing/params)
         self.pending_migration: Dict[str, Dict[str, Any]] = {}
@@ -253,7 +254,16 @@ class Worker:
         c = self.peer_conns.pop(peer_name, None)
         if c:
             await c.close()
+        if self.mig is not None:
+            await self._run_on_engine(
+                lambda: self.mig.close_peer(peer_name))
         return True
+
+    async def rpc_export_cache(self, conn) -> dict:
+        """Prefill-side: IPC handles of the KV cache for xGMI migration."""
+        from .kv_migration import export_cache_handles
+        return await self._run_on_engine(
+            lambda: export_cache_handles(self.engine))
 
     async def _peer_conn(self, name: str) -> Optional[msgrpc.Connection]:
         c = self.peer_conns.get(name)
@@ -320,15 +330,26 @@ class Worker:
         meta = self.req_meta.setdefault(service_request_id, {})
         meta.update(prompt_len=len(prompt_token_ids), params=params)
 
+        if transport == "xgmi":
+            # open the peer's cache over IPC once, then pull over xGMI
+            from .kv_migration import MigrationManager
+            if self.mig is None:
+                self.mig = MigrationManager(self.engine)
+            if not self.mig.has_peer(src_name):
+                pconn = await self._peer_conn(src_name)
+                if pconn is None:
+                    raise RuntimeError(f"peer {src_name} unreachable")
+                exported = await pconn.call("export_cache", timeout=30.0)
+                await self._run_on_engine(
+                    lambda: self.mig.open_peer(src_name, exported))
+
         def _recv():
             blocks = self.engine.alloc_migration_blocks(n_blocks)
             try:
                 if transport == "bytes":
                     self.engine.import_block_bytes(blocks, data)
                 elif transport == "xgmi":
-                    from .kv_migration import migrate_in_xgmi
-                    migrate_in_xgmi(self.engine, self.peers.get(src_name),
-                                    src_blocks, blocks)
+                    self.mig.pull_blocks(src_name, src_blocks, blocks)
                 else:
                     raise ValueError(f"unknown transport {transport}")
             except Exception:
@@ -479,3 +500,45 @@ class Worker:
                 self._tbt_samples.clear()
             except Exception:
                 pass
+
+
+def main():
+    """CLI: python -m xllm_service_amd.engine.worker --name w0 ..."""
+    import argparse
+    ap = argparse.ArgumentParser(description="xllm-service-amd worker")
+    ap.add_argument("--name", required=True)
+    ap.add_argument("--type", default="DEFAULT",
+                    choices=[t.value for t in InstanceType])
+    ap.add_argument("--model", default="llama-3-8b")
+    ap.add_argument("--device", default=None)
+    ap.add_argument("--registry-host", default="127.0.0.1")
+    ap.add_argument("--registry-port", type=int, required=True)
+    ap.add_argument("--rpc-host", default="127.0.0.1")
+    ap.add_argument("--rpc-port", type=int, default=0)
+    ap.add_argument("--max-kv-blocks", type=int, default=None)
+    ap.add_argument("--max-num-seqs", type=int, default=256)
+    ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--no-graphs", action="store_true")
+    args = ap.parse_args()
+    logging.basicConfig(level=logging.INFO)
+
+    worker = Worker(
+        args.name, args.type, model=args.model, device=args.device,
+        registry_host=args.registry_host, registry_port=args.registry_port,
+        rpc_host=args.rpc_host, rpc_port=args.rpc_port,
+        max_kv_blocks=args.max_kv_blocks,
+        engine_kwargs=dict(seed=args.seed, max_num_seqs=args.max_num_seqs,
+                           enable_graphs=not args.no_graphs))
+
+    async def run():
+        await worker.start()
+        try:
+            await asyncio.Event().wait()
+        finally:
+            await worker.stop()
+
+    asyncio.run(run())
+
+
+if __name__ == "__main__":
+    main()
