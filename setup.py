@@ -31,6 +31,7 @@ ext_modules = [
                 "paged_attn_decode.hip",
                 "paged_attn_prefill.hip",
                 "sampling.hip",
+                "gemm.hip",
             )
         ],
         extra_compile_args={

@@ -23,8 +23,8 @@ def anyio_backend():
     return "asyncio"
 
 
-def make_master(policy="RR", **kw):
-    opts = MasterOptions(load_balance_policy=policy, model_id="llama-tiny",
+def make_master(policy="RR", model_id="llama-tiny", **kw):
+    opts = MasterOptions(load_balance_policy=policy, model_id=model_id,
                          host_registry=True, registry_port=0, rpc_port=0,
                          **kw)
     return Master(opts)
@@ -39,8 +39,8 @@ async def wait_for(cond, timeout=10.0, interval=0.05):
     raise TimeoutError("condition not met")
 
 
-def worker_kwargs(master, **kw):
-    base = dict(model="llama-tiny", device="cpu",
+def worker_kwargs(master, model="llama-tiny", **kw):
+    base = dict(model=model, device="cpu",
                 registry_host="127.0.0.1",
                 registry_port=master.opts.registry_port,
                 max_kv_blocks=256, heartbeat_s=0.2, lease_ttl_s=0.6,

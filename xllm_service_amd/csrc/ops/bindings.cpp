@@ -38,6 +38,9 @@ void launch_paged_attn_prefill(unsigned short*, const unsigned short*,
                                const int*, int, float, int, int, int, long,
                                long, hipStream_t);
 void launch_greedy_sample(long*, const unsigned short*, int, int, hipStream_t);
+void launch_mfma_gemm(unsigned short*, const unsigned short*,
+                      const unsigned short*, const unsigned short*, int, int,
+                      int, hipStream_t);
 void launch_mfma_probe(float*, const unsigned short*, const unsigned short*,
                        hipStream_t);
 }  // namespace xllm
@@ -209,6 +212,24 @@ torch::Tensor greedy_sample(torch::Tensor logits) {
   return out;
 }
 
+torch::Tensor mfma_gemm(torch::Tensor a, torch::Tensor b,
+                        c10::optional<torch::Tensor> bias) {
+  // a: [M, K]; b: [N, K] (torch weight layout); returns [M, N]
+  CHECK_BF16_CUDA(a); CHECK_BF16_CUDA(b);
+  TORCH_CHECK(a.size(1) == b.size(1), "K mismatch");
+  TORCH_CHECK(a.size(1) % 8 == 0, "K must be a multiple of 8");
+  const int M = a.size(0), K = a.size(1), N = b.size(0);
+  auto c = torch::empty({M, N}, a.options());
+  const unsigned short* bp = nullptr;
+  if (bias.has_value()) {
+    CHECK_BF16_CUDA(bias.value());
+    bp = u16c(bias.value());
+  }
+  xllm::launch_mfma_gemm(u16(c), u16c(a), u16c(b), bp, M, N, K,
+                         cur_stream());
+  return c;
+}
+
 torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b) {
   CHECK_BF16_CUDA(a); CHECK_BF16_CUDA(b);
   TORCH_CHECK(a.size(0) == 16 && a.size(1) == 32);
@@ -344,6 +365,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attn_prefill", &paged_attn_prefill);
   m.def("greedy_sample", &greedy_sample);
   m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
+  m.def("mfma_gemm", &mfma_gemm);
   m.def("migrate_blocks_peer", &migrate_blocks_peer);
   m.def("ipc_get_handle", &ipc_get_handle);
   m.def("ipc_open_handle", &ipc_open_handle);

@@ -115,7 +115,8 @@ class LLMEngine:
                     params: Optional[SamplingParams] = None,
                     priority: int = 0,
                     eos_token_id: Optional[int] = None,
-                    hold_blocks: bool = False) -> None:
+                    hold_blocks: bool = False,
+                    mm_embeds=None) -> None:
         if request_id in self.seqs:
             raise ValueError(f"duplicate request_id {request_id}")
         seq = Sequence(request_id=request_id,
@@ -125,6 +126,10 @@ class LLMEngine:
                        else self.eos_token_id,
                        priority=priority,
                        hold_blocks=hold_blocks)
+        if mm_embeds is not None:
+            seq.mm_embeds = mm_embeds
+            seq.mm_placeholder = getattr(self.model, "image_pad_token_id",
+                                         None)
         self.seqs[request_id] = seq
         self.scheduler.add(seq)
 

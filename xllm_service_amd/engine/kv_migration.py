@@ -3,67 +3,75 @@
 Design: on one 8xMI355X node every GPU pair is one hop over
 7x~153 GB/s xGMI links, so the reference's device-network KV-transfer
 (LinkInstance with device_ips/ports) degenerates to:
-  * at link time, the prefill worker exports hipIpc handles for its KV cache
-    tensors (one per layer, k and v), sent to the decode worker over RPC
-  * the decode worker opens them once (hipIpcOpenMemHandle with lazy peer
-    access = hipDeviceEnablePeerAccess under the hood)
-  * per migration, the decode worker PULLS the prompt's blocks with
-    hipMemcpyPeerAsync per contiguous run, per layer, on a dedicated side
-    stream so copies overlap with the decode compute stream
+  * at link time the prefill worker exports its per-layer KV cache tensors
+    with torch's CUDA-IPC storage sharing (dmabuf handles on this driver —
+    HSA_ENABLE_IPC_MODE_LEGACY=0); sharing through the storage layer keeps
+    allocator-pool offsets correct, which raw hipIpcGetMemHandle on a
+    tensor data_ptr would lose
+  * the decode worker materialises tensor views of the peer's caches once,
+    then PULLS blocks per migration with hipMemcpyPeerAsync per contiguous
+    run (ops.migrate_blocks_peer), on a dedicated side stream so copies
+    overlap the decode compute stream
 
 The serialized-bytes RPC transport (engine.export_block_bytes) remains the
-fallback for cross-node peers and for the CPU test path.
+fallback for cross-node peers and the CPU test path.
 """
 from __future__ import annotations
 
 import logging
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import torch
+
+from xllm_service_amd import ops
 
 log = logging.getLogger("xllm.kv_migration")
 
 
 def available() -> bool:
-    try:
-        from xllm_service_amd import _ops
-        return torch.cuda.is_available() and hasattr(_ops, "ipc_open_handle")
-    except ImportError:
-        return False
+    return torch.cuda.is_available() and ops.HAS_EXT
+
+
+def _share_tensor(t: torch.Tensor) -> dict:
+    storage = t.untyped_storage()
+    (device, handle, storage_size, storage_offset, ref_handle, ref_offset,
+     event_handle, event_sync) = storage._share_cuda_()
+    return dict(device=device, handle=bytes(handle),
+                storage_size=storage_size, storage_offset=storage_offset,
+                ref_handle=bytes(ref_handle) if ref_handle else b"",
+                ref_offset=ref_offset,
+                event_handle=bytes(event_handle) if event_handle else b"",
+                event_sync=bool(event_sync),
+                shape=list(t.shape), dtype=str(t.dtype).split(".")[-1],
+                tensor_offset=t.storage_offset())
+
+
+def _open_tensor(d: dict) -> torch.Tensor:
+    storage = torch.UntypedStorage._new_shared_cuda(
+        d["device"], d["handle"], d["storage_size"], d["storage_offset"],
+        d["ref_handle"], d["ref_offset"], d["event_handle"], d["event_sync"])
+    dtype = getattr(torch, d["dtype"])
+    t = torch.empty(0, dtype=dtype, device=f"cuda:{d['device']}")
+    t.set_(storage, d["tensor_offset"], d["shape"])
+    return t
 
 
 def export_cache_handles(engine) -> dict:
-    """Prefill-side: IPC handles + geometry for every layer's k/v cache."""
-    from xllm_service_amd import _ops
+    """Prefill-side: shareable descriptors for every layer's k/v cache."""
     handles = []
     for (kc, vc) in engine.runner.kv_caches:
-        handles.append([bytes(_ops.ipc_get_handle(kc)),
-                        bytes(_ops.ipc_get_handle(vc))])
-    kc0 = engine.runner.kv_caches[0][0]
-    return dict(
-        handles=handles,
-        device=engine.device.index or 0,
-        num_blocks=int(kc0.shape[0]),
-        shape=list(kc0.shape[1:]),
-    )
+        handles.append([_share_tensor(kc), _share_tensor(vc)])
+    return dict(handles=handles, device=engine.device.index or 0)
 
 
 @dataclass
 class PeerCache:
-    """Decode-side view of one peer's opened cache (per layer k/v ptrs)."""
     src_device: int
-    ptrs: List[List[int]] = field(default_factory=list)  # [layer][k,v]
+    views: List[List[torch.Tensor]] = field(default_factory=list)
 
     def close(self):
-        from xllm_service_amd import _ops
-        for pk, pv in self.ptrs:
-            try:
-                _ops.ipc_close_handle(pk)
-                _ops.ipc_close_handle(pv)
-            except Exception:
-                pass
-        self.ptrs = []
+        self.views = []
 
 
 class MigrationManager:
@@ -76,22 +84,19 @@ class MigrationManager:
                        if engine.device.type == "cuda" else None)
 
     def open_peer(self, name: str, exported: dict):
-        from xllm_service_amd import _ops
         self.close_peer(name)
         pc = PeerCache(src_device=int(exported["device"]))
         my_dev = self.engine.device.index or 0
         if pc.src_device != my_dev:
             try:
-                _ops.enable_peer_access(my_dev, pc.src_device)
-            except Exception as e:  # same-device IPC needs no peer access
+                ops.enable_peer_access(my_dev, pc.src_device)
+            except Exception as e:
                 log.warning("peer access %d->%d: %s", my_dev, pc.src_device, e)
         for hk, hv in exported["handles"]:
-            pk = _ops.ipc_open_handle(list(hk), my_dev)
-            pv = _ops.ipc_open_handle(list(hv), my_dev)
-            pc.ptrs.append([pk, pv])
+            pc.views.append([_open_tensor(hk), _open_tensor(hv)])
         self.peers[name] = pc
-        log.info("opened IPC cache of peer %s (device %d, %d layers)", name,
-                 pc.src_device, len(pc.ptrs))
+        log.info("opened shared cache of peer %s (device %d, %d layers)",
+                 name, pc.src_device, len(pc.views))
 
     def close_peer(self, name: str):
         pc = self.peers.pop(name, None)
@@ -103,22 +108,16 @@ class MigrationManager:
 
     def pull_blocks(self, src_name: str, src_blocks: List[int],
                     dst_blocks: List[int]):
-        """Copy blocks from the peer's cache into ours (all layers), on the
-        side stream; synchronizes before returning so the caller may
-        activate the sequence immediately."""
-        from xllm_service_amd import _ops
+        """Copy blocks from the peer cache into ours (all layers) on the side
+        stream; synchronizes before returning so the sequence can be
+        activated immediately after."""
         pc = self.peers[src_name]
         my_dev = self.engine.device.index or 0
         with torch.cuda.stream(self.stream):
             for layer, (kc, vc) in enumerate(self.engine.runner.kv_caches):
-                pk, pv = pc.ptrs[layer]
-                _ops.migrate_blocks_from_ptr(kc, pk, pc.src_device, my_dev,
-                                             list(src_blocks), list(dst_blocks))
-                _ops.migrate_blocks_from_ptr(vc, pv, pc.src_device, my_dev,
-                                             list(src_blocks), list(dst_blocks))
+                sk, sv = pc.views[layer]
+                ops.migrate_blocks_peer(kc, my_dev, sk, pc.src_device,
+                                        src_blocks, dst_blocks)
+                ops.migrate_blocks_peer(vc, my_dev, sv, pc.src_device,
+                                        src_blocks, dst_blocks)
         self.stream.synchronize()
-
-
-def migrate_in_xgmi(engine, peer_meta, src_blocks, dst_blocks):
-    raise RuntimeError(
-        "migrate_in_xgmi requires the worker's MigrationManager path")

@@ -179,7 +179,11 @@ class ModelRunner:
                         for s in plan.decodes)):
             return self._execute_decode_graph(plan, bm)
         input_ids, positions, meta = self._build_batch(plan, bm)
-        hidden = self.model(input_ids, positions, self.kv_caches, meta)
+        inputs_embeds = None
+        if any(sp.seq.mm_embeds is not None for sp in plan.prefills):
+            inputs_embeds = self._merge_mm_embeds(plan, input_ids, meta)
+        hidden = self.model(input_ids, positions, self.kv_caches, meta,
+                            inputs_embeds=inputs_embeds)
 
         # rows that need logits: last token of each COMPLETED prefill chunk,
         # plus every decode row
@@ -199,6 +203,28 @@ class ModelRunner:
         logits = self.model.compute_logits(sel)
         tokens = self._sample(logits, seqs)
         return {seq.request_id: tok for seq, tok in zip(seqs, tokens)}
+
+    def _merge_mm_embeds(self, plan: StepPlan, input_ids, meta):
+        """Replace image-placeholder rows of the token embeddings with the
+        per-sequence vision embeddings (EPD E->P handoff). Chunked prefill
+        is handled by counting placeholders before the chunk start."""
+        embeds = self.model.embed(input_ids)
+        off = 0
+        for i, sp in enumerate(plan.prefills):
+            seq = sp.seq
+            chunk = slice(int(meta.cu_q[i]), int(meta.cu_q[i + 1]))
+            if seq.mm_embeds is None:
+                continue
+            ph = seq.mm_placeholder
+            mm = seq.mm_embeds.to(embeds.device, embeds.dtype)
+            n_before = sum(1 for t in seq.prompt_token_ids[:sp.chunk_start]
+                           if t == ph)
+            chunk_ids = input_ids[chunk]
+            mask = chunk_ids == ph
+            n_here = int(mask.sum())
+            if n_here:
+                embeds[chunk][mask] = mm[n_before:n_before + n_here]
+        return embeds
 
     def _execute_decode_graph(self, plan: StepPlan, bm) -> Dict[str, int]:
         import numpy as np

@@ -37,6 +37,10 @@ class Sequence:
     block_table: List[int] = field(default_factory=list)
     num_computed_tokens: int = 0          # prompt tokens already prefilled
     preempt_count: int = 0
+    # multimodal: pre-computed image embeddings substituted at placeholder
+    # token positions during prefill (EPD E->P handoff)
+    mm_embeds: Optional["object"] = None      # torch.Tensor [n, hidden]
+    mm_placeholder: Optional[int] = None
     # PD-disaggregation: set on a decode instance receiving a migrated prefill
     migrated_in: bool = False
     # PD-disaggregation: keep KV blocks alive after finish (prefill side

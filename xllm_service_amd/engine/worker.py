@@ -39,6 +39,38 @@ from xllm_service_amd.utils import msgrpc
 from .engine import LLMEngine
 from .sampling import SamplingParams
 
+
+class VisionEncoder:
+    """Stage-E vision tower runner (ENCODE instances; also used in-process
+    by colocated DEFAULT instances serving multimodal models)."""
+
+    def __init__(self, model_name: str, device: Optional[str], seed: int = 0):
+        from xllm_service_amd.models.config import get_config
+        from xllm_service_amd.models.qwen2_vl import Qwen2VisionTransformer
+        self.cfg = get_config(model_name)
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = torch.device(device)
+        dtype = torch.bfloat16 if self.device.type == "cuda" else torch.float32
+        self.tower = Qwen2VisionTransformer(self.cfg, dtype).to(self.device)
+        self.tower.random_init(seed)
+        self.tower.eval()
+
+    @torch.inference_mode()
+    def encode(self, images: List[Dict[str, Any]]) -> torch.Tensor:
+        """images: [{grid_h, grid_w, seed}] with synthetic deterministic
+        pixels (offline environment); returns [sum tokens, hidden]."""
+        outs = []
+        for img in images:
+            gh, gw = int(img["grid_h"]), int(img["grid_w"])
+            gen = torch.Generator().manual_seed(int(img.get("seed", 0)))
+            patches = torch.randn(gh * gw, self.tower.patch_dim,
+                                  generator=gen) * 0.5
+            patches = patches.to(self.device, next(
+                self.tower.parameters()).dtype)
+            outs.append(self.tower(patches, gh, gw))
+        return torch.cat(outs, dim=0)
+
 log = logging.getLogger("xllm.worker")
 
 
@@ -87,6 +119,7 @@ class Worker:
         self.peers: Dict[str, InstanceMetaInfo] = {}
         self.peer_conns: Dict[str, msgrpc.Connection] = {}
         self.mig = None  # decode-side xGMI MigrationManager (lazy)
+        self.encoder: Optional[VisionEncoder] = None  # ENCODE stage / VL
         # requests this PREFILL instance must migrate after the first token:
         # rid -> dict(routing/params)
         self.pending_migration: Dict[str, Dict[str, Any]] = {}
@@ -106,12 +139,18 @@ class Worker:
     # ------------------------------------------------------------------ setup
     async def start(self):
         self._loop = asyncio.get_running_loop()
-        # engine init is slow (weights); do it off-loop
-        self.engine = await self._loop.run_in_executor(None, self._make_engine)
-        self._engine_thread = threading.Thread(target=self._engine_loop,
-                                               daemon=True,
-                                               name=f"engine-{self.name}")
-        self._engine_thread.start()
+        if self.itype == InstanceType.ENCODE:
+            # stage-E instance: vision tower only, no LM engine
+            self.encoder = await self._loop.run_in_executor(
+                None, lambda: VisionEncoder(self.model, self.device))
+        else:
+            # engine init is slow (weights); do it off-loop
+            self.engine = await self._loop.run_in_executor(
+                None, self._make_engine)
+            self._engine_thread = threading.Thread(target=self._engine_loop,
+                                                   daemon=True,
+                                                   name=f"engine-{self.name}")
+            self._engine_thread.start()
 
         self.rpc_server = msgrpc.Server(lambda conn: self, self.rpc_host,
                                         self.rpc_port)
@@ -151,6 +190,7 @@ class Worker:
             cluster_ids=[dev] if dev >= 0 else [],
             num_kv_blocks=self.engine.block_manager.num_blocks
             if self.engine else 0,
+            dp_size=1,
             block_size=16, model=self.model,
             incarnation_id=self.incarnation,
             k_cache_ids=list(range(self.engine.cfg.num_layers))
@@ -285,10 +325,31 @@ class Worker:
                                  routing: dict, offline: bool = False,
                                  multimodal: Optional[dict] = None):
         routing = routing or {}
+        # ---- stage E: run the vision tower, then hand off to prefill ------
+        if self.itype == InstanceType.ENCODE:
+            await self._encode_and_forward(service_request_id, token_ids,
+                                           params, routing, offline,
+                                           multimodal)
+            return
         sp = params_from_dict(params)
         self.req_meta[service_request_id] = dict(
             params=params, routing=routing, prompt_len=len(token_ids),
             offline=offline, multimodal=multimodal)
+        mm_embeds = None
+        if multimodal:
+            if multimodal.get("embeds_b") is not None:
+                import numpy as np
+                raw = np.frombuffer(multimodal["embeds_b"], dtype=np.uint8)
+                mm_embeds = torch.from_numpy(raw.copy()).view(
+                    torch.bfloat16 if multimodal.get("dtype") == "bfloat16"
+                    else torch.float32).reshape(multimodal["embeds_shape"])
+            elif multimodal.get("images"):
+                # colocated multimodal: run the vision tower in-process
+                if self.encoder is None:
+                    self.encoder = await self._loop.run_in_executor(
+                        None, lambda: VisionEncoder(self.model, self.device))
+                mm_embeds = await self._loop.run_in_executor(
+                    None, lambda: self.encoder.encode(multimodal["images"]))
         decode_name = routing.get("decode_name")
         do_migrate = (self.itype == InstanceType.PREFILL
                       and decode_name and decode_name != self.name)
@@ -303,12 +364,44 @@ class Worker:
             self._post_to_engine(
                 lambda: self.engine.add_request(
                     service_request_id, token_ids, first_sp,
-                    priority=1 if offline else 0, hold_blocks=True))
+                    priority=1 if offline else 0, hold_blocks=True,
+                    mm_embeds=mm_embeds))
         else:
             self._post_to_engine(
                 lambda: self.engine.add_request(
                     service_request_id, token_ids, sp,
-                    priority=1 if offline else 0))
+                    priority=1 if offline else 0, mm_embeds=mm_embeds))
+
+    async def _encode_and_forward(self, rid, token_ids, params, routing,
+                                  offline, multimodal):
+        target = routing.get("prefill_name") or routing.get("decode_name")
+        try:
+            embeds = await self._loop.run_in_executor(
+                None, lambda: self.encoder.encode(
+                    (multimodal or {}).get("images", [])))
+            conn = await self._peer_conn(target)
+            if conn is None:
+                raise RuntimeError(f"prefill peer {target} unreachable")
+            payload_mm = dict(
+                embeds_b=embeds.cpu().view(torch.uint8).numpy().tobytes()
+                if embeds.dtype != torch.float32
+                else embeds.cpu().numpy().tobytes(),
+                dtype=str(embeds.dtype).split(".")[-1],
+                embeds_shape=list(embeds.shape))
+            await conn.notify("execute_request",
+                              service_request_id=rid, token_ids=token_ids,
+                              params=params, routing=routing,
+                              offline=offline, multimodal=payload_mm)
+        except Exception as e:
+            log.warning("encode stage failed for %s: %s", rid, e)
+            if self.master_conn:
+                try:
+                    await self.master_conn.notify("generations", gens=[dict(
+                        service_request_id=rid, token_ids=[], finished=True,
+                        finish_reason="abort",
+                        error=f"vision encode failed: {e}")])
+                except Exception:
+                    pass
 
     def on_abort_request(self, conn, service_request_id: str):
         self.pending_migration.pop(service_request_id, None)
@@ -482,6 +575,15 @@ class Worker:
                 if self.master_conn is None or self.master_conn.closed.is_set():
                     continue
             try:
+                if self.engine is None:  # ENCODE stage: trivial heartbeat
+                    await self.master_conn.call(
+                        "heartbeat", timeout=5.0, name=self.name,
+                        incarnation=self.incarnation,
+                        load=dict(waiting_requests_num=0,
+                                  running_requests_num=0,
+                                  gpu_cache_usage_perc=0.0),
+                        latency={}, kv_stored=[], kv_removed=[])
+                    continue
                 ev = self.engine.block_manager.events.drain()
                 st = self.engine.stats
                 await self.master_conn.call(

@@ -22,6 +22,7 @@ class ModelConfig:
     hidden_act: str = "silu"
     # vision tower (qwen2_vl only)
     vision: dict = field(default_factory=dict)
+    image_pad_token_id: int = -1
 
     @property
     def q_size(self) -> int:
@@ -57,9 +58,17 @@ PRESETS = {
         name="qwen2-vl-7b", architecture="qwen2_vl", vocab_size=152064,
         hidden_size=3584, intermediate_size=18944, num_layers=28,
         num_heads=28, num_kv_heads=4, head_dim=128, rope_theta=1000000.0,
-        max_position=8192,
+        max_position=8192, image_pad_token_id=151655,
         vision=dict(depth=32, embed_dim=1280, num_heads=16, patch_size=14,
                     spatial_merge_size=2, out_hidden_size=3584)),
+    # tiny multimodal config for CPU EPD tests
+    "qwen2-vl-tiny": ModelConfig(
+        name="qwen2-vl-tiny", architecture="qwen2_vl", vocab_size=1024,
+        hidden_size=256, intermediate_size=512, num_layers=2, num_heads=4,
+        num_kv_heads=2, head_dim=64, rope_theta=10000.0, max_position=2048,
+        image_pad_token_id=9,
+        vision=dict(depth=2, embed_dim=64, num_heads=4, patch_size=14,
+                    spatial_merge_size=2, out_hidden_size=256)),
 }
 
 

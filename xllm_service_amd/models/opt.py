@@ -79,12 +79,15 @@ class OPTForCausalLM(nn.Module):
 
     def random_init(self, seed: int = 0):
         gen = torch.Generator().manual_seed(seed)
-        std = 0.02
         for name, p in self.named_parameters():
-            if "ln" in name or "bias" in name:
-                continue
             with torch.no_grad():
-                p.copy_((torch.randn(p.shape, generator=gen) * std).to(p.dtype))
+                if "ln" in name and name.endswith("weight"):
+                    p.fill_(1.0)
+                elif "ln" in name or "bias" in name:
+                    p.zero_()
+                else:
+                    p.copy_((torch.randn(p.shape, generator=gen) * 0.02
+                             ).to(p.dtype))
 
     def forward(self, input_ids, positions, kv_caches: List[Tuple],
                 meta: AttnMetadata, inputs_embeds=None):

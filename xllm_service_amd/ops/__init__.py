@@ -162,6 +162,17 @@ def enable_peer_access(device: int, peer: int):
     _ops.enable_peer_access(device, peer)
 
 
+def mfma_gemm(a, b, bias=None):
+    """C = a @ b.T (+ bias): the hand-written vision-encoder GEMM on GPU;
+    torch on CPU."""
+    if a.is_cuda:
+        _require_ext()
+        return _ops.mfma_gemm(a, b, bias)
+    return torch.nn.functional.linear(a.float(), b.float(),
+                                      bias.float() if bias is not None
+                                      else None).to(a.dtype)
+
+
 def mfma_probe_16x16x32(a, b):
     _require_ext()
     return _ops.mfma_probe_16x16x32(a, b)

@@ -143,6 +143,43 @@ def build_app(master) -> FastAPI:
             return JSONResponse(
                 {"error": {"message": f"chat template error: {e}",
                            "type": "invalid_request_error"}}, status_code=400)
+        # multimodal: collect image parts; offline environment accepts the
+        # synthetic form {"type": "image", "grid": [gh, gw], "seed": n}
+        images = []
+        for msg in body.messages:
+            content = msg.get("content")
+            if not isinstance(content, list):
+                continue
+            for part in content:
+                if part.get("type") == "image":
+                    gh, gw = part.get("grid", [28, 28])
+                    images.append(dict(grid_h=int(gh), grid_w=int(gw),
+                                       seed=int(part.get("seed", 0))))
+                elif part.get("type") == "image_url":
+                    return JSONResponse(
+                        {"error": {"message":
+                                   "image_url fetch unavailable offline; "
+                                   "use {'type':'image','grid':[h,w]}",
+                                   "type": "invalid_request_error"}},
+                        status_code=400)
+        multimodal = None
+        if images:
+            try:
+                from xllm_service_amd.models.config import get_config
+                mcfg = get_config(body.model or master.model_id)
+                merge = mcfg.vision.get("spatial_merge_size", 2)
+                pad_id = mcfg.image_pad_token_id
+                assert pad_id >= 0
+            except Exception:
+                return JSONResponse(
+                    {"error": {"message": "model is not multimodal",
+                               "type": "invalid_request_error"}},
+                    status_code=400)
+            n_img_tokens = sum((im["grid_h"] // merge) * (im["grid_w"] // merge)
+                               for im in images)
+            # vision tokens precede the text tokens
+            token_ids = [pad_id] * n_img_tokens + token_ids
+            multimodal = {"images": images}
         params = _sampling_dict(body)
         if body.max_completion_tokens is not None:
             params["max_tokens"] = body.max_completion_tokens
@@ -152,7 +189,7 @@ def build_app(master) -> FastAPI:
             service_request_id=make_request_id("chatcmpl"),
             kind="chat", model=body.model or master.model_id,
             stream=body.stream, token_ids=token_ids, prompt_text=prompt_text,
-            params=params, offline=body.offline)
+            params=params, offline=body.offline, multimodal=multimodal)
         return await _run(req, request, body.stream, chat=True)
 
     def _sampling_dict(body) -> Dict[str, Any]:

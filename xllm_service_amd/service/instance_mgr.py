@@ -168,18 +168,29 @@ class InstanceMgr:
             # link fan-out: P<->D peers exchange cluster info
             links: List[Tuple[Instance, InstanceMetaInfo]] = []
             it = inst.itype
-            if it in (InstanceType.PREFILL, InstanceType.MIX,
-                      InstanceType.ENCODE):
+            if it == InstanceType.ENCODE:
+                for other in self.instances.values():
+                    if other.name != meta.name and other.itype in (
+                            InstanceType.DEFAULT, InstanceType.PREFILL,
+                            InstanceType.MIX):
+                        links.append((inst, other.meta))
+                        links.append((other, meta))
+            elif it in (InstanceType.PREFILL, InstanceType.MIX):
                 for other in self.instances.values():
                     if other.name != meta.name and other.itype in (
                             InstanceType.DECODE, InstanceType.MIX,
-                            InstanceType.PREFILL):
+                            InstanceType.PREFILL, InstanceType.ENCODE):
                         links.append((other, meta))
                         links.append((inst, other.meta))
             elif it == InstanceType.DECODE:
                 for other in self.instances.values():
                     if other.name != meta.name and other.itype in (
                             InstanceType.PREFILL, InstanceType.MIX):
+                        links.append((inst, other.meta))
+                        links.append((other, meta))
+            elif it == InstanceType.DEFAULT:
+                for other in self.instances.values():
+                    if other.name != meta.name and                             other.itype == InstanceType.ENCODE:
                         links.append((inst, other.meta))
                         links.append((other, meta))
             done: List[Tuple[Instance, InstanceMetaInfo]] = []

@@ -64,11 +64,20 @@ class ServiceScheduler:
             chat_template_kwargs=chat_template_kwargs)
         return prompt, self.tokenizer.encode(prompt)
 
+    _rr_encode = 0
+
     def schedule(self, req: ServiceRequest) -> None:
-        """Bind req to an instance pair; raises SchedulerError if none."""
+        """Bind req to an instance pair (plus the E-stage instance for
+        multimodal requests); raises SchedulerError if none."""
         pair = self.policy.select_instances_pair(req.token_ids)
         if not pair.ok:
             raise SchedulerError("no available instances")
+        if req.multimodal and req.multimodal.get("images"):
+            encodes = self.mgr.schedulable_encodes()
+            if encodes:  # E/P/D three-stage split; else colocated vision
+                ServiceScheduler._rr_encode += 1
+                req.encode_name = encodes[
+                    ServiceScheduler._rr_encode % len(encodes)].name
         if pair.prefill is not None:
             req.prefill_name = pair.prefill.name
             req.prefill_incarnation = pair.prefill.meta.incarnation_id
@@ -88,8 +97,9 @@ class ServiceScheduler:
         req.scheduled_at = time.monotonic()
 
     async def dispatch(self, req: ServiceRequest) -> None:
-        """Forward the scheduled request to its prefill instance."""
-        target_name = req.prefill_name or req.decode_name
+        """Forward the request to its first-stage instance (E for
+        multimodal when an encode pool exists, else P)."""
+        target_name = req.encode_name or req.prefill_name or req.decode_name
         inst = self.mgr.get(target_name)
         if inst is None or inst.conn is None:
             raise SchedulerError(f"instance {target_name} unavailable")
