@@ -73,3 +73,23 @@ def test_gpu_prefix_cache_and_chunked_prefill():
     from xllm_service_amd.engine.sequence import Sequence
     probe = Sequence("probe", shared + [9, 9, 9], SamplingParams())
     assert eng.block_manager.match_prefix(probe) >= 64
+
+
+def test_gpu_qwen2_vl_engine():
+    """Multimodal engine on GPU: vision embeds + paged LM decode."""
+    from xllm_service_amd.engine.worker import VisionEncoder
+    eng = LLMEngine("qwen2-vl-debug", device="cuda:0", max_kv_blocks=128,
+                    seed=4, enable_graphs=False)
+    enc = VisionEncoder("qwen2-vl-debug", "cuda:0", seed=4)
+    mm = enc.encode([dict(grid_h=4, grid_w=4, seed=1)])
+    assert mm.shape == (4, 512)
+    cfg = get_config("qwen2-vl-debug")
+    prompt = [9] * 4 + list(range(100, 120))
+    eng.add_request("vl", prompt,
+                    SamplingParams(max_tokens=5, ignore_eos=True),
+                    mm_embeds=mm)
+    toks = []
+    while eng.has_work():
+        for o in eng.step():
+            toks.extend(o.new_token_ids)
+    assert len(toks) == 5

@@ -158,6 +158,19 @@ def test_paged_attn_prefill(dev, ctx):
     _assert_close(got, want, atol=3e-2, label=f"prefill ctx={ctx}")
 
 
+@pytest.mark.parametrize("M,N,K", [(128, 128, 64), (100, 1280, 1184),
+                                   (256, 512, 1024), (64, 3584, 5120)])
+def test_mfma_gemm(dev, M, N, K):
+    torch.manual_seed(M + N)
+    a = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+    b = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
+    bias = torch.randn(N, device=dev, dtype=torch.bfloat16)
+    got = ops.mfma_gemm(a, b, bias)
+    want = (a.float() @ b.float().t()) + bias.float()
+    _assert_close(got, want, atol=0.05 + 0.02 * (K / 1024),
+                  label=f"mfma_gemm {M}x{N}x{K}")
+
+
 def test_greedy_sample(dev):
     torch.manual_seed(0)
     logits = torch.randn(33, 128256, device=dev, dtype=torch.bfloat16)

@@ -72,3 +72,38 @@ class RowParallelLinear(nn.Module):
         if self.bias is not None:
             y = y + self.bias
         return y
+
+
+def shard_llama_state_dict(full_sd, cfg, tp: int, rank: int):
+    """Shard a full (tp=1) Llama-family state dict for TP rank `rank`.
+
+    Column-parallel weights (qkv, gate_up, lm_head) split by output rows per
+    constituent; row-parallel weights (o_proj, down) split by input columns.
+    Used by tests (TP == single-GPU equivalence) and by checkpoint loading.
+    """
+    import torch
+    q_sz = cfg.q_size
+    kv_sz = cfg.kv_size
+    out = {}
+    for k, v in full_sd.items():
+        if "qkv_proj" in k:  # weight or bias: rows [q | k | v]
+            q, kk, vv = torch.split(v, [q_sz, kv_sz, kv_sz], dim=0)
+            parts = []
+            for t, sz in ((q, q_sz), (kk, kv_sz), (vv, kv_sz)):
+                s = sz // tp
+                parts.append(t[rank * s:(rank + 1) * s])
+            out[k] = torch.cat(parts, dim=0)
+        elif "gate_up" in k:
+            g, u = v.chunk(2, dim=0)
+            s = g.shape[0] // tp
+            out[k] = torch.cat([g[rank * s:(rank + 1) * s],
+                                u[rank * s:(rank + 1) * s]], dim=0)
+        elif "lm_head" in k:
+            s = v.shape[0] // tp
+            out[k] = v[rank * s:(rank + 1) * s]
+        elif ("o_proj.weight" in k) or ("down.weight" in k):
+            s = v.shape[1] // tp
+            out[k] = v[:, rank * s:(rank + 1) * s]
+        else:
+            out[k] = v
+    return out

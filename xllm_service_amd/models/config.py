@@ -61,6 +61,14 @@ PRESETS = {
         max_position=8192, image_pad_token_id=151655,
         vision=dict(depth=32, embed_dim=1280, num_heads=16, patch_size=14,
                     spatial_merge_size=2, out_hidden_size=3584)),
+    # small multimodal config with head_dim 128 for GPU tests
+    "qwen2-vl-debug": ModelConfig(
+        name="qwen2-vl-debug", architecture="qwen2_vl", vocab_size=1024,
+        hidden_size=512, intermediate_size=1024, num_layers=2, num_heads=4,
+        num_kv_heads=2, head_dim=128, rope_theta=10000.0, max_position=2048,
+        image_pad_token_id=9,
+        vision=dict(depth=2, embed_dim=256, num_heads=4, patch_size=14,
+                    spatial_merge_size=2, out_hidden_size=512)),
     # tiny multimodal config for CPU EPD tests
     "qwen2-vl-tiny": ModelConfig(
         name="qwen2-vl-tiny", architecture="qwen2_vl", vocab_size=1024,
