@@ -80,7 +80,10 @@ def test_tp2_matches_single(tmp_path):
     sd_path = str(tmp_path / "full.pt")
     out_path = str(tmp_path / "tp_out.pt")
     torch.save(sd, sd_path)
-    port = 29641
+    import socket
+    with socket.socket() as sock:   # pick a free port (avoid suite clashes)
+        sock.bind(("127.0.0.1", 0))
+        port = sock.getsockname()[1]
     mp.spawn(_tp_worker, args=(2, port, sd_path, out_path), nprocs=2,
              join=True)
     got = torch.load(out_path)
