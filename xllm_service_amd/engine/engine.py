@@ -27,6 +27,7 @@ class StepOutput:
     new_token_ids: List[int]
     finished: bool
     finish_reason: Optional[str] = None   # "stop" | "length" | "abort"
+    logprobs: Optional[List[dict]] = None  # per new token (when requested)
     num_prompt_tokens: int = 0
     num_output_tokens: int = 0
     first_token: bool = False
@@ -198,16 +199,19 @@ class LLMEngine:
         new_tokens = self.runner.execute(plan, self.block_manager)
 
         outputs: List[StepOutput] = []
-        for rid, tok in new_tokens.items():
+        for rid, (tok, lp) in new_tokens.items():
             seq = self.seqs.get(rid)
             if seq is None:
                 continue
             first = not seq.output_token_ids
             seq.append_token(tok)
+            if lp is not None:
+                seq.cumulative_logprob += lp["token_logprob"]
             finished = seq.check_finish()
             outputs.append(StepOutput(
                 request_id=rid,
                 new_token_ids=[tok],
+                logprobs=[lp] if lp is not None else None,
                 finished=finished,
                 finish_reason=(
                     "stop" if seq.status == SeqStatus.FINISHED_STOP else

@@ -135,6 +135,26 @@ class ModelRunner:
         return input_ids, pos_t, meta
 
     # ---- sampling -----------------------------------------------------------
+    def _logprobs(self, logits: torch.Tensor, seqs: List[Sequence],
+                  tokens: List[int]):
+        """Per-token logprob + top-N alternatives for requesting seqs."""
+        if not any(s.params.logprobs is not None for s in seqs):
+            return {}
+        lp = torch.log_softmax(logits.float(), dim=-1)
+        out = {}
+        for i, seq in enumerate(seqs):
+            n = seq.params.logprobs
+            if n is None:
+                continue
+            row = lp[i]
+            chosen = float(row[tokens[i]])
+            top = {}
+            if n > 0:
+                vals, idx = torch.topk(row, min(n, row.numel()))
+                top = {int(t): float(v) for t, v in zip(idx, vals)}
+            out[seq.request_id] = {"token_logprob": chosen, "top": top}
+        return out
+
     def _sample(self, logits: torch.Tensor, seqs: List[Sequence]) -> List[int]:
         greedy = all(s.params.greedy for s in seqs)
         if greedy:
@@ -202,7 +222,9 @@ class ModelRunner:
         sel = hidden[torch.tensor(rows, dtype=torch.long, device=hidden.device)]
         logits = self.model.compute_logits(sel)
         tokens = self._sample(logits, seqs)
-        return {seq.request_id: tok for seq, tok in zip(seqs, tokens)}
+        lps = self._logprobs(logits, seqs, tokens)
+        return {seq.request_id: (tok, lps.get(seq.request_id))
+                for seq, tok in zip(seqs, tokens)}
 
     def _merge_mm_embeds(self, plan: StepPlan, input_ids, meta):
         """Replace image-placeholder rows of the token embeddings with the
@@ -245,7 +267,6 @@ class ModelRunner:
             bt_rows.append(gr.block_row(seq))
         logits = gr.run(input_ids, positions, slots, seq_lens, bt_rows)
         tokens = self._sample(logits, seqs)
-        out = {}
-        for seq, tok in zip(seqs, tokens):
-            out[seq.request_id] = tok
-        return out
+        lps = self._logprobs(logits, seqs, tokens)
+        return {seq.request_id: (tok, lps.get(seq.request_id))
+                for seq, tok in zip(seqs, tokens)}

@@ -260,7 +260,24 @@ class Worker:
             except queue.Empty:
                 pass
             if self.engine.has_work():
-                outs = self.engine.step()
+                try:
+                    outs = self.engine.step()
+                except Exception:
+                    # a poisoned batch must not kill the worker: drop the
+                    # running sequences with an abort so the master errors
+                    # them out instead of hanging the clients
+                    log.exception("engine step failed; aborting running seqs")
+                    from .engine import StepOutput
+                    bad = [s.request_id
+                           for s in list(self.engine.scheduler.running)
+                           ] + [s.request_id
+                                for s in list(self.engine.scheduler.waiting)]
+                    outs = []
+                    for rid in bad:
+                        self.engine.abort_request(rid)
+                        outs.append(StepOutput(
+                            request_id=rid, new_token_ids=[], finished=True,
+                            finish_reason="abort"))
                 if outs:
                     self._out_q.put(outs)
                 did = True
@@ -481,6 +498,7 @@ class Worker:
                     service_request_id=rid,
                     token_ids=o.new_token_ids,
                     finished=o.finished,
+                    logprobs=o.logprobs,
                     finish_reason=o.finish_reason,
                     finished_on_prefill=(
                         o.first_token and

@@ -170,3 +170,27 @@ class RegistryClient:
     async def close(self):
         if self.conn:
             await self.conn.close()
+
+
+def main():
+    """Standalone registry daemon: python -m xllm_service_amd.registry.server
+    (multi-replica deployments keep the registry outside any master)."""
+    import argparse
+    import logging
+    ap = argparse.ArgumentParser(description="xllm-service-amd registry")
+    ap.add_argument("--host", default="127.0.0.1")
+    ap.add_argument("--port", type=int, default=12379)
+    args = ap.parse_args()
+    logging.basicConfig(level=logging.INFO)
+
+    async def run():
+        svc = RegistryService(args.host, args.port)
+        port = await svc.start()
+        logging.info("registry listening on %s:%d", args.host, port)
+        await asyncio.Event().wait()
+
+    asyncio.run(run())
+
+
+if __name__ == "__main__":
+    main()

@@ -93,6 +93,29 @@ def build_app(master) -> FastAPI:
                           "owned_by": "xllm-service-amd"}
                          for m in master.served_models()]}
 
+    @app.post("/admin/reload_flags")
+    async def reload_flags(request: Request):
+        """Runtime-reloadable knobs (reference: brpc-reloadable target_ttft /
+        target_tpot gflags)."""
+        body = await request.json()
+        changed = {}
+        from .policies import SloAwarePolicy
+        pol = scheduler().policy
+        if "target_ttft_ms" in body:
+            master.opts.target_ttft_ms = float(body["target_ttft_ms"])
+            if isinstance(pol, SloAwarePolicy):
+                pol.target_ttft_ms = master.opts.target_ttft_ms
+            changed["target_ttft_ms"] = master.opts.target_ttft_ms
+        if "target_tpot_ms" in body:
+            master.opts.target_tpot_ms = float(body["target_tpot_ms"])
+            if isinstance(pol, SloAwarePolicy):
+                pol.target_tpot_ms = master.opts.target_tpot_ms
+            changed["target_tpot_ms"] = master.opts.target_tpot_ms
+        if "enable_request_trace" in body:
+            master.tracer.enabled = bool(body["enable_request_trace"])
+            changed["enable_request_trace"] = master.tracer.enabled
+        return {"reloaded": changed}
+
     @app.post("/v1/embeddings")
     async def embeddings():
         return JSONResponse({"error": {"message": "not support embeddings",
@@ -196,7 +219,8 @@ def build_app(master) -> FastAPI:
         d = dict(temperature=body.temperature, top_p=body.top_p,
                  top_k=body.top_k, max_tokens=body.max_tokens or 16,
                  min_tokens=body.min_tokens, seed=body.seed,
-                 ignore_eos=body.ignore_eos)
+                 ignore_eos=body.ignore_eos,
+                 logprobs=getattr(body, "logprobs", None))
         if body.stream_options:
             d["include_usage"] = bool(
                 body.stream_options.get("include_usage"))

@@ -208,6 +208,11 @@ class Master:
     async def stop(self):
         for t in self._tasks:
             t.cancel()
+        if self.registry and self._lease_id:
+            try:  # release the service key + master election key promptly
+                await self.registry.revoke_lease(self._lease_id)
+            except Exception:
+                pass
         if self._uvicorn:
             self._uvicorn.should_exit = True
         if self.instance_mgr:

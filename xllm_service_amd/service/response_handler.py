@@ -131,10 +131,20 @@ class ResponseHandler:
                 "id": tc.id, "type": "function",
                 "function": {"name": tc.name, "arguments": tc.arguments}}
                 for tc in tool_calls]
+        choice = {"index": 0, "message": msg, "finish_reason": finish}
+        raw_lps = getattr(req, "_collected_logprobs", [])
+        if raw_lps:
+            choice["logprobs"] = {"content": [
+                {"token": self.tokenizer.decode([t]),
+                 "logprob": lp["token_logprob"] if lp else None,
+                 "top_logprobs": [
+                     {"token": self.tokenizer.decode([tt]), "logprob": vv}
+                     for tt, vv in (lp.get("top") or {}).items()]}
+                for t, lp in zip(token_ids, raw_lps)]}
         return {
             "id": req.service_request_id, "object": "chat.completion",
             "created": int(req.created), "model": req.model,
-            "choices": [{"index": 0, "message": msg, "finish_reason": finish}],
+            "choices": [choice],
             "usage": usage}
 
     # ------------------------------------------------------ completion paths
@@ -183,16 +193,22 @@ class ResponseHandler:
         token_ids, usage, finish, err = await self._collect(req)
         if err:
             return {"error": {"message": err, "type": "server_error"}}
+        choice = {"index": 0, "text": self.tokenizer.decode(token_ids),
+                  "finish_reason": finish}
+        lp = self._completion_logprobs(
+            token_ids, getattr(req, "_collected_logprobs", []))
+        if lp is not None:
+            choice["logprobs"] = lp
         return {
             "id": req.service_request_id, "object": "text_completion",
             "created": int(req.created), "model": req.model,
-            "choices": [{"index": 0, "text": self.tokenizer.decode(token_ids),
-                         "finish_reason": finish}],
+            "choices": [choice],
             "usage": usage}
 
     # ---------------------------------------------------------------- common
     async def _collect(self, req: ServiceRequest):
         token_ids = []
+        logprobs = []
         usage = {"prompt_tokens": 0, "completion_tokens": 0, "total_tokens": 0}
         finish = "stop"
         while True:
@@ -201,10 +217,27 @@ class ResponseHandler:
             if gen.error:
                 return token_ids, usage, finish, gen.error
             token_ids.extend(gen.token_ids)
+            if gen.logprobs:
+                logprobs.extend(gen.logprobs)
             if gen.finished:
                 finish = gen.finish_reason or "stop"
                 usage = {"prompt_tokens": gen.usage_prompt_tokens,
                          "completion_tokens": gen.usage_completion_tokens,
                          "total_tokens": gen.usage_prompt_tokens +
                          gen.usage_completion_tokens}
+                req._collected_logprobs = logprobs
                 return token_ids, usage, finish, None
+
+    def _completion_logprobs(self, token_ids, logprobs):
+        """OpenAI completion-style logprobs block."""
+        if not logprobs:
+            return None
+        return {
+            "tokens": [self.tokenizer.decode([t]) for t in token_ids],
+            "token_logprobs": [lp["token_logprob"] if lp else None
+                               for lp in logprobs],
+            "top_logprobs": [
+                {self.tokenizer.decode([t]): v
+                 for t, v in (lp.get("top") or {}).items()} if lp else None
+                for lp in logprobs],
+        }

@@ -21,8 +21,11 @@ class RequestTracer:
             self._fh = open(path, "a", buffering=1)
 
     def trace(self, service_request_id: str, direction: str, data: Any):
-        if not self.enabled or self._fh is None:
+        if not self.enabled:
             return
+        if self._fh is None:  # enabled at runtime via /admin/reload_flags
+            os.makedirs(os.path.dirname(self.path) or ".", exist_ok=True)
+            self._fh = open(self.path, "a", buffering=1)
         rec = {"timestamp": time.time(),
                "service_request_id": service_request_id,
                "direction": direction, "data": data}

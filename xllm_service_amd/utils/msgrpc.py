@@ -75,7 +75,7 @@ class Connection:
                 if ln > MAX_FRAME:
                     raise RpcError(f"frame too large: {ln}")
                 data = await self.reader.readexactly(ln)
-                msg = msgpack.unpackb(data, raw=False)
+                msg = msgpack.unpackb(data, raw=False, strict_map_key=False)
                 kind = msg[0]
                 if kind == 0:
                     asyncio.create_task(self._handle_request(msg[1], msg[2],
