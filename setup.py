@@ -32,6 +32,7 @@ ext_modules = [
                 "paged_attn_prefill.hip",
                 "sampling.hip",
                 "gemm.hip",
+                "skinny_gemm.hip",
             )
         ],
         extra_compile_args={

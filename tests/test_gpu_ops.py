@@ -171,6 +171,20 @@ def test_mfma_gemm(dev, M, N, K):
                   label=f"mfma_gemm {M}x{N}x{K}")
 
 
+@pytest.mark.parametrize("M,N,K", [(1, 6144, 4096), (64, 6144, 4096),
+                                   (64, 4096, 4096), (100, 4096, 14336),
+                                   (33, 512, 1536)])
+def test_skinny_gemm(dev, M, N, K):
+    torch.manual_seed(M)
+    a = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+    b = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
+    bias = torch.randn(N, device=dev, dtype=torch.bfloat16)
+    got = ops.skinny_gemm(a, b, bias)
+    want = torch.nn.functional.linear(a.float(), b.float(), bias.float())
+    _assert_close(got, want, atol=0.05 + 0.02 * (K / 1024),
+                  label=f"skinny {M}x{N}x{K}")
+
+
 def test_greedy_sample(dev):
     torch.manual_seed(0)
     logits = torch.randn(33, 128256, device=dev, dtype=torch.bfloat16)

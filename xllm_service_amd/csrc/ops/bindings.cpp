@@ -41,6 +41,9 @@ void launch_greedy_sample(long*, const unsigned short*, int, int, hipStream_t);
 void launch_mfma_gemm(unsigned short*, const unsigned short*,
                       const unsigned short*, const unsigned short*, int, int,
                       int, hipStream_t);
+void launch_skinny_gemm(unsigned short*, const unsigned short*,
+                        const unsigned short*, const unsigned short*, float*,
+                        int, int, int, hipStream_t);
 void launch_mfma_probe(float*, const unsigned short*, const unsigned short*,
                        hipStream_t);
 }  // namespace xllm
@@ -230,6 +233,23 @@ torch::Tensor mfma_gemm(torch::Tensor a, torch::Tensor b,
   return c;
 }
 
+torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor b,
+                          c10::optional<torch::Tensor> bias,
+                          torch::Tensor ws) {
+  // a: [M, K]; b: [N, K]; ws: fp32 [SK, ceil(M/16)*16, N]; returns [M, N]
+  CHECK_BF16_CUDA(a); CHECK_BF16_CUDA(b);
+  TORCH_CHECK(ws.dtype() == torch::kFloat && ws.is_cuda());
+  const int M = a.size(0), K = a.size(1), N = b.size(0);
+  TORCH_CHECK(a.size(1) == b.size(1) && M <= 128);
+  TORCH_CHECK(K % 32 == 0 && N % 4 == 0);
+  auto c = torch::empty({M, N}, a.options());
+  const unsigned short* bp = nullptr;
+  if (bias.has_value()) bp = u16c(bias.value());
+  xllm::launch_skinny_gemm(u16(c), u16c(a), u16c(b), bp,
+                           ws.data_ptr<float>(), M, N, K, cur_stream());
+  return c;
+}
+
 torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b) {
   CHECK_BF16_CUDA(a); CHECK_BF16_CUDA(b);
   TORCH_CHECK(a.size(0) == 16 && a.size(1) == 32);
@@ -366,6 +386,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("greedy_sample", &greedy_sample);
   m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
   m.def("mfma_gemm", &mfma_gemm);
+  m.def("skinny_gemm", &skinny_gemm);
   m.def("migrate_blocks_peer", &migrate_blocks_peer);
   m.def("ipc_get_handle", &ipc_get_handle);
   m.def("ipc_open_handle", &ipc_open_handle);

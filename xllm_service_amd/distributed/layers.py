@@ -13,6 +13,16 @@ import torch.nn as nn
 from . import parallel_state as ps
 
 
+def _maybe_skinny(x, weight, bias):
+    """Decode batches (rows <= 128) route to the split-K weight-streaming
+    kernel when the layer opted in (ops/skinny_gemm.hip); else hipBLASLt."""
+    from xllm_service_amd import ops as xops
+    if (x.is_cuda and 0 < x.shape[0] <= 128 and xops.HAS_EXT
+            and weight.shape[1] % 32 == 0 and weight.shape[0] % 4 == 0):
+        return xops.skinny_gemm(x, weight, bias)
+    return torch.nn.functional.linear(x, weight, bias)
+
+
 class ColumnParallelLinear(nn.Module):
     """Y = X W^T with W sharded along the output dim."""
 
@@ -27,8 +37,11 @@ class ColumnParallelLinear(nn.Module):
         self.weight = nn.Parameter(
             torch.empty(self.shard_out, in_features, dtype=dtype))
         self.bias = nn.Parameter(torch.zeros(self.shard_out, dtype=dtype)) if bias else None
+        self.use_skinny = False  # opt-in decode fast path (see _maybe_skinny)
 
     def forward(self, x):
+        if self.use_skinny:
+            return _maybe_skinny(x, self.weight, self.bias)
         return torch.nn.functional.linear(x, self.weight, self.bias)
 
 
@@ -65,9 +78,13 @@ class RowParallelLinear(nn.Module):
             torch.empty(out_features, self.shard_in, dtype=dtype))
         # bias added once (rank 0's contribution) to keep the sum correct
         self.bias = nn.Parameter(torch.zeros(out_features, dtype=dtype)) if bias else None
+        self.use_skinny = False
 
     def forward(self, x):
-        y = torch.nn.functional.linear(x, self.weight)
+        if self.use_skinny:
+            y = _maybe_skinny(x, self.weight, None)
+        else:
+            y = torch.nn.functional.linear(x, self.weight)
         y = ps.tp_all_reduce(y)
         if self.bias is not None:
             y = y + self.bias

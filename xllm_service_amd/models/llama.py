@@ -36,6 +36,11 @@ class LlamaAttention(nn.Module):
             [cfg.q_size, cfg.kv_size, cfg.kv_size],
             dtype=dtype)
         self.o_proj = RowParallelLinear(cfg.q_size, cfg.hidden_size, dtype=dtype)
+        # qkv/o are the projections the library GEMMs under-parallelize at
+        # decode batches (bench_kernels.py); gate_up/down are already at the
+        # stream roofline there
+        self.qkv_proj.use_skinny = True
+        self.o_proj.use_skinny = True
 
     def forward(self, x, positions, kv_cache, meta: AttnMetadata, cos_sin):
         qkv = self.qkv_proj(x)

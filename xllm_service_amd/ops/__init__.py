@@ -173,6 +173,18 @@ def mfma_gemm(a, b, bias=None):
                                       else None).to(a.dtype)
 
 
+def skinny_gemm(a, b, bias=None):
+    """Split-K weight-streaming GEMM for decode batches (M <= 128).
+    GPU-only; callers fall back to F.linear elsewhere."""
+    _require_ext()
+    import math
+    sk = (a.shape[1] + 1023) // 1024
+    mpad = (a.shape[0] + 15) // 16 * 16
+    ws = torch.empty(sk, mpad, b.shape[0], dtype=torch.float32,
+                     device=a.device)
+    return _ops.skinny_gemm(a, b, bias, ws)
+
+
 def mfma_probe_16x16x32(a, b):
     _require_ext()
     return _ops.mfma_probe_16x16x32(a, b)
