@@ -241,7 +241,7 @@ torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor b,
   TORCH_CHECK(ws.dtype() == torch::kFloat && ws.is_cuda());
   const int M = a.size(0), K = a.size(1), N = b.size(0);
   TORCH_CHECK(a.size(1) == b.size(1) && M <= 128);
-  TORCH_CHECK(K % 32 == 0 && N % 4 == 0);
+  TORCH_CHECK(K % 64 == 0 && N % 4 == 0);
   auto c = torch::empty({M, N}, a.options());
   const unsigned short* bp = nullptr;
   if (bias.has_value()) bp = u16c(bias.value());

@@ -44,25 +44,36 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 #pragma unroll
   for (int m = 0; m < MT; m++) acc[m] = f32x4{0, 0, 0, 0};
 
-  const long wrow = (long)(col_ok ? col : 0) * K;
-  for (int k0 = k_begin; k0 < k_end; k0 += 32) {
-    bf16x8 bk;
-    if (col_ok) {
-      bk = *reinterpret_cast<const bf16x8*>(w + wrow + k0 + fcol8);
-    } else {
-      bk = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-    }
+  // Guards are hoisted into CLAMPED base pointers: an in-loop conditional
+  // around each load makes hipcc branch + drain vmcnt per element (guide
+  // §5 ".s-level traps" (c)). Clamped rows produce garbage partials only
+  // for output rows/cols the epilogue never stores.
+  const unsigned short* wbase =
+      w + (long)(col_ok ? col : N - 1) * K + fcol8;
+  const unsigned short* abase[MT];
+#pragma unroll
+  for (int m = 0; m < MT; m++) {
+    const int row = m * 16 + frow;
+    abase[m] = a + (long)(row < M ? row : M - 1) * K + fcol8;
+  }
+  // unroll by 2 K-steps so 2 B-loads + 2*MT A-loads stay in flight
+  for (int k0 = k_begin; k0 < k_end; k0 += 64) {
+    bf16x8 bk0 = *reinterpret_cast<const bf16x8*>(wbase + k0);
+    bf16x8 bk1 = *reinterpret_cast<const bf16x8*>(wbase + k0 + 32);
+    bf16x8 av0[MT], av1[MT];
 #pragma unroll
     for (int m = 0; m < MT; m++) {
-      const int row = m * 16 + frow;
-      bf16x8 av;
-      if (row < M) {
-        av = *reinterpret_cast<const bf16x8*>(a + (long)row * K + k0 + fcol8);
-      } else {
-        av = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-      }
-      acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bk, acc[m], 0, 0, 0);
+      av0[m] = *reinterpret_cast<const bf16x8*>(abase[m] + k0);
+      av1[m] = *reinterpret_cast<const bf16x8*>(abase[m] + k0 + 32);
     }
+#pragma unroll
+    for (int m = 0; m < MT; m++)
+      acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av0[m], bk0, acc[m],
+                                                       0, 0, 0);
+#pragma unroll
+    for (int m = 0; m < MT; m++)
+      acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av1[m], bk1, acc[m],
+                                                       0, 0, 0);
   }
 
   // partials in C layout: ws[sk][m*16 + crow4 + r][col]

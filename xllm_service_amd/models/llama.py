@@ -36,11 +36,11 @@ class LlamaAttention(nn.Module):
             [cfg.q_size, cfg.kv_size, cfg.kv_size],
             dtype=dtype)
         self.o_proj = RowParallelLinear(cfg.q_size, cfg.hidden_size, dtype=dtype)
-        # qkv/o are the projections the library GEMMs under-parallelize at
-        # decode batches (bench_kernels.py); gate_up/down are already at the
-        # stream roofline there
-        self.qkv_proj.use_skinny = True
-        self.o_proj.use_skinny = True
+        # NOTE: an experimental split-K weight-streaming GEMM
+        # (csrc/ops/skinny_gemm.hip) was measured against hipBLASLt for the
+        # decode qkv/o shapes and LOST (1.1-1.6 TB/s vs the library's ~3.5
+        # TB/s aggregate in-graph) — library GEMMs stay on this path;
+        # use_skinny remains available per-layer for future experiments.
 
     def forward(self, x, positions, kv_cache, meta: AttnMetadata, cos_sin):
         qkv = self.qkv_proj(x)
