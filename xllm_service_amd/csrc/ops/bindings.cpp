@@ -30,8 +30,10 @@ void launch_gather_blocks(unsigned short*, unsigned short*, const long*, int,
                           long, bool, hipStream_t);
 void launch_paged_attn_decode(unsigned short*, const unsigned short*,
                               const unsigned short*, const unsigned short*,
-                              const int*, const int*, float, int, int, int,
-                              int, int, long, long, hipStream_t);
+                              const int*, const int*, float*, float*, float,
+                              int, int, int, int, int, long, long,
+                              hipStream_t);
+int paged_attn_decode_partitions(int, int);
 void launch_paged_attn_prefill(unsigned short*, const unsigned short*,
                                const unsigned short*, const unsigned short*,
                                const int*, const int*, const int*, const int*,
@@ -175,9 +177,21 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(D <= 128 && D % 8 == 0, "head_dim must be <=128, mult of 8");
   TORCH_CHECK(n_qheads % n_kv == 0 && n_qheads / n_kv <= 8,
               "GQA group must be <= 8");
+  const int G = n_qheads / n_kv;
+  const int W = xllm::paged_attn_decode_partitions(num_seqs, n_kv) * 4;
+  auto fopt = q.options().dtype(torch::kFloat);
+  torch::Tensor ws_ml, ws_o;
+  if (D == 128) {
+    ws_ml = torch::empty({(long)num_seqs * n_kv * W * G * 2}, fopt);
+    ws_o = torch::empty({(long)num_seqs * n_kv * W * G * 128}, fopt);
+  } else {  // small-head fallback kernel merges in-workgroup
+    ws_ml = torch::empty({1}, fopt);
+    ws_o = torch::empty({1}, fopt);
+  }
   xllm::launch_paged_attn_decode(
       u16(out), u16c(q), u16c(k_cache), u16c(v_cache),
-      block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), (float)scale,
+      block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
+      ws_ml.data_ptr<float>(), ws_o.data_ptr<float>(), (float)scale,
       num_seqs, n_qheads, n_kv, D, block_tables.size(1), q.stride(0),
       out.stride(0), cur_stream());
 }

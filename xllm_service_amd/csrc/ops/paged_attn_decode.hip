@@ -32,7 +32,8 @@ typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4;
 
 template <int G>
 __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
-    unsigned short* __restrict__ out,        // [num_seqs, n_qheads, D] bf16
+    float* __restrict__ ws_ml,               // [S*n_kv*W][G][2] partial m,l
+    float* __restrict__ ws_o,                // [S*n_kv*W][G][D] partial O
     const unsigned short* __restrict__ q,    // [num_seqs, n_qheads, D] bf16
     const unsigned short* __restrict__ k_cache,  // [blocks, n_kv, BS, D]
     const unsigned short* __restrict__ v_cache,
@@ -40,12 +41,15 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
     const int* __restrict__ seq_lens,        // [num_seqs]
     const float scale,
     const int n_kv_heads, const int max_blocks_per_seq,
-    const long q_stride, const long out_stride) {
-  const int seq = blockIdx.x / n_kv_heads;
-  const int kvh = blockIdx.x % n_kv_heads;
+    const long q_stride, const int P) {     // P partitions per (seq, kvh)
+  const int sk = blockIdx.x / P;            // (seq, kvh) unit
+  const int wgp = blockIdx.x % P;           // partition within the unit
+  const int seq = sk / n_kv_heads;
+  const int kvh = sk % n_kv_heads;
   const int n_qheads = n_kv_heads * G;
   const int seq_len = seq_lens[seq];
   const int n_tiles = (seq_len + PA_KBLK - 1) / PA_KBLK;
+  const int nwaves = P * 4;                 // total partials per unit
 
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
@@ -53,11 +57,14 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
   const int fcol8 = (lane >> 4) * 8;
   const int crow4 = (lane >> 4) * 4;
 
-  // per-wave LDS scratch + cross-wave merge buffers
+  // per-wave LDS scratch + cross-wave merge buffers.
+  // vt_lds holds the tr16-compatible V image: per n-tile (16 d) and k-half
+  // t, four [4 k][16 d] row-major blocks (one per 16-lane group), so P·V
+  // B-fragments come from ds_read_b64_tr_b16 (guide T10) instead of a
+  // scalar transpose. Element (k, d) lives at
+  //   (d>>4)*512 + ((k>>2)&1)*256 + (k>>3)*64 + (k&3)*16 + (d&15).
   __shared__ unsigned short p_lds[4][16][PA_KBLK + PA_PAD];
-  __shared__ unsigned short vt_lds[4][PA_D][PA_KBLK + PA_PAD];
-  __shared__ float m_merge[4][16][2];              // (m, l) per wave per row
-  __shared__ float o_merge[4][16][PA_D];           // rescaled O per wave
+  __shared__ unsigned short vt_lds[4][4096];
 
   // ---- Q as A-fragment: row = head-in-group (zero-padded to 16) -----------
   bf16x8 qf[4];
@@ -81,7 +88,7 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
   const int* btab = block_tables + (long)seq * max_blocks_per_seq;
   const long head_stride = (long)PA_BS * PA_D;
 
-  for (int tile = wid; tile < n_tiles; tile += 4) {
+  for (int tile = wgp * 4 + wid; tile < n_tiles; tile += nwaves) {
     const int t0 = tile * PA_KBLK;
     const long base0 = ((long)btab[t0 / PA_BS] * n_kv_heads + kvh) * head_stride;
     const long base1 = (t0 + 16 < seq_len)
@@ -107,8 +114,7 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
       }
     }
 
-    // ---- stage V^T for this tile (per-wave private; vector read, scalar
-    // transpose write) ------------------------------------------------------
+    // ---- stage V into the tr16 image (vector loads AND vector writes) -----
     {
       // lane covers token tv = lane>>1, d-half dv = (lane&1)*64
       const int tv = lane >> 1;
@@ -116,8 +122,10 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
       const bool vok = (t0 + tv) < seq_len;
       const long vbase = ((tv < 16) ? base0 + (long)tv * PA_D
                                     : base1 + (long)(tv - 16) * PA_D) + dv;
+      const int koff = ((tv >> 2) & 1) * 256 + (tv >> 3) * 64 + (tv & 3) * 16;
 #pragma unroll
       for (int mseg = 0; mseg < 8; mseg++) {
+        const int d0 = dv + mseg * 8;
         ushort8_t vv;
         if (vok) {
           vv = *reinterpret_cast<const ushort8_t*>(v_cache + vbase + mseg * 8);
@@ -125,9 +133,8 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
 #pragma unroll
           for (int j = 0; j < 8; j++) vv.x[j] = 0;
         }
-#pragma unroll
-        for (int j = 0; j < 8; j++)
-          vt_lds[wid][dv + mseg * 8 + j][tv] = vv.x[j];
+        *reinterpret_cast<ushort8_t*>(
+            &vt_lds[wid][(d0 >> 4) * 512 + koff + (d0 & 15)]) = vv;
       }
     }
 
@@ -175,49 +182,128 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
       for (int r = 0; r < 4; r++) o_acc[n][r] *= alpha[r];
     }
     bf16x8 pa = *reinterpret_cast<const bf16x8*>(&p_lds[wid][frow][fcol8]);
-    // ---- O += P V ---------------------------------------------------------
+    // ---- O += P V (B-fragments via hardware transpose reads) --------------
+    // per tr read a 16-lane group gathers one [4 k][16 d] block; lane
+    // address = image + (l>>4)*128B + (l&15)*8B, sub-tiles via offset:
+    {
+      typedef __attribute__((__vector_size__(2 * sizeof(unsigned)))) unsigned u32x2;
+      // low 32 bits of a generic LDS pointer ARE the LDS byte offset
+      // (the shared aperture is 2^32-aligned on amdgcn)
+      const unsigned vaddr =
+          (unsigned)(unsigned long long)(&vt_lds[wid][0]) +
+          ((lane >> 4) * 128u + (lane & 15) * 8u);
+      // two half-batches of 8 tr reads keep 16 (not 32) result VGPRs live:
+      // the kernel is VGPR-occupancy-bound (3 waves/SIMD at <=168 regs)
+#define PA_TR8(OFF0)                                                       \
+      asm volatile(                                                        \
+          "ds_read_b64_tr_b16 %[t0], %[a] offset:" #OFF0 "+0\n\t"          \
+          "ds_read_b64_tr_b16 %[t1], %[a] offset:" #OFF0 "+512\n\t"        \
+          "ds_read_b64_tr_b16 %[t2], %[a] offset:" #OFF0 "+1024\n\t"       \
+          "ds_read_b64_tr_b16 %[t3], %[a] offset:" #OFF0 "+1536\n\t"       \
+          "ds_read_b64_tr_b16 %[t4], %[a] offset:" #OFF0 "+2048\n\t"       \
+          "ds_read_b64_tr_b16 %[t5], %[a] offset:" #OFF0 "+2560\n\t"       \
+          "ds_read_b64_tr_b16 %[t6], %[a] offset:" #OFF0 "+3072\n\t"       \
+          "ds_read_b64_tr_b16 %[t7], %[a] offset:" #OFF0 "+3584\n\t"       \
+          "s_waitcnt lgkmcnt(0)"                                           \
+          : [t0] "=&v"(tr[0]), [t1] "=&v"(tr[1]), [t2] "=&v"(tr[2]),       \
+            [t3] "=&v"(tr[3]), [t4] "=&v"(tr[4]), [t5] "=&v"(tr[5]),       \
+            [t6] "=&v"(tr[6]), [t7] "=&v"(tr[7])                           \
+          : [a] "v"(vaddr)                                                 \
+          : "memory")
+      u32x2 tr[8];
+      u32x2 tr2[8];
 #pragma unroll
-    for (int n = 0; n < 8; n++) {
-      bf16x8 bv = *reinterpret_cast<const bf16x8*>(
-          &vt_lds[wid][n * 16 + frow][fcol8]);
-      o_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bv, o_acc[n], 0, 0, 0);
+      for (int half = 0; half < 2; half++) {
+        if (half == 0) { PA_TR8(0); } else {
+          // second half into separate regs so both batches stay in flight
+#define PA_SWAP_TR tr2
+          asm volatile(
+              "ds_read_b64_tr_b16 %[t0], %[a] offset:4096+0\n\t"
+              "ds_read_b64_tr_b16 %[t1], %[a] offset:4096+512\n\t"
+              "ds_read_b64_tr_b16 %[t2], %[a] offset:4096+1024\n\t"
+              "ds_read_b64_tr_b16 %[t3], %[a] offset:4096+1536\n\t"
+              "ds_read_b64_tr_b16 %[t4], %[a] offset:4096+2048\n\t"
+              "ds_read_b64_tr_b16 %[t5], %[a] offset:4096+2560\n\t"
+              "ds_read_b64_tr_b16 %[t6], %[a] offset:4096+3072\n\t"
+              "ds_read_b64_tr_b16 %[t7], %[a] offset:4096+3584\n\t"
+              "s_waitcnt lgkmcnt(0)"
+              : [t0] "=&v"(tr2[0]), [t1] "=&v"(tr2[1]), [t2] "=&v"(tr2[2]),
+                [t3] "=&v"(tr2[3]), [t4] "=&v"(tr2[4]), [t5] "=&v"(tr2[5]),
+                [t6] "=&v"(tr2[6]), [t7] "=&v"(tr2[7])
+              : [a] "v"(vaddr)
+              : "memory");
+        }
+        __builtin_amdgcn_sched_barrier(0);  // MFMAs stay below the wait
+#pragma unroll
+        for (int nn = 0; nn < 4; nn++) {
+          const int n = half * 4 + nn;
+          bf16x8 bv;
+          unsigned* bw = reinterpret_cast<unsigned*>(&bv);
+          u32x2* src = half ? tr2 : tr;
+          bw[0] = src[2 * nn][0];
+          bw[1] = src[2 * nn][1];
+          bw[2] = src[2 * nn + 1][0];
+          bw[3] = src[2 * nn + 1][1];
+          o_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bv, o_acc[n],
+                                                             0, 0, 0);
+        }
+      }
+#undef PA_TR8
     }
   }
 
-  // ---- cross-wave merge ----------------------------------------------------
-  // each wave writes its (m, l) and O to LDS (C layout: row crow4+r, col frow)
+  // ---- write this wave's partial (no cross-wave sync: the tiny merge
+  // kernel below combines the P*4 partials per (seq, kvh) unit) ------------
+  const long part = ((long)sk * nwaves + wgp * 4 + wid);
   if (frow == 0) {
 #pragma unroll
     for (int r = 0; r < 4; r++) {
-      m_merge[wid][crow4 + r][0] = m_r[r];
-      m_merge[wid][crow4 + r][1] = l_r[r];
+      const int row = crow4 + r;
+      if (row < G) {
+        ws_ml[(part * G + row) * 2 + 0] = m_r[r];
+        ws_ml[(part * G + row) * 2 + 1] = l_r[r];
+      }
     }
   }
 #pragma unroll
   for (int n = 0; n < 8; n++) {
 #pragma unroll
-    for (int r = 0; r < 4; r++)
-      o_merge[wid][crow4 + r][n * 16 + frow] = o_acc[n][r];
+    for (int r = 0; r < 4; r++) {
+      const int row = crow4 + r;
+      if (row < G)
+        ws_o[(part * G + row) * PA_D + n * 16 + frow] = o_acc[n][r];
+    }
   }
-  __syncthreads();
+}
 
-  // 256 threads cover (g, d) pairs: G*128 <= 2048 elements
-  for (int idx = threadIdx.x; idx < G * PA_D; idx += 256) {
-    const int g = idx / PA_D;
-    const int d = idx % PA_D;
+// Combine the W = P*4 per-wave partials of each (seq, kvh, g) row.
+__global__ void paged_attn_decode_merge_kernel(
+    unsigned short* __restrict__ out,        // [S, n_qheads, D]
+    const float* __restrict__ ws_ml,         // [S*n_kv*W][G][2]
+    const float* __restrict__ ws_o,          // [S*n_kv*W][G][D]
+    const int G, const int n_kv_heads, const int W, const long out_stride,
+    const long total) {                      // total = S * n_kv * G * D
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const int d = (int)(idx % PA_D);
+    long t = idx / PA_D;
+    const int g = (int)(t % G);
+    t /= G;
+    const int kvh = (int)(t % n_kv_heads);
+    const long seq = t / n_kv_heads;
+    const long unit = (seq * n_kv_heads + kvh);
     float m_star = -INFINITY;
-#pragma unroll
-    for (int w = 0; w < 4; w++) m_star = fmaxf(m_star, m_merge[w][g][0]);
+    for (int w = 0; w < W; w++)
+      m_star = fmaxf(m_star, ws_ml[((unit * W + w) * G + g) * 2]);
     float l_tot = 0.0f, o = 0.0f;
-#pragma unroll
-    for (int w = 0; w < 4; w++) {
-      const float mw = m_merge[w][g][0];
+    for (int w = 0; w < W; w++) {
+      const float mw = ws_ml[((unit * W + w) * G + g) * 2];
       const float wgt = (mw == -INFINITY) ? 0.0f : __expf(mw - m_star);
-      l_tot += wgt * m_merge[w][g][1];
-      o += wgt * o_merge[w][g][d];
+      l_tot += wgt * ws_ml[((unit * W + w) * G + g) * 2 + 1];
+      o += wgt * ws_o[((unit * W + w) * G + g) * PA_D + d];
     }
     const float inv = (l_tot > 0.0f) ? 1.0f / l_tot : 0.0f;
-    out[(long)seq * out_stride + (long)(kvh * G + g) * PA_D + d] =
+    out[seq * out_stride + (long)(kvh * G + g) * PA_D + d] =
         f32_to_bf16(o * inv);
   }
 }
@@ -337,28 +423,48 @@ __global__ __launch_bounds__(256) void paged_attn_decode_small_kernel(
 
 #define PA_DISPATCH_G(GV)                                                      \
   do {                                                                         \
-    if (D == PA_D)                                                             \
+    if (D == PA_D) {                                                           \
       hipLaunchKernelGGL((paged_attn_decode_kernel<GV>), grid, block, 0,       \
-                         stream, out, q, k_cache, v_cache, block_tables,       \
-                         seq_lens, scale, n_kv_heads, max_blocks_per_seq,      \
-                         q_stride, out_stride);                                \
-    else                                                                       \
+                         stream, ws_ml, ws_o, q, k_cache, v_cache,             \
+                         block_tables, seq_lens, scale, n_kv_heads,            \
+                         max_blocks_per_seq, q_stride, P);                     \
+      hipLaunchKernelGGL(paged_attn_decode_merge_kernel, mgrid, block, 0,      \
+                         stream, out, ws_ml, ws_o, GV, n_kv_heads, P * 4,      \
+                         out_stride, total);                                   \
+    } else {                                                                   \
       hipLaunchKernelGGL((paged_attn_decode_small_kernel<GV>), grid, block, 0, \
                          stream, out, q, k_cache, v_cache, block_tables,       \
                          seq_lens, scale, n_kv_heads, D, max_blocks_per_seq,   \
                          q_stride, out_stride);                                \
+    }                                                                          \
   } while (0)
+
+int paged_attn_decode_partitions(int num_seqs, int n_kv_heads) {
+  // partition only when there are too few (seq, kv_head) units to fill the
+  // chip: extra partitions shrink per-wave work and cost merge overhead
+  const int units = num_seqs * n_kv_heads;
+  int P = (512 + units - 1) / units;
+  if (P < 1) P = 1;
+  if (P > 8) P = 8;
+  return P;
+}
 
 void launch_paged_attn_decode(unsigned short* out, const unsigned short* q,
                               const unsigned short* k_cache,
                               const unsigned short* v_cache,
                               const int* block_tables, const int* seq_lens,
+                              float* ws_ml, float* ws_o,
                               float scale, int num_seqs, int n_qheads,
                               int n_kv_heads, int D, int max_blocks_per_seq,
                               long q_stride, long out_stride,
                               hipStream_t stream) {
-  dim3 grid(num_seqs * n_kv_heads), block(256);
+  const int P = paged_attn_decode_partitions(num_seqs, n_kv_heads);
+  dim3 grid(num_seqs * n_kv_heads * (D == PA_D ? P : 1)), block(256);
   const int G = n_qheads / n_kv_heads;
+  const long total = (long)num_seqs * n_kv_heads * G * PA_D;
+  long mg = (total + 255) / 256;
+  if (mg > 2048) mg = 2048;
+  dim3 mgrid((unsigned)mg);
   switch (G) {
     case 1: PA_DISPATCH_G(1); break;
     case 2: PA_DISPATCH_G(2); break;
