@@ -78,7 +78,9 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_kernel(
   const int wg_kmax = ctx + wg_q_end;            // == max over waves
 
   __shared__ unsigned short k_lds[PF_KBLK][PF_D + PF_KPAD];
-  __shared__ unsigned short vt_lds[PF_D][PF_KBLK + PF_VPAD];
+  // tr16-compatible V image (see paged_attn_decode.hip): element (k, d) at
+  // (d>>4)*512 + ((k>>2)&1)*256 + (k>>3)*64 + (k&3)*16 + (d&15)
+  __shared__ unsigned short vt_lds[4096];
   __shared__ unsigned short p_lds[4][PF_QBLK][PF_KBLK + PF_VPAD];
 
   // ---- load Q fragments (lane holds row frow, cols ks*32+fcol8..+8) ----------
@@ -126,8 +128,7 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_kernel(
         }
         *reinterpret_cast<ushort8_t*>(&k_lds[kt][d]) = val;
       }
-      // V^T: element [d][kt] <- V[tok][d]; vectorize over d (8 per thread),
-      // scatter as scalars into the transposed image
+      // V into the tr16 image: vector loads AND vector writes
       for (int idx = threadIdx.x * 8; idx < PF_KBLK * PF_D; idx += 256 * 8) {
         const int kt = idx / PF_D;
         const int d = idx % PF_D;
@@ -142,8 +143,10 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_kernel(
 #pragma unroll
           for (int j = 0; j < 8; j++) val.x[j] = 0;
         }
-#pragma unroll
-        for (int j = 0; j < 8; j++) vt_lds[d + j][kt] = val.x[j];
+        const int koff =
+            ((kt >> 2) & 1) * 256 + (kt >> 3) * 64 + (kt & 3) * 16;
+        *reinterpret_cast<ushort8_t*>(
+            &vt_lds[(d >> 4) * 512 + koff + (d & 15)]) = val;
       }
     }
     __syncthreads();
@@ -215,12 +218,54 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_kernel(
       }
       // A frag of P: lane holds P[frow][fcol8 + j]
       bf16x8 pa = *reinterpret_cast<const bf16x8*>(&p_lds[wid][frow][fcol8]);
-      // ---- O += P V ----------------------------------------------------------
+      // ---- O += P V (B-fragments via hardware transpose reads) --------------
+      {
+        typedef __attribute__((__vector_size__(2 * sizeof(unsigned)))) unsigned u32x2;
+        const unsigned vaddr =
+            (unsigned)(unsigned long long)(&vt_lds[0]) +
+            ((lane >> 4) * 128u + (lane & 15) * 8u);
+        u32x2 tr[8];
+        u32x2 tr2[8];
+        asm volatile(
+            "ds_read_b64_tr_b16 %[t0], %[a] offset:0\n\t"
+            "ds_read_b64_tr_b16 %[t1], %[a] offset:512\n\t"
+            "ds_read_b64_tr_b16 %[t2], %[a] offset:1024\n\t"
+            "ds_read_b64_tr_b16 %[t3], %[a] offset:1536\n\t"
+            "ds_read_b64_tr_b16 %[t4], %[a] offset:2048\n\t"
+            "ds_read_b64_tr_b16 %[t5], %[a] offset:2560\n\t"
+            "ds_read_b64_tr_b16 %[t6], %[a] offset:3072\n\t"
+            "ds_read_b64_tr_b16 %[t7], %[a] offset:3584\n\t"
+            "ds_read_b64_tr_b16 %[u0], %[a] offset:4096\n\t"
+            "ds_read_b64_tr_b16 %[u1], %[a] offset:4608\n\t"
+            "ds_read_b64_tr_b16 %[u2], %[a] offset:5120\n\t"
+            "ds_read_b64_tr_b16 %[u3], %[a] offset:5632\n\t"
+            "ds_read_b64_tr_b16 %[u4], %[a] offset:6144\n\t"
+            "ds_read_b64_tr_b16 %[u5], %[a] offset:6656\n\t"
+            "ds_read_b64_tr_b16 %[u6], %[a] offset:7168\n\t"
+            "ds_read_b64_tr_b16 %[u7], %[a] offset:7680\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : [t0] "=&v"(tr[0]), [t1] "=&v"(tr[1]), [t2] "=&v"(tr[2]),
+              [t3] "=&v"(tr[3]), [t4] "=&v"(tr[4]), [t5] "=&v"(tr[5]),
+              [t6] "=&v"(tr[6]), [t7] "=&v"(tr[7]),
+              [u0] "=&v"(tr2[0]), [u1] "=&v"(tr2[1]), [u2] "=&v"(tr2[2]),
+              [u3] "=&v"(tr2[3]), [u4] "=&v"(tr2[4]), [u5] "=&v"(tr2[5]),
+              [u6] "=&v"(tr2[6]), [u7] "=&v"(tr2[7])
+            : [a] "v"(vaddr)
+            : "memory");
+        __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
-      for (int n = 0; n < 8; n++) {
-        // B frag: V[k = fcol8+j][d = n*16+frow] = vt_lds[d][k]
-        bf16x8 bv = *reinterpret_cast<const bf16x8*>(&vt_lds[n * 16 + frow][fcol8]);
-        o_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bv, o_acc[n], 0, 0, 0);
+        for (int n = 0; n < 8; n++) {
+          bf16x8 bv;
+          unsigned* bw = reinterpret_cast<unsigned*>(&bv);
+          u32x2* src = (n < 4) ? tr : tr2;
+          const int nn = n & 3;
+          bw[0] = src[2 * nn][0];
+          bw[1] = src[2 * nn][1];
+          bw[2] = src[2 * nn + 1][0];
+          bw[3] = src[2 * nn + 1][1];
+          o_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bv, o_acc[n],
+                                                             0, 0, 0);
+        }
       }
     }
     __syncthreads();
