@@ -93,3 +93,21 @@ def test_gpu_qwen2_vl_engine():
         for o in eng.step():
             toks.extend(o.new_token_ids)
     assert len(toks) == 5
+
+
+def test_gpu_swap_tier():
+    """KV swap to pinned host memory and back on the GPU path."""
+    eng = LLMEngine("llama-debug-128", device="cuda:0", max_kv_blocks=9,
+                    seed=3, enable_prefix_caching=False, enable_graphs=False)
+    cfg = get_config("llama-debug-128")
+    torch.manual_seed(31)
+    prompts = [torch.randint(0, cfg.vocab_size, (60,)).tolist()
+               for _ in range(2)]
+    got = eng.generate(prompts, SamplingParams(max_tokens=8, ignore_eos=True))
+    assert eng.scheduler.num_swap_outs > 0 and eng.scheduler.num_swap_ins > 0
+    # swap must be lossless: same outputs as an unconstrained engine
+    eng2 = LLMEngine("llama-debug-128", device="cuda:0", max_kv_blocks=256,
+                     seed=3, enable_prefix_caching=False, enable_graphs=False)
+    want = eng2.generate(prompts, SamplingParams(max_tokens=8,
+                                                 ignore_eos=True))
+    assert got == want

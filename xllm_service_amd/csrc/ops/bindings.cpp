@@ -373,6 +373,40 @@ void migrate_blocks_from_ptr(torch::Tensor dst_cache, int64_t src_ptr,
   }
 }
 
+// ---- KV block swap GPU<->CPU (the "dram" cache tier) ---------------------
+// Contiguous (src,dst) runs collapse into single async copies on the
+// current stream; callers use a side stream + synchronize.
+void swap_blocks(torch::Tensor dst_cache, torch::Tensor src_cache,
+                 std::vector<long> src_blocks, std::vector<long> dst_blocks) {
+  TORCH_CHECK(src_blocks.size() == dst_blocks.size());
+  TORCH_CHECK(src_cache.element_size() == dst_cache.element_size());
+  const long numel =
+      (long)src_cache.size(1) * src_cache.size(2) * src_cache.size(3);
+  const long bytes = numel * src_cache.element_size();
+  char* dst = reinterpret_cast<char*>(dst_cache.data_ptr());
+  const char* src = reinterpret_cast<const char*>(src_cache.data_ptr());
+  hipMemcpyKind kind =
+      src_cache.is_cuda()
+          ? (dst_cache.is_cuda() ? hipMemcpyDeviceToDevice
+                                 : hipMemcpyDeviceToHost)
+          : hipMemcpyHostToDevice;
+  hipStream_t stream = cur_stream();
+  size_t i = 0;
+  while (i < src_blocks.size()) {
+    size_t j = i + 1;
+    while (j < src_blocks.size() && src_blocks[j] == src_blocks[j - 1] + 1 &&
+           dst_blocks[j] == dst_blocks[j - 1] + 1)
+      j++;
+    const long n = (long)(j - i);
+    hipError_t err = hipMemcpyAsync(dst + dst_blocks[i] * bytes,
+                                    src + src_blocks[i] * bytes,
+                                    (size_t)(n * bytes), kind, stream);
+    TORCH_CHECK(err == hipSuccess, "swap_blocks copy failed: ",
+                hipGetErrorString(err));
+    i = j;
+  }
+}
+
 void enable_peer_access(long device, long peer) {
   int can = 0;
   hipError_t err = hipDeviceCanAccessPeer(&can, (int)device, (int)peer);
@@ -406,5 +440,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ipc_open_handle", &ipc_open_handle);
   m.def("ipc_close_handle", &ipc_close_handle);
   m.def("migrate_blocks_from_ptr", &migrate_blocks_from_ptr);
+  m.def("swap_blocks", &swap_blocks);
   m.def("enable_peer_access", &enable_peer_access);
 }

@@ -99,12 +99,14 @@ class BlockManager:
         max_match = (seq.prompt_len - 1) // self.block_size
         return min(n, max_match) * self.block_size
 
-    def can_allocate(self, seq: Sequence, num_tokens: int) -> bool:
-        """Can we hold the first num_tokens of this sequence?"""
+    def can_allocate(self, seq: Sequence, num_tokens: int,
+                     reserve_blocks: int = 0) -> bool:
+        """Can we hold the first num_tokens of this sequence while leaving
+        reserve_blocks free (blocks already promised to scheduled decodes)?"""
         cached = self.match_prefix(seq) if not seq.block_table else 0
         need = (num_tokens + self.block_size - 1) // self.block_size
         need -= cached // self.block_size
-        return need <= self.num_free
+        return need <= self.num_free - reserve_blocks
 
     def allocate_prefill(self, seq: Sequence) -> int:
         """Allocate blocks for the whole prompt; returns cached-token count.
@@ -141,10 +143,13 @@ class BlockManager:
                 self.hash_to_block[hashes[i]] = blk
                 self.events.stored.add(hashes[i])
 
-    def can_append(self, seq: Sequence) -> bool:
+    def needs_append_block(self, seq: Sequence) -> bool:
         # decode writes KV for position total_len - 1 (the freshly-fed token)
-        need = 1 if (seq.total_len - 1) // self.block_size >= len(seq.block_table) else 0
-        return need <= self.num_free
+        return (seq.total_len - 1) // self.block_size >= len(seq.block_table)
+
+    def can_append(self, seq: Sequence, reserve_blocks: int = 0) -> bool:
+        need = 1 if self.needs_append_block(seq) else 0
+        return need <= self.num_free - reserve_blocks
 
     def append_slot(self, seq: Sequence) -> int:
         """Slot index for the token being decoded (block alloc on boundary)."""

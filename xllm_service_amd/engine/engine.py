@@ -53,6 +53,7 @@ class LLMEngine:
                  enable_prefix_caching: bool = True,
                  enable_graphs: bool = True,
                  max_model_len: int = 4096,
+                 swap_space_mb: int = 1024,
                  seed: int = 0):
         self.cfg: ModelConfig = get_config(model_name)
         if device is None:
@@ -76,14 +77,31 @@ class LLMEngine:
             max_blocks=max_kv_blocks)
         self.block_manager = BlockManager(num_blocks, BLOCK_SIZE,
                                           enable_prefix_caching)
-        self.scheduler = EngineScheduler(self.block_manager,
-                                         max_num_seqs=max_num_seqs,
-                                         max_batched_tokens=max_batched_tokens)
+        # host-DRAM KV tier ("dram" in the reference's hbm/dram/ssd cache
+        # tiers): preempted sequences swap out instead of recomputing
+        n_kv = n_kv_local
+        per_block_bytes = (2 * n_kv * BLOCK_SIZE * self.cfg.head_dim *
+                           (2 if dtype == torch.bfloat16 else 4) *
+                           self.cfg.num_layers)
+        num_cpu_blocks = max((swap_space_mb << 20) // per_block_bytes, 0)
+        self.cpu_block_manager = (
+            BlockManager(num_cpu_blocks, BLOCK_SIZE,
+                         enable_prefix_caching=False)
+            if num_cpu_blocks > 0 else None)
+        self.scheduler = EngineScheduler(
+            self.block_manager, max_num_seqs=max_num_seqs,
+            max_batched_tokens=max_batched_tokens,
+            swap_out=self._swap_out if self.cpu_block_manager else None,
+            swap_in=self._swap_in if self.cpu_block_manager else None)
+        if self.cpu_block_manager:
+            self.scheduler.free_cpu_blocks = self._free_cpu_blocks
         self.runner = ModelRunner(self.model, self.cfg, self.device,
                                   num_blocks, dtype=dtype,
                                   enable_graphs=enable_graphs,
                                   max_model_len=max_model_len,
                                   max_graph_batch=min(max_num_seqs, 256))
+        if self.cpu_block_manager:
+            self.runner.alloc_cpu_caches(self.cpu_block_manager.num_blocks)
         if self.device.type == "cuda":
             self.runner.capture_graphs()
         self.seqs: Dict[str, Sequence] = {}
@@ -133,6 +151,41 @@ class LLMEngine:
                                          None)
         self.seqs[request_id] = seq
         self.scheduler.add(seq)
+
+    # ---- host-DRAM KV tier (swap) ------------------------------------------
+    def _swap_out(self, seq):
+        from xllm_service_amd import ops as xops
+        cbm = self.cpu_block_manager
+        n = len(seq.block_table)
+        if cbm.num_free < n:
+            return None  # dram tier full: caller falls back to recompute
+        cpu_blocks = cbm.allocate_raw(n)
+        for (kc, vc), (ck, cv) in zip(self.runner.kv_caches,
+                                      self.runner.cpu_kv_caches):
+            xops.swap_blocks(ck, kc, seq.block_table, cpu_blocks)
+            xops.swap_blocks(cv, vc, seq.block_table, cpu_blocks)
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+        return cpu_blocks
+
+    def _swap_in(self, seq):
+        from xllm_service_amd import ops as xops
+        n = len(seq.cpu_block_table)
+        gpu_blocks = self.block_manager.allocate_raw(n)
+        for (kc, vc), (ck, cv) in zip(self.runner.kv_caches,
+                                      self.runner.cpu_kv_caches):
+            xops.swap_blocks(kc, ck, seq.cpu_block_table, gpu_blocks)
+            xops.swap_blocks(vc, cv, seq.cpu_block_table, gpu_blocks)
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+        self._free_cpu_blocks(seq)
+        seq.block_table = gpu_blocks
+
+    def _free_cpu_blocks(self, seq):
+        tmp = Sequence("_cpu_tmp", [], SamplingParams())
+        tmp.block_table = list(seq.cpu_block_table)
+        self.cpu_block_manager.free(tmp)
+        seq.cpu_block_table = []
 
     # ---- PD-disaggregation support -----------------------------------------
     def held_block_table(self, request_id: str) -> List[int]:

@@ -42,6 +42,19 @@ class ModelRunner:
                          dtype=dtype, device=device))
             for _ in range(cfg.num_layers)
         ]
+        self.cpu_kv_caches: List[Tuple[torch.Tensor, torch.Tensor]] = []
+
+    def alloc_cpu_caches(self, num_cpu_blocks: int):
+        """Pinned host-DRAM tier for KV swapping."""
+        n_kv = self.kv_caches[0][0].shape[1]
+        pin = self.device.type == "cuda"
+        self.cpu_kv_caches = [
+            (torch.zeros(num_cpu_blocks, n_kv, BLOCK_SIZE, self.cfg.head_dim,
+                         dtype=self.dtype, pin_memory=pin),
+             torch.zeros(num_cpu_blocks, n_kv, BLOCK_SIZE, self.cfg.head_dim,
+                         dtype=self.dtype, pin_memory=pin))
+            for _ in range(self.cfg.num_layers)
+        ]
 
     @staticmethod
     def kv_cache_blocks_for(cfg: ModelConfig, device: torch.device,

@@ -42,13 +42,21 @@ class EngineScheduler:
     def __init__(self, block_manager: BlockManager,
                  max_num_seqs: int = 256,
                  max_batched_tokens: int = 8192,
-                 enable_chunked_prefill: bool = True):
+                 enable_chunked_prefill: bool = True,
+                 swap_out=None, swap_in=None):
         self.bm = block_manager
         self.max_num_seqs = max_num_seqs
         self.max_batched_tokens = max_batched_tokens
         self.enable_chunked_prefill = enable_chunked_prefill
+        # swap_out(seq) -> cpu_blocks | None; swap_in(seq) -> None
+        # (engine wires these to the host-DRAM KV tier; None = recompute)
+        self.swap_out = swap_out
+        self.swap_in = swap_in
         self.waiting: Deque[Sequence] = deque()
+        self.swapped: List[Sequence] = []
         self.running: List[Sequence] = []
+        self.num_swap_outs = 0
+        self.num_swap_ins = 0
 
     # ---- API ----------------------------------------------------------------
     def add(self, seq: Sequence):
@@ -77,16 +85,49 @@ class EngineScheduler:
                     self.bm.free(s)
                 del self.waiting[i]
                 return s
+        for i, s in enumerate(self.swapped):
+            if s.request_id == request_id:
+                s.status = SeqStatus.FINISHED_ABORT
+                if self.swap_in is not None:
+                    # free the dram blocks without copying back
+                    self.free_cpu_blocks(s)
+                del self.swapped[i]
+                return s
         return None
 
+    # engine injects this (frees dram blocks of an aborted swapped seq)
+    free_cpu_blocks = staticmethod(lambda seq: None)
+
     def has_work(self) -> bool:
-        return bool(self.waiting or self.running)
+        return bool(self.waiting or self.running or self.swapped)
 
     @property
     def num_waiting(self) -> int:
         return len(self.waiting)
 
     # ---- core ---------------------------------------------------------------
+    def _do_preempt(self, victim: Sequence):
+        """Swap the victim's KV to the dram tier when possible, else free +
+        recompute (reference behaviour is client-side retry; we keep the
+        request and restore it)."""
+        self.running.remove(victim)
+        victim.preempt_count += 1
+        if self.swap_out is not None and victim.prefill_done:
+            cpu_blocks = self.swap_out(victim)
+            if cpu_blocks is not None:
+                victim.cpu_block_table = cpu_blocks
+                self.bm.free(victim)
+                victim.status = SeqStatus.SWAPPED
+                self.swapped.append(victim)
+                self.num_swap_outs += 1
+                return
+        self.bm.free(victim)
+        victim.status = SeqStatus.PREEMPTED
+        victim.num_computed_tokens = 0
+        victim.prompt_token_ids = victim.all_token_ids()
+        victim.output_token_ids = []
+        self.waiting.appendleft(victim)
+
     def _preempt_one(self) -> bool:
         """Preempt the lowest-priority, most recent running sequence."""
         if not self.running:
@@ -99,42 +140,32 @@ class EngineScheduler:
                 break
         if victim_idx is None:
             victim_idx = len(self.running) - 1
-        victim = self.running.pop(victim_idx)
-        self.bm.free(victim)
-        victim.status = SeqStatus.PREEMPTED
-        victim.num_computed_tokens = 0
-        victim.preempt_count += 1
-        # recompute path: prompt grows by generated tokens so far
-        victim.prompt_token_ids = victim.all_token_ids()
-        victim.output_token_ids = []
-        self.waiting.appendleft(victim)
+        victim = self.running[victim_idx]
+        self._do_preempt(victim)
         return True
 
     def schedule(self) -> StepPlan:
         plan = StepPlan()
         budget = self.max_batched_tokens
 
-        # 1. decodes for all running seqs (preempt on OOM)
+        # 1. decodes for all running seqs (preempt on OOM). `reserved`
+        # counts free blocks already promised to earlier decodes this step.
+        reserved = 0
         for seq in list(self.running):
             if seq not in self.running:      # became a preemption victim
                 continue
             if not seq.prefill_done:
                 continue                     # mid-chunked-prefill: step 2
-            while (not self.bm.can_append(seq)
+            while (not self.bm.can_append(seq, reserved)
                    and self._preempt_victim_excluding(seq, plan)):
                 pass
-            if self.bm.can_append(seq):
+            if self.bm.can_append(seq, reserved):
                 plan.decodes.append(seq)
+                if self.bm.needs_append_block(seq):
+                    reserved += 1
                 budget -= 1
             else:                            # pool exhausted: preempt self
-                self.running.remove(seq)
-                self.bm.free(seq)
-                seq.status = SeqStatus.PREEMPTED
-                seq.num_computed_tokens = 0
-                seq.preempt_count += 1
-                seq.prompt_token_ids = seq.all_token_ids()
-                seq.output_token_ids = []
-                self.waiting.appendleft(seq)
+                self._do_preempt(seq)
                 plan.preempted.append(seq)
 
         # 2. continue chunked prefills already running
@@ -147,12 +178,30 @@ class EngineScheduler:
                     ScheduledPrefill(seq, seq.num_computed_tokens, chunk))
                 budget -= chunk
 
-        # 3. admit waiting sequences
+        # 2.6 resume swapped sequences (dram -> hbm) before new admissions
+        for seq in list(self.swapped):
+            need = len(seq.cpu_block_table) + 1   # +1: next decode block
+            if len(self.running) >= self.max_num_seqs or \
+                    self.bm.num_free - reserved < need:
+                break
+            reserved += 1
+            self.swap_in(seq)
+            self.num_swap_ins += 1
+            seq.status = SeqStatus.RUNNING
+            self.swapped.remove(seq)
+            self.running.append(seq)
+            if seq.prefill_done:
+                budget -= 1
+                plan.decodes.append(seq)
+
+        # 3. admit waiting sequences (leaving the blocks this step's decodes
+        # will take in append_slot untouched)
         while self.waiting and budget > 0 and len(self.running) < self.max_num_seqs:
             seq = self.waiting[0]
             first_alloc = not seq.block_table
             if first_alloc:
-                if not self.bm.can_allocate(seq, seq.prompt_len):
+                if not self.bm.can_allocate(seq, seq.prompt_len,
+                                            reserve_blocks=reserved):
                     break
                 self.bm.allocate_prefill(seq)
             remaining = seq.prompt_len - seq.num_computed_tokens
@@ -187,14 +236,7 @@ class EngineScheduler:
                 break
         if victim is None:
             victim = candidates[-1]
-        self.running.remove(victim)
-        self.bm.free(victim)
-        victim.status = SeqStatus.PREEMPTED
-        victim.num_computed_tokens = 0
-        victim.preempt_count += 1
-        victim.prompt_token_ids = victim.all_token_ids()
-        victim.output_token_ids = []
-        self.waiting.appendleft(victim)
+        self._do_preempt(victim)
         plan.preempted.append(victim)
         return True
 

@@ -13,6 +13,7 @@ class SeqStatus(enum.Enum):
     WAITING = "waiting"
     RUNNING = "running"
     PREEMPTED = "preempted"
+    SWAPPED = "swapped"         # KV offloaded to the host-DRAM tier
     FINISHED_STOP = "finished_stop"      # stop/eos token
     FINISHED_LENGTH = "finished_length"  # max_tokens reached
     FINISHED_ABORT = "finished_abort"    # cancelled (client disconnect etc.)
@@ -35,6 +36,7 @@ class Sequence:
     status: SeqStatus = SeqStatus.WAITING
     output_token_ids: List[int] = field(default_factory=list)
     block_table: List[int] = field(default_factory=list)
+    cpu_block_table: List[int] = field(default_factory=list)  # dram tier
     num_computed_tokens: int = 0          # prompt tokens already prefilled
     preempt_count: int = 0
     # multimodal: pre-computed image embeddings substituted at placeholder

@@ -150,6 +150,17 @@ def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
     return ref.greedy_sample(logits)
 
 
+def swap_blocks(dst_cache, src_cache, src_blocks, dst_blocks):
+    """Move whole KV blocks between the GPU cache and the host-DRAM tier
+    (or CPU<->CPU on the test path)."""
+    if HAS_EXT and (src_cache.is_cuda or dst_cache.is_cuda):
+        _ops.swap_blocks(dst_cache, src_cache, list(src_blocks),
+                         list(dst_blocks))
+        return
+    for s_, d_ in zip(src_blocks, dst_blocks):
+        dst_cache[d_].copy_(src_cache[s_])
+
+
 def migrate_blocks_peer(dst_cache, dst_device, src_cache, src_device,
                         src_blocks, dst_blocks):
     _require_ext()
