@@ -49,6 +49,7 @@ async def run(args):
                    "--seed", "0"]
             if device:
                 cmd += ["--device", device]
+            cmd += ["--max-batched-tokens", str(args.chunk_tokens)]
             procs.append(subprocess.Popen(cmd, cwd=ROOT,
                                           stdout=subprocess.PIPE,
                                           stderr=subprocess.STDOUT))
@@ -102,9 +103,21 @@ async def run(args):
         await asyncio.gather(*[one_request() for _ in range(args.concurrency)])
         results.clear()
         t0 = time.monotonic()
-        per_client = max(args.requests // args.concurrency, 1)
-        await asyncio.gather(*[client_loop(per_client)
-                               for _ in range(args.concurrency)])
+        if args.arrival_rate > 0:
+            # open-loop Poisson arrivals (the fair SLO measurement): one
+            # task per request, spaced by exponential gaps
+            async def open_loop():
+                tasks = []
+                for _ in range(args.requests):
+                    tasks.append(asyncio.create_task(one_request()))
+                    await asyncio.sleep(rnd.expovariate(args.arrival_rate))
+                done = await asyncio.gather(*tasks)
+                results.extend(r for r in done if r is not None)
+            await open_loop()
+        else:
+            per_client = max(args.requests // args.concurrency, 1)
+            await asyncio.gather(*[client_loop(per_client)
+                                   for _ in range(args.concurrency)])
         wall = time.monotonic() - t0
         await client.aclose()
 
@@ -130,6 +143,8 @@ async def run(args):
             "model": args.model,
             "input_len": args.input_len,
             "output_len": args.output_len,
+            "arrival_rate": args.arrival_rate,
+            "chunk_tokens": args.chunk_tokens,
             "data": "synthetic",
         }))
     finally:
@@ -155,6 +170,12 @@ def main():
     ap.add_argument("--output-len", type=int, default=128)
     ap.add_argument("--slo-ttft-ms", type=float, default=1000.0,
                     help="reference default target_ttft")
+    ap.add_argument("--arrival-rate", type=float, default=0.0,
+                    help="requests/s for open-loop Poisson arrivals "
+                         "(0 = closed-loop burst)")
+    ap.add_argument("--chunk-tokens", type=int, default=2048,
+                    help="worker max batched tokens per step (chunked "
+                         "prefill interleave granularity)")
     ap.add_argument("--startup-timeout", type=float, default=600.0)
     args = ap.parse_args()
     asyncio.run(run(args))

@@ -637,6 +637,7 @@ def main():
     ap.add_argument("--rpc-port", type=int, default=0)
     ap.add_argument("--max-kv-blocks", type=int, default=None)
     ap.add_argument("--max-num-seqs", type=int, default=256)
+    ap.add_argument("--max-batched-tokens", type=int, default=8192)
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--no-graphs", action="store_true")
     args = ap.parse_args()
@@ -648,6 +649,7 @@ def main():
         rpc_host=args.rpc_host, rpc_port=args.rpc_port,
         max_kv_blocks=args.max_kv_blocks,
         engine_kwargs=dict(seed=args.seed, max_num_seqs=args.max_num_seqs,
+                           max_batched_tokens=args.max_batched_tokens,
                            enable_graphs=not args.no_graphs))
 
     async def run():
