@@ -112,11 +112,11 @@ def paged_attn_decode(q, k_cache, v_cache, block_tables, seq_lens, scale,
 
 
 def _prefill_tiles(cu_q):
-    """Host-side tile decomposition for the prefill kernel (64 q rows/WG)."""
+    """Host-side tile decomposition for the prefill kernel (128 q rows/WG)."""
     tile_seq, tile_q0 = [], []
     for s in range(len(cu_q) - 1):
         qlen = int(cu_q[s + 1]) - int(cu_q[s])
-        for q0 in range(0, qlen, 64):
+        for q0 in range(0, qlen, 128):
             tile_seq.append(s)
             tile_q0.append(q0)
     return tile_seq, tile_q0
