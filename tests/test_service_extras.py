@@ -105,3 +105,41 @@ async def test_replica_sync_and_master_takeover():
         await worker.stop()
         await b.stop()
         await registry.stop()
+
+
+@pytest.mark.anyio
+async def test_anthropic_messages_api():
+    master = make_master(policy="RR")
+    await master.start(serve_http=False)
+    worker = Worker("w0", "DEFAULT", **worker_kwargs(master))
+    try:
+        await worker.start()
+        await wait_for(lambda: master.instance_mgr.get("w0"))
+        client = await http_client(master)
+        r = await client.post("/v1/messages", json={
+            "model": "llama-tiny", "max_tokens": 5,
+            "system": "be brief",
+            "messages": [{"role": "user", "content": "hi"}],
+            "temperature": 0.0})
+        assert r.status_code == 200, r.text
+        body = r.json()
+        assert body["type"] == "message" and body["role"] == "assistant"
+        assert body["usage"]["output_tokens"] == 5
+        assert body["stop_reason"] == "max_tokens"
+        # streaming variant
+        events = []
+        async with client.stream("POST", "/v1/messages", json={
+                "model": "llama-tiny", "max_tokens": 4, "stream": True,
+                "messages": [{"role": "user", "content": "hi"}],
+                "temperature": 0.0}) as resp:
+            assert resp.status_code == 200
+            async for line in resp.aiter_lines():
+                if line.startswith("event: "):
+                    events.append(line[7:])
+        assert events[0] == "message_start"
+        assert "content_block_delta" in events
+        assert events[-1] == "message_stop"
+        await client.aclose()
+    finally:
+        await worker.stop()
+        await master.stop()

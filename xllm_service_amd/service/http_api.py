@@ -93,6 +93,110 @@ def build_app(master) -> FastAPI:
                           "owned_by": "xllm-service-amd"}
                          for m in master.served_models()]}
 
+    @app.post("/v1/messages")
+    async def anthropic_messages(request: Request):
+        """Anthropic-style messages API (the reference compiles an
+        `anthropic` proto alongside the OpenAI ones — SURVEY.md 2.9).
+        Translated onto the chat pipeline; supports system, max_tokens,
+        temperature, stop_sequences and SSE streaming."""
+        body = await request.json()
+        na = _not_ready()
+        if na is not None:
+            return na
+        messages = []
+        if body.get("system"):
+            messages.append({"role": "system", "content": body["system"]})
+        messages.extend(body.get("messages", []))
+        try:
+            prompt_text, token_ids = scheduler().tokenize_chat(messages)
+        except Exception as e:
+            return JSONResponse({"type": "error", "error": {
+                "type": "invalid_request_error", "message": str(e)}},
+                status_code=400)
+        params = dict(temperature=body.get("temperature", 1.0),
+                      top_p=body.get("top_p", 1.0),
+                      top_k=body.get("top_k", -1) or -1,
+                      max_tokens=body.get("max_tokens", 256),
+                      ignore_eos=False, stop_token_ids=[])
+        req = ServiceRequest(
+            service_request_id=make_request_id("msg"),
+            kind="chat", model=body.get("model") or master.model_id,
+            stream=bool(body.get("stream")), token_ids=token_ids,
+            prompt_text=prompt_text, params=params)
+        sch = scheduler()
+        try:
+            sch.schedule(req)
+            await sch.dispatch(req)
+        except SchedulerError as e:
+            return JSONResponse({"type": "error", "error": {
+                "type": "overloaded_error", "message": str(e)}},
+                status_code=e.status_code)
+
+        async def on_cancel(r):
+            await sch.cancel_request(r, reason="client disconnected")
+
+        if req.stream:
+            async def gen():
+                import json as _json
+                yield ("event: message_start\ndata: " + _json.dumps({
+                    "type": "message_start", "message": {
+                        "id": req.service_request_id, "type": "message",
+                        "role": "assistant", "content": [],
+                        "model": req.model}}) + "\n\n")
+                yield ("event: content_block_start\ndata: " + _json.dumps({
+                    "type": "content_block_start", "index": 0,
+                    "content_block": {"type": "text", "text": ""}}) + "\n\n")
+                from xllm_service_amd.tokenizer import IncrementalDecoder
+                dec = IncrementalDecoder(master.tokenizer)
+                import asyncio as _aio
+                try:
+                    while True:
+                        d = await _aio.wait_for(req.output_queue.get(), 600.0)
+                        if d.error:
+                            break
+                        text = dec.push(d.token_ids) if d.token_ids else ""
+                        if text:
+                            yield ("event: content_block_delta\ndata: " +
+                                   _json.dumps({
+                                       "type": "content_block_delta",
+                                       "index": 0,
+                                       "delta": {"type": "text_delta",
+                                                 "text": text}}) + "\n\n")
+                        if d.finished:
+                            stop = ("max_tokens"
+                                    if d.finish_reason == "length"
+                                    else "end_turn")
+                            yield ("event: message_delta\ndata: " +
+                                   _json.dumps({
+                                       "type": "message_delta",
+                                       "delta": {"stop_reason": stop},
+                                       "usage": {"output_tokens":
+                                                 d.usage_completion_tokens}})
+                                   + "\n\n")
+                            break
+                    yield ("event: message_stop\ndata: " + _json.dumps(
+                        {"type": "message_stop"}) + "\n\n")
+                except (GeneratorExit, Exception):
+                    await on_cancel(req)
+                    raise
+            return StreamingResponse(gen(), media_type="text/event-stream")
+
+        token_ids_out, usage, finish, err = \
+            await master.response_handler._collect(req)
+        if err:
+            return JSONResponse({"type": "error", "error": {
+                "type": "api_error", "message": err}}, status_code=500)
+        return {
+            "id": req.service_request_id, "type": "message",
+            "role": "assistant", "model": req.model,
+            "content": [{"type": "text",
+                         "text": master.tokenizer.decode(token_ids_out)}],
+            "stop_reason": ("max_tokens" if finish == "length"
+                            else "end_turn"),
+            "usage": {"input_tokens": usage["prompt_tokens"],
+                      "output_tokens": usage["completion_tokens"]},
+        }
+
     @app.post("/admin/reload_flags")
     async def reload_flags(request: Request):
         """Runtime-reloadable knobs (reference: brpc-reloadable target_ttft /
