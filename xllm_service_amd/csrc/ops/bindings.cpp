@@ -16,6 +16,11 @@ void launch_fused_add_rmsnorm(unsigned short*, unsigned short*,
                               hipStream_t);
 void launch_rope(unsigned short*, unsigned short*, const long*, const float*,
                  int, int, int, int, int, long, long, hipStream_t);
+void launch_fused_rope_cache(unsigned short*, const unsigned short*,
+                             const unsigned short*, unsigned short*,
+                             unsigned short*, const long*, const long*,
+                             const float*, int, int, int, int, int, long,
+                             long, int, hipStream_t);
 void launch_silu_and_mul(unsigned short*, const unsigned short*, int, int,
                          hipStream_t);
 void launch_gelu_and_mul(unsigned short*, const unsigned short*, int, int,
@@ -106,6 +111,29 @@ void rope(torch::Tensor positions, torch::Tensor q, torch::Tensor k,
   xllm::launch_rope(u16(q), u16(k), positions.data_ptr<long>(),
                     cos_sin.data_ptr<float>(), T, n_q, n_k, (int)head_dim,
                     (int)rot_dim, q.stride(0), k.stride(0), cur_stream());
+}
+
+void fused_rope_cache(torch::Tensor positions, torch::Tensor q,
+                      torch::Tensor k, torch::Tensor v,
+                      torch::Tensor k_cache, torch::Tensor v_cache,
+                      torch::Tensor slot_mapping, torch::Tensor cos_sin,
+                      long rot_dim) {
+  CHECK_BF16_ROWVIEW(q); CHECK_BF16_ROWVIEW(k); CHECK_BF16_ROWVIEW(v);
+  CHECK_BF16_CUDA(k_cache); CHECK_BF16_CUDA(v_cache);
+  TORCH_CHECK(positions.dtype() == torch::kLong && positions.is_cuda());
+  TORCH_CHECK(slot_mapping.dtype() == torch::kLong && slot_mapping.is_cuda());
+  TORCH_CHECK(cos_sin.dtype() == torch::kFloat && cos_sin.is_cuda());
+  TORCH_CHECK(k.stride(0) == v.stride(0));
+  const int T = positions.size(0);
+  const int n_kv = k_cache.size(1);
+  const int bs = k_cache.size(2);
+  const int D = k_cache.size(3);
+  const int n_q = q.numel() / T / D;
+  xllm::launch_fused_rope_cache(
+      u16(q), u16c(k), u16c(v), u16(k_cache), u16(v_cache),
+      positions.data_ptr<long>(), slot_mapping.data_ptr<long>(),
+      cos_sin.data_ptr<float>(), T, n_q, n_kv, D, (int)rot_dim, q.stride(0),
+      k.stride(0), bs, cur_stream());
 }
 
 void silu_and_mul(torch::Tensor out, torch::Tensor x) {
@@ -424,6 +452,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm);
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
   m.def("rope", &rope);
+  m.def("fused_rope_cache", &fused_rope_cache);
   m.def("silu_and_mul", &silu_and_mul);
   m.def("gelu_and_mul", &gelu_and_mul);
   m.def("reshape_and_cache", &reshape_and_cache);

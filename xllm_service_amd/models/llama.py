@@ -54,10 +54,9 @@ class LlamaAttention(nn.Module):
             -1, (self.n_kv_heads, self.head_dim))
         vh = qkv[:, q_sz + kv_sz:].unflatten(
             -1, (self.n_kv_heads, self.head_dim))
-        qh, kh = ops.rope(positions, qh, kh, cos_sin, self.head_dim,
-                          self.head_dim)
         k_cache, v_cache = kv_cache
-        ops.reshape_and_cache(kh, vh, k_cache, v_cache, meta.slot_mapping)
+        qh = ops.fused_rope_cache(positions, qh, kh, vh, k_cache, v_cache,
+                                  meta.slot_mapping, cos_sin, self.head_dim)
 
         np_, nd = meta.num_prefill_tokens, meta.num_decode_tokens
         out = torch.empty(T, q_sz, dtype=qkv.dtype, device=qkv.device)

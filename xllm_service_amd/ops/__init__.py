@@ -56,6 +56,21 @@ def rope(positions, q, k, cos_sin, head_dim, rot_dim):
     return ref.rope(positions, q, k, cos_sin, head_dim, rot_dim)
 
 
+def fused_rope_cache(positions, q, k, v, k_cache, v_cache, slot_mapping,
+                     cos_sin, rot_dim):
+    """GPU: one kernel ropes q in place and scatters roped-k + v into the
+    paged cache. CPU: composed from the reference ops."""
+    if q.is_cuda:
+        _require_ext()
+        _ops.fused_rope_cache(positions, q, k, v, k_cache, v_cache,
+                              slot_mapping, cos_sin, rot_dim)
+        return q
+    D = k_cache.shape[3]
+    q2, k2 = ref.rope(positions, q, k, cos_sin, D, rot_dim)
+    ref.reshape_and_cache(k2, v, k_cache, v_cache, slot_mapping)
+    return q2
+
+
 def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
     if x.is_cuda:
         _require_ext()
