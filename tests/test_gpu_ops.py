@@ -80,6 +80,31 @@ def test_rope(dev):
     _assert_close(got_k, want_k, atol=3e-2, label="rope k")
 
 
+def test_fused_rope_cache(dev):
+    torch.manual_seed(4)
+    T, Hq, Hkv, D, bs, blocks = 11, 32, 8, 128, 16, 8
+    qkv = torch.randn(T, (Hq + 2 * Hkv) * D, device=dev, dtype=torch.bfloat16)
+    q = qkv[:, :Hq * D].unflatten(-1, (Hq, D))
+    k = qkv[:, Hq * D:(Hq + Hkv) * D].unflatten(-1, (Hkv, D))
+    v = qkv[:, (Hq + Hkv) * D:].unflatten(-1, (Hkv, D))
+    kc = torch.zeros(blocks, Hkv, bs, D, device=dev, dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    pos = torch.randint(0, 2048, (T,), device=dev)
+    slots = torch.randperm(blocks * bs, device=dev)[:T]
+    table = ref.rope_table(D, 4096).to(dev)
+    # reference on CPU fp32
+    q_ref, k_ref = ref.rope(pos.cpu(), q.float().cpu().reshape(T, -1),
+                            k.float().cpu().reshape(T, -1), table.cpu(), D, D)
+    kc_ref = torch.zeros(blocks, Hkv, bs, D)
+    vc_ref = torch.zeros_like(kc_ref)
+    ref.reshape_and_cache(k_ref.reshape(T, Hkv, D), v.float().cpu(),
+                          kc_ref, vc_ref, slots.cpu())
+    ops.fused_rope_cache(pos, q, k, v, kc, vc, slots, table, D)
+    _assert_close(q.reshape(T, -1), q_ref, atol=3e-2, label="fused q")
+    _assert_close(kc, kc_ref, atol=3e-2, label="fused k_cache")
+    _assert_close(vc, vc_ref, atol=3e-2, label="fused v_cache")
+
+
 def test_silu_and_mul(dev):
     x = torch.randn(37, 2 * 14336, device=dev, dtype=torch.bfloat16)
     got = ops.silu_and_mul(x)
