@@ -88,32 +88,50 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
   const int* btab = block_tables + (long)seq * max_blocks_per_seq;
   const long head_stride = (long)PA_BS * PA_D;
 
-  for (int tile = wgp * 4 + wid; tile < n_tiles; tile += nwaves) {
+  // K fragments of a tile: clamped-address loads (no per-load branch —
+  // guide trap 4(c)); garbage lanes beyond seq_len are masked to -inf in
+  // the softmax. Prefetched one tile ahead so the HBM latency of K hides
+  // under the previous tile's V-stage + softmax + PV.
+  auto load_k = [&](bf16x8 (&kb)[2][4], int tile) {
+    const int t0 = tile * PA_KBLK;
+#pragma unroll
+    for (int n = 0; n < 2; n++) {
+      const int tok = t0 + n * 16 + frow;
+      const int tok_c = tok < seq_len ? tok : seq_len - 1;
+      const long rbase =
+          ((long)btab[tok_c / PA_BS] * n_kv_heads + kvh) * head_stride +
+          (long)(tok_c % PA_BS) * PA_D;
+#pragma unroll
+      for (int ks = 0; ks < 4; ks++)
+        kb[n][ks] = *reinterpret_cast<const bf16x8*>(
+            k_cache + rbase + ks * 32 + fcol8);
+    }
+  };
+
+  bf16x8 kb_a[2][4], kb_b[2][4];
+  const int tile0 = wgp * 4 + wid;
+  if (tile0 < n_tiles) load_k(kb_a, tile0);
+
+  for (int tile = tile0, phase = 0; tile < n_tiles;
+       tile += nwaves, phase ^= 1) {
+    bf16x8 (&kb)[2][4] = phase ? kb_b : kb_a;
+    bf16x8 (&kb_next)[2][4] = phase ? kb_a : kb_b;
+    if (tile + nwaves < n_tiles) load_k(kb_next, tile + nwaves);
     const int t0 = tile * PA_KBLK;
     const long base0 = ((long)btab[t0 / PA_BS] * n_kv_heads + kvh) * head_stride;
     const long base1 = (t0 + 16 < seq_len)
         ? ((long)btab[t0 / PA_BS + 1] * n_kv_heads + kvh) * head_stride
         : base0;
 
-    // ---- S = Q K^T (B-fragment loads directly from the paged cache) ------
+    // ---- S = Q K^T from the prefetched fragments --------------------------
     f32x4 s[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
 #pragma unroll
     for (int n = 0; n < 2; n++) {
-      const long rbase = (n ? base1 : base0) + (long)frow * PA_D;
-      const bool tok_ok = (t0 + n * 16 + frow) < seq_len;
 #pragma unroll
-      for (int ks = 0; ks < 4; ks++) {
-        bf16x8 bk;
-        if (tok_ok) {
-          bk = *reinterpret_cast<const bf16x8*>(
-              k_cache + rbase + ks * 32 + fcol8);
-        } else {
-          bk = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-        }
-        s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[ks], bk, s[n], 0, 0, 0);
-      }
+      for (int ks = 0; ks < 4; ks++)
+        s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[ks], kb[n][ks],
+                                                       s[n], 0, 0, 0);
     }
-
     // ---- stage V into the tr16 image (vector loads AND vector writes) -----
     {
       // lane covers token tv = lane>>1, d-half dv = (lane&1)*64
