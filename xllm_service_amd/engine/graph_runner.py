@@ -33,6 +33,9 @@ class DecodeGraphRunner:
         self.graphs: Dict[int, Tuple[torch.cuda.CUDAGraph, dict,
                                      torch.Tensor]] = {}
         self.pool = None
+        # pinned host staging (one per bucket) so the per-replay H2D copies
+        # are truly async; unpinned numpy->GPU copies serialize the stream
+        self._pinned: Dict[int, dict] = {}
         # per-request padded block-table rows (numpy, cheap incremental update)
         self._bt_rows: Dict[str, Tuple[np.ndarray, int]] = {}
 
@@ -76,6 +79,12 @@ class DecodeGraphRunner:
         if self.pool is None:
             self.pool = graph.pool()
         self.graphs[bs] = (graph, static, logits)
+        self._pinned[bs] = dict(
+            i64=torch.empty(3, bs, dtype=torch.long, pin_memory=True),
+            seq_lens=torch.empty(bs, dtype=torch.int32, pin_memory=True),
+            block_tables=torch.empty(bs, self.max_blocks, dtype=torch.int32,
+                                     pin_memory=True),
+        )
 
     # ------------------------------------------------------------------ run
     def bucket_for(self, n: int) -> Optional[int]:
@@ -108,19 +117,22 @@ class DecodeGraphRunner:
         n = len(input_ids)
         bs = self.bucket_for(n)
         graph, static, logits = self.graphs[bs]
-        dev = self.device
-        static["input_ids"][:n].copy_(
-            torch.from_numpy(input_ids), non_blocking=True)
-        static["positions"][:n].copy_(
-            torch.from_numpy(positions), non_blocking=True)
-        static["slot_mapping"][:n].copy_(
-            torch.from_numpy(slots), non_blocking=True)
+        pin = self._pinned[bs]
+        i64 = pin["i64"].numpy()
+        i64[0, :n] = input_ids
+        i64[1, :n] = positions
+        i64[2, :n] = slots
         if n < bs:  # neutralize padding rows
-            static["slot_mapping"][n:bs].fill_(-1)
-            static["seq_lens"][n:bs].fill_(16)
-        static["seq_lens"][:n].copy_(
-            torch.from_numpy(seq_lens), non_blocking=True)
-        bt = torch.from_numpy(np.stack(bt_rows))
-        static["block_tables"][:n].copy_(bt, non_blocking=True)
+            i64[2, n:bs] = -1
+        pin["seq_lens"].numpy()[:n] = seq_lens
+        if n < bs:
+            pin["seq_lens"].numpy()[n:bs] = 16
+        np.stack(bt_rows, out=pin["block_tables"].numpy()[:n])
+        static["input_ids"].copy_(pin["i64"][0], non_blocking=True)
+        static["positions"].copy_(pin["i64"][1], non_blocking=True)
+        static["slot_mapping"].copy_(pin["i64"][2], non_blocking=True)
+        static["seq_lens"].copy_(pin["seq_lens"], non_blocking=True)
+        static["block_tables"][:n].copy_(pin["block_tables"][:n],
+                                         non_blocking=True)
         graph.replay()
         return logits[:n]
