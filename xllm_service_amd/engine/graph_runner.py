@@ -79,10 +79,14 @@ class DecodeGraphRunner:
         if self.pool is None:
             self.pool = graph.pool()
         self.graphs[bs] = (graph, static, logits)
+        # zero-init: padding rows of the FULL-width copies must hold valid
+        # token ids / block ids (stale entries from a larger batch are fine
+        # — they were valid once)
         self._pinned[bs] = dict(
-            i64=torch.empty(3, bs, dtype=torch.long, pin_memory=True),
-            seq_lens=torch.empty(bs, dtype=torch.int32, pin_memory=True),
-            block_tables=torch.empty(bs, self.max_blocks, dtype=torch.int32,
+            i64=torch.zeros(3, bs, dtype=torch.long, pin_memory=True),
+            seq_lens=torch.full((bs,), 16, dtype=torch.int32,
+                                pin_memory=True),
+            block_tables=torch.zeros(bs, self.max_blocks, dtype=torch.int32,
                                      pin_memory=True),
         )
 
