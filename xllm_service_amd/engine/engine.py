@@ -6,7 +6,7 @@ engine/worker.py (RPC); bench.py and tests drive it directly.
 from __future__ import annotations
 
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 import torch

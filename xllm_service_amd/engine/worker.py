@@ -24,7 +24,6 @@ import asyncio
 import concurrent.futures
 import logging
 import queue
-import socket
 import threading
 import time
 from typing import Any, Dict, List, Optional

@@ -15,8 +15,6 @@ bench shapes are identical — noted as a fidelity simplification).
 from __future__ import annotations
 
 import math
-from typing import List, Optional, Tuple
-
 import torch
 import torch.nn as nn
 

@@ -203,7 +203,6 @@ def skinny_gemm(a, b, bias=None):
     """Split-K weight-streaming GEMM for decode batches (M <= 128).
     GPU-only; callers fall back to F.linear elsewhere."""
     _require_ext()
-    import math
     sk = (a.shape[1] + 1023) // 1024
     mpad = (a.shape[0] + 15) // 16 * 16
     ws = torch.empty(sk, mpad, b.shape[0], dtype=torch.float32,

@@ -14,8 +14,8 @@ import argparse
 import asyncio
 import logging
 import socket
-from dataclasses import dataclass, field
-from typing import Any, Dict, List, Optional
+from dataclasses import dataclass
+from typing import List, Optional
 
 import uvicorn
 

@@ -7,8 +7,8 @@ from __future__ import annotations
 
 import logging
 import time
-from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Tuple
+from dataclasses import dataclass
+from typing import Dict, List, Optional
 
 from .instance_mgr import Instance, InstanceMgr
 from .kvcache_mgr import GlobalKVCacheMgr

@@ -6,7 +6,6 @@ from __future__ import annotations
 
 import asyncio
 import json
-import time
 from typing import AsyncIterator, Optional
 
 from xllm_service_amd.tokenizer import IncrementalDecoder, Tokenizer

@@ -15,7 +15,6 @@ Responsibilities (reference: scheduler/scheduler.{h,cpp}):
 """
 from __future__ import annotations
 
-import asyncio
 import logging
 import time
 from typing import Any, Dict, List, Optional
@@ -29,7 +28,6 @@ from .kvcache_mgr import GlobalKVCacheMgr
 from .policies import LoadBalancePolicy, SloAwarePolicy
 from .request import GenerationDelta, ServiceRequest
 from .tracer import RequestTracer
-from .types import InstanceType
 
 log = logging.getLogger("xllm.scheduler")
 

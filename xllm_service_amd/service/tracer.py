@@ -6,7 +6,7 @@ import json
 import os
 import threading
 import time
-from typing import Any, Optional
+from typing import Any
 
 
 class RequestTracer:
