@@ -54,8 +54,19 @@ class LLMEngine:
                  enable_graphs: bool = True,
                  max_model_len: int = 4096,
                  swap_space_mb: int = 1024,
+                 tp_size: int = 1,
+                 load_state_path: Optional[str] = None,
                  seed: int = 0):
         self.cfg: ModelConfig = get_config(model_name)
+        # tensor parallelism: every rank of the TP group builds its shard;
+        # rank 0 drives scheduling, followers replay broadcast batches
+        # (model_runner.follower_step). torch.distributed must carry
+        # RANK/WORLD_SIZE (torchrun-style) when tp_size > 1.
+        self.tp_size = tp_size
+        if tp_size > 1:
+            from xllm_service_amd.distributed import parallel_state as ps
+            ps.init_distributed()
+            ps.init_tensor_parallel(tp_size)
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
         self.device = torch.device(device)
@@ -66,7 +77,17 @@ class LLMEngine:
         self.dtype = dtype
 
         self.model = create_model(self.cfg, dtype=dtype).to(self.device)
-        self.model.random_init(seed)
+        if load_state_path:
+            import os
+            full_sd = torch.load(load_state_path, map_location="cpu")
+            if tp_size > 1:
+                from xllm_service_amd.distributed import parallel_state as ps
+                from xllm_service_amd.distributed.layers import                     shard_llama_state_dict
+                full_sd = shard_llama_state_dict(full_sd, self.cfg, tp_size,
+                                                 ps.tp_rank())
+            self.model.load_state_dict(full_sd)
+        else:
+            self.model.random_init(seed)
         self.model = self.model.eval()
 
         n_kv_local = getattr(self.model, "local_kv_heads", self.cfg.num_kv_heads)
@@ -315,6 +336,11 @@ class LLMEngine:
                     len(block_ids), *c.shape[1:])
                 c[idx] = t.to(self.device)
                 off += per
+
+    def follower_loop(self) -> None:
+        """TP ranks > 0: replay broadcast batches until rank 0 stops."""
+        while self.runner.follower_step():
+            pass
 
     # ---- convenience (tests, smoke) ----------------------------------------
     def generate(self, prompts: List[List[int]],

@@ -130,6 +130,31 @@ class ModelRunner:
 
         dev = self.device
         np_ = cu_q[-1]
+        from xllm_service_amd.distributed import parallel_state as ps
+        if ps.tp_size() > 1 and ps.tp_rank() == 0:
+            import numpy as _np
+            def tables_np(tables):
+                if not tables:
+                    return None
+                w = max(len(t) for t in tables)
+                out = _np.zeros((len(tables), w), dtype=_np.int32)
+                for i, t in enumerate(tables):
+                    out[i, :len(t)] = t
+                return out
+            self._tp_batch = dict(
+                kind="eager", np=np_, nd=len(plan.decodes),
+                input_ids=_np.asarray(tokens, dtype=_np.int64),
+                positions=_np.asarray(positions, dtype=_np.int64),
+                slots=_np.asarray(slots, dtype=_np.int64),
+                cu_q=_np.asarray(cu_q, dtype=_np.int32)
+                if plan.prefills else None,
+                p_seq_lens=_np.asarray(p_seq_lens, dtype=_np.int32)
+                if plan.prefills else None,
+                p_tables=tables_np(p_tables) if plan.prefills else None,
+                d_seq_lens=_np.asarray(d_seq_lens, dtype=_np.int32)
+                if plan.decodes else None,
+                d_tables=tables_np(d_tables) if plan.decodes else None,
+            )
         meta = AttnMetadata(
             num_prefill_tokens=np_,
             num_decode_tokens=len(plan.decodes),
@@ -200,6 +225,64 @@ class ModelRunner:
             out.append(int(torch.multinomial(probs, 1, generator=gen)))
         return out
 
+    # ---- tensor-parallel batch broadcast ------------------------------------
+    def _tp_bcast(self, obj):
+        from xllm_service_amd.distributed import parallel_state as ps
+        if ps.tp_size() <= 1:
+            return
+        import torch.distributed as dist
+        box = [obj]
+        dist.broadcast_object_list(box, src=0, group=ps.tp_group())
+
+    def follower_step(self) -> bool:
+        """TP rank > 0: receive one batch from rank 0 and run the forward
+        (participating in the TP collectives). Returns False on stop."""
+        from xllm_service_amd.distributed import parallel_state as ps
+        import torch.distributed as dist
+        box = [None]
+        dist.broadcast_object_list(box, src=0, group=ps.tp_group())
+        obj = box[0]
+        if obj is None or obj.get("kind") == "stop":
+            return False
+        with torch.inference_mode():
+            if obj["kind"] == "graph":
+                logits = self.graph_runner.run(
+                    obj["input_ids"], obj["positions"], obj["slots"],
+                    obj["seq_lens"], list(obj["bt_rows"]))
+            else:
+                dev = self.device
+                meta = AttnMetadata(
+                    num_prefill_tokens=obj["np"],
+                    num_decode_tokens=obj["nd"],
+                    slot_mapping=torch.from_numpy(obj["slots"]).to(dev),
+                    cu_q=(torch.from_numpy(obj["cu_q"]).to(dev)
+                          if obj["cu_q"] is not None else None),
+                    prefill_seq_lens=(
+                        torch.from_numpy(obj["p_seq_lens"]).to(dev)
+                        if obj["p_seq_lens"] is not None else None),
+                    prefill_block_tables=(
+                        torch.from_numpy(obj["p_tables"]).to(dev)
+                        if obj["p_tables"] is not None else None),
+                    decode_seq_lens=(
+                        torch.from_numpy(obj["d_seq_lens"]).to(dev)
+                        if obj["d_seq_lens"] is not None else None),
+                    decode_block_tables=(
+                        torch.from_numpy(obj["d_tables"]).to(dev)
+                        if obj["d_tables"] is not None else None),
+                )
+                input_ids = torch.from_numpy(obj["input_ids"]).to(dev)
+                positions = torch.from_numpy(obj["positions"]).to(dev)
+                hidden = self.model(input_ids, positions, self.kv_caches,
+                                    meta)
+                if obj["rows"]:
+                    sel = hidden[torch.tensor(obj["rows"], dtype=torch.long,
+                                              device=dev)]
+                    self.model.compute_logits(sel)  # joins the all-gather
+        return True
+
+    def stop_followers(self):
+        self._tp_bcast({"kind": "stop"})
+
     # ---- one step -----------------------------------------------------------
     @torch.inference_mode()
     def execute(self, plan: StepPlan, bm) -> Dict[str, int]:
@@ -212,6 +295,19 @@ class ModelRunner:
                         for s in plan.decodes)):
             return self._execute_decode_graph(plan, bm)
         input_ids, positions, meta = self._build_batch(plan, bm)
+        tp_batch = getattr(self, "_tp_batch", None)
+        self._tp_batch = None
+        if tp_batch is not None:
+            # rows that will need logits (followers must join the logits
+            # all-gather with the same selection size)
+            rows_pre = []
+            for i, sp in enumerate(plan.prefills):
+                if sp.chunk_start + sp.chunk_len >= sp.seq.prompt_len:
+                    rows_pre.append(int(meta.cu_q[i + 1]) - 1)
+            rows_pre.extend(meta.num_prefill_tokens + j
+                            for j in range(len(plan.decodes)))
+            tp_batch["rows"] = rows_pre
+            self._tp_bcast(tp_batch)
         inputs_embeds = None
         if any(sp.seq.mm_embeds is not None for sp in plan.prefills):
             inputs_embeds = self._merge_mm_embeds(plan, input_ids, meta)
@@ -278,6 +374,11 @@ class ModelRunner:
             slots[i] = bm.append_slot(seq)
             seq_lens[i] = seq.total_len
             bt_rows.append(gr.block_row(seq))
+        from xllm_service_amd.distributed import parallel_state as ps
+        if ps.tp_size() > 1 and ps.tp_rank() == 0:
+            self._tp_bcast(dict(kind="graph", input_ids=input_ids,
+                                positions=positions, slots=slots,
+                                seq_lens=seq_lens, bt_rows=bt_rows))
         logits = gr.run(input_ids, positions, slots, seq_lens, bt_rows)
         tokens = self._sample(logits, seqs)
         lps = self._logprobs(logits, seqs, tokens)

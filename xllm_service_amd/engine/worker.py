@@ -23,6 +23,7 @@ from __future__ import annotations
 import asyncio
 import concurrent.futures
 import logging
+import os
 import queue
 import threading
 import time
@@ -639,8 +640,29 @@ def main():
     ap.add_argument("--max-batched-tokens", type=int, default=8192)
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--no-graphs", action="store_true")
+    ap.add_argument("--tp", type=int, default=1,
+                    help="tensor-parallel group size; launch one process "
+                         "per rank with RANK/WORLD_SIZE/MASTER_ADDR set "
+                         "(torchrun-style); rank 0 serves RPC, ranks >0 "
+                         "run the follower loop")
+    ap.add_argument("--load-state", default=None,
+                    help="full (tp=1) state-dict .pt to shard-load")
     args = ap.parse_args()
     logging.basicConfig(level=logging.INFO)
+
+    if args.tp > 1 and int(os.environ.get("RANK", "0")) > 0:
+        # TP follower rank: shard engine + replay loop, no service plumbing
+        from .engine import LLMEngine
+        eng = LLMEngine(args.model, device=args.device,
+                        max_kv_blocks=args.max_kv_blocks,
+                        tp_size=args.tp, load_state_path=args.load_state,
+                        seed=args.seed,
+                        enable_graphs=not args.no_graphs,
+                        max_num_seqs=args.max_num_seqs,
+                        max_batched_tokens=args.max_batched_tokens)
+        logging.info("TP follower rank %s up", os.environ["RANK"])
+        eng.follower_loop()
+        return
 
     worker = Worker(
         args.name, args.type, model=args.model, device=args.device,
@@ -649,7 +671,9 @@ def main():
         max_kv_blocks=args.max_kv_blocks,
         engine_kwargs=dict(seed=args.seed, max_num_seqs=args.max_num_seqs,
                            max_batched_tokens=args.max_batched_tokens,
-                           enable_graphs=not args.no_graphs))
+                           enable_graphs=not args.no_graphs,
+                           tp_size=args.tp,
+                           load_state_path=args.load_state))
 
     async def run():
         await worker.start()
