@@ -73,6 +73,7 @@ def _tp_worker(rank, world, port, sd_path, out_path):
         logits = model.compute_logits(hidden[-1:])
     if rank == 0:
         torch.save({"hidden": hidden, "logits": logits}, out_path)
+    ps.shutdown()
 
 
 def test_tp2_matches_single(tmp_path):

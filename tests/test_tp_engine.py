@@ -50,6 +50,7 @@ def _tp_rank(rank, world, port, ckpt, out_path):
     cfg = get_config(MODEL)
     eng = LLMEngine(MODEL, device="cpu", max_kv_blocks=128, tp_size=world,
                     load_state_path=ckpt)
+    from xllm_service_amd.distributed import parallel_state as ps
     if rank == 0:
         out = eng.generate(_prompts(cfg),
                            SamplingParams(max_tokens=6, ignore_eos=True))
@@ -57,6 +58,7 @@ def _tp_rank(rank, world, port, ckpt, out_path):
         torch.save(out, out_path)
     else:
         eng.follower_loop()
+    ps.shutdown()
 
 
 def test_tp2_engine_group_matches_single(tmp_path):

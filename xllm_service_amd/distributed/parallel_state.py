@@ -33,6 +33,18 @@ def init_distributed(backend: str | None = None, timeout_s: int = 120) -> None:
         torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
 
 
+def shutdown() -> None:
+    """Barrier + destroy the default group. Spawned worker processes must
+    call this before exiting: gloo's background threads abort the process
+    ("terminate called without an active exception") if torn down by
+    process exit instead."""
+    global _TP_GROUP, _TP_RANK, _TP_SIZE
+    if dist.is_initialized():
+        dist.barrier()
+        dist.destroy_process_group()
+    _TP_GROUP, _TP_RANK, _TP_SIZE = None, 0, 1
+
+
 def init_tensor_parallel(tp_size: int = 1, group=None) -> None:
     """Declare the TP group for model layers created afterwards."""
     global _TP_GROUP, _TP_RANK, _TP_SIZE
