@@ -24,8 +24,8 @@ class SeqStatus(enum.Enum):
                         SeqStatus.FINISHED_ABORT)
 
 
-@dataclass
-class Sequence:
+@dataclass(eq=False)   # identity eq/hash: `seq in running` must not compare
+class Sequence:        # token lists element-wise (O(len) per membership test)
     request_id: str
     prompt_token_ids: List[int]
     params: SamplingParams
