@@ -88,40 +88,85 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
   const int* btab = block_tables + (long)seq * max_blocks_per_seq;
   const long head_stride = (long)PA_BS * PA_D;
 
-  // K fragments of a tile: clamped-address loads (no per-load branch —
-  // guide trap 4(c)); garbage lanes beyond seq_len are masked to -inf in
-  // the softmax. Prefetched one tile ahead so the HBM latency of K hides
-  // under the previous tile's V-stage + softmax + PV.
-  auto load_k = [&](bf16x8 (&kb)[2][4], int tile) {
+  // K/V loads of a tile: clamped-address loads (no per-load branch — guide
+  // trap 4(c)); lanes beyond seq_len re-read the last token. K garbage is
+  // masked to -inf in the softmax; V garbage is harmless because its P
+  // weight is exactly 0 (finite * 0 = 0 — clamping, unlike zero-fill via a
+  // branch, keeps the load stream unconditional).
+  // Cache-block index per 16-token half: min(t0/16+n, last_blk) is exact
+  // for clamped lanes too (the boundary tile's in-range lanes live in the
+  // same block the clamp resolves to) and — crucially — UNIFORM across the
+  // wave, so btab[...] compiles to a scalar load instead of a per-lane
+  // VMEM load sitting in the K/V address chain every tile.
+  const int last_blk = (seq_len - 1) / PA_BS;
+  auto load_blks = [&](int tile, int& b0, int& b1) {
+    // uniform per tile: min(2*tile+n, last_blk) is exact for clamped lanes
+    b0 = btab[min(tile * (PA_KBLK / PA_BS), last_blk)];
+    b1 = btab[min(tile * (PA_KBLK / PA_BS) + 1, last_blk)];
+  };
+  auto base_of = [&](int blk) -> long {
+    return ((long)blk * n_kv_heads + kvh) * head_stride;
+  };
+  auto load_k = [&](bf16x8 (&kb)[2][4], int tile, int b0, int b1) {
     const int t0 = tile * PA_KBLK;
 #pragma unroll
     for (int n = 0; n < 2; n++) {
       const int tok = t0 + n * 16 + frow;
       const int tok_c = tok < seq_len ? tok : seq_len - 1;
-      const long rbase =
-          ((long)btab[tok_c / PA_BS] * n_kv_heads + kvh) * head_stride +
-          (long)(tok_c % PA_BS) * PA_D;
+      const long rbase = base_of(n ? b1 : b0) +
+                         (long)(tok_c % PA_BS) * PA_D;
 #pragma unroll
       for (int ks = 0; ks < 4; ks++)
         kb[n][ks] = *reinterpret_cast<const bf16x8*>(
             k_cache + rbase + ks * 32 + fcol8);
     }
   };
+  // lane covers token tv = lane>>1, d-half dv = (lane&1)*64
+  const int tv = lane >> 1;
+  const int dv = (lane & 1) * 64;
+  const int koff = ((tv >> 2) & 1) * 256 + (tv >> 3) * 64 + (tv & 3) * 16;
+  auto stage_v = [&](int tile, int b0, int b1) {
+    // V is loaded and written to the tr16 LDS image in-tile with transient
+    // registers (8-16 live): K owns the persistent prefetch buffers, and
+    // holding V too pushes past 256 regs into scratch spills (measured).
+    const int tok = tile * PA_KBLK + tv;
+    const int tok_c = tok < seq_len ? tok : seq_len - 1;
+    const long vbase = base_of(tv >> 4 ? b1 : b0) +
+                       (long)(tok_c % PA_BS) * PA_D + dv;
+#pragma unroll
+    for (int mseg = 0; mseg < 8; mseg++) {
+      const ushort8_t vv = *reinterpret_cast<const ushort8_t*>(
+          v_cache + vbase + mseg * 8);
+      const int d0 = dv + mseg * 8;
+      *reinterpret_cast<ushort8_t*>(
+          &vt_lds[wid][(d0 >> 4) * 512 + koff + (d0 & 15)]) = vv;
+    }
+  };
 
+  // Pipeline: K is double-buffered (each tile's K loads issue a FULL tile
+  // ahead), V stages in-tile through transient regs, and the uniform block
+  // ids both need are scalar-loaded TWO iterations ahead (cb/fb/gb
+  // rotation) so the btab lookup never sits in any load's address chain.
   bf16x8 kb_a[2][4], kb_b[2][4];
   const int tile0 = wgp * 4 + wid;
-  if (tile0 < n_tiles) load_k(kb_a, tile0);
+  int cb0 = 0, cb1 = 0;      // ids for the tile being computed
+  int fb0 = 0, fb1 = 0;      // ids for the next tile (K prefetch target)
+  if (tile0 < n_tiles) {
+    load_blks(tile0, cb0, cb1);
+    load_k(kb_a, tile0, cb0, cb1);
+    if (tile0 + nwaves < n_tiles) load_blks(tile0 + nwaves, fb0, fb1);
+  }
 
   for (int tile = tile0, phase = 0; tile < n_tiles;
        tile += nwaves, phase ^= 1) {
+    const int t0 = tile * PA_KBLK;
     bf16x8 (&kb)[2][4] = phase ? kb_b : kb_a;
     bf16x8 (&kb_next)[2][4] = phase ? kb_a : kb_b;
-    if (tile + nwaves < n_tiles) load_k(kb_next, tile + nwaves);
-    const int t0 = tile * PA_KBLK;
-    const long base0 = ((long)btab[t0 / PA_BS] * n_kv_heads + kvh) * head_stride;
-    const long base1 = (t0 + 16 < seq_len)
-        ? ((long)btab[t0 / PA_BS + 1] * n_kv_heads + kvh) * head_stride
-        : base0;
+    // ids for tile+2*nwaves: a full iteration before their first use
+    int gb0 = 0, gb1 = 0;
+    if (tile + 2 * nwaves < n_tiles) load_blks(tile + 2 * nwaves, gb0, gb1);
+    // K loads for the next tile: full-tile latency distance
+    if (tile + nwaves < n_tiles) load_k(kb_next, tile + nwaves, fb0, fb1);
 
     // ---- S = Q K^T from the prefetched fragments --------------------------
     f32x4 s[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
@@ -132,29 +177,9 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
         s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[ks], kb[n][ks],
                                                        s[n], 0, 0, 0);
     }
-    // ---- stage V into the tr16 image (vector loads AND vector writes) -----
-    {
-      // lane covers token tv = lane>>1, d-half dv = (lane&1)*64
-      const int tv = lane >> 1;
-      const int dv = (lane & 1) * 64;
-      const bool vok = (t0 + tv) < seq_len;
-      const long vbase = ((tv < 16) ? base0 + (long)tv * PA_D
-                                    : base1 + (long)(tv - 16) * PA_D) + dv;
-      const int koff = ((tv >> 2) & 1) * 256 + (tv >> 3) * 64 + (tv & 3) * 16;
-#pragma unroll
-      for (int mseg = 0; mseg < 8; mseg++) {
-        const int d0 = dv + mseg * 8;
-        ushort8_t vv;
-        if (vok) {
-          vv = *reinterpret_cast<const ushort8_t*>(v_cache + vbase + mseg * 8);
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; j++) vv.x[j] = 0;
-        }
-        *reinterpret_cast<ushort8_t*>(
-            &vt_lds[wid][(d0 >> 4) * 512 + koff + (d0 & 15)]) = vv;
-      }
-    }
+    // ---- stage V into the tr16 image (in-tile, transient regs) ------------
+    stage_v(tile, cb0, cb1);
+    cb0 = fb0; cb1 = fb1; fb0 = gb0; fb1 = gb1;
 
     // ---- mask + online softmax (rows are heads) ---------------------------
     float p_val[2][4];
@@ -210,8 +235,8 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
       const unsigned vaddr =
           (unsigned)(unsigned long long)(&vt_lds[wid][0]) +
           ((lane >> 4) * 128u + (lane & 15) * 8u);
-      // two half-batches of 8 tr reads keep 16 (not 32) result VGPRs live:
-      // the kernel is VGPR-occupancy-bound (3 waves/SIMD at <=168 regs)
+      // one half-batch of 8 tr reads live at a time (16 result VGPRs,
+      // reused across halves: the kernel is register-occupancy-bound)
 #define PA_TR8(OFF0)                                                       \
       asm volatile(                                                        \
           "ds_read_b64_tr_b16 %[t0], %[a] offset:" #OFF0 "+0\n\t"          \
@@ -229,39 +254,19 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
           : [a] "v"(vaddr)                                                 \
           : "memory")
       u32x2 tr[8];
-      u32x2 tr2[8];
 #pragma unroll
       for (int half = 0; half < 2; half++) {
-        if (half == 0) { PA_TR8(0); } else {
-          // second half into separate regs so both batches stay in flight
-#define PA_SWAP_TR tr2
-          asm volatile(
-              "ds_read_b64_tr_b16 %[t0], %[a] offset:4096+0\n\t"
-              "ds_read_b64_tr_b16 %[t1], %[a] offset:4096+512\n\t"
-              "ds_read_b64_tr_b16 %[t2], %[a] offset:4096+1024\n\t"
-              "ds_read_b64_tr_b16 %[t3], %[a] offset:4096+1536\n\t"
-              "ds_read_b64_tr_b16 %[t4], %[a] offset:4096+2048\n\t"
-              "ds_read_b64_tr_b16 %[t5], %[a] offset:4096+2560\n\t"
-              "ds_read_b64_tr_b16 %[t6], %[a] offset:4096+3072\n\t"
-              "ds_read_b64_tr_b16 %[t7], %[a] offset:4096+3584\n\t"
-              "s_waitcnt lgkmcnt(0)"
-              : [t0] "=&v"(tr2[0]), [t1] "=&v"(tr2[1]), [t2] "=&v"(tr2[2]),
-                [t3] "=&v"(tr2[3]), [t4] "=&v"(tr2[4]), [t5] "=&v"(tr2[5]),
-                [t6] "=&v"(tr2[6]), [t7] "=&v"(tr2[7])
-              : [a] "v"(vaddr)
-              : "memory");
-        }
+        if (half == 0) { PA_TR8(0); } else { PA_TR8(4096); }
         __builtin_amdgcn_sched_barrier(0);  // MFMAs stay below the wait
 #pragma unroll
         for (int nn = 0; nn < 4; nn++) {
           const int n = half * 4 + nn;
           bf16x8 bv;
           unsigned* bw = reinterpret_cast<unsigned*>(&bv);
-          u32x2* src = half ? tr2 : tr;
-          bw[0] = src[2 * nn][0];
-          bw[1] = src[2 * nn][1];
-          bw[2] = src[2 * nn + 1][0];
-          bw[3] = src[2 * nn + 1][1];
+          bw[0] = tr[2 * nn][0];
+          bw[1] = tr[2 * nn][1];
+          bw[2] = tr[2 * nn + 1][0];
+          bw[3] = tr[2 * nn + 1][1];
           o_acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bv, o_acc[n],
                                                              0, 0, 0);
         }
