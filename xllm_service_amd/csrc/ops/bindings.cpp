@@ -209,7 +209,7 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
   const int W = xllm::paged_attn_decode_partitions(num_seqs, n_kv) * 4;
   auto fopt = q.options().dtype(torch::kFloat);
   torch::Tensor ws_ml, ws_o;
-  if (D == 128) {
+  if (D == 128 && W > 4) {  // P == 1 merges in-kernel: no workspace
     ws_ml = torch::empty({(long)num_seqs * n_kv * W * G * 2}, fopt);
     ws_o = torch::empty({(long)num_seqs * n_kv * W * G * 128}, fopt);
   } else {  // small-head fallback kernel merges in-workgroup

@@ -41,7 +41,9 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
     const int* __restrict__ seq_lens,        // [num_seqs]
     const float scale,
     const int n_kv_heads, const int max_blocks_per_seq,
-    const long q_stride, const int P) {     // P partitions per (seq, kvh)
+    const long q_stride, const int P,       // P partitions per (seq, kvh)
+    unsigned short* __restrict__ out,       // [S, n_qheads, D] (P == 1)
+    const long out_stride) {
   const int sk = blockIdx.x / P;            // (seq, kvh) unit
   const int wgp = blockIdx.x % P;           // partition within the unit
   const int seq = sk / n_kv_heads;
@@ -275,6 +277,59 @@ __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
     }
   }
 
+  if (P == 1) {
+    // ---- fused merge: combine the 4 per-wave partials through LDS and
+    // write the output directly — saves the merge-kernel launch plus the
+    // global ws_o/ws_ml round-trip (~8 MB at batch 64). vt_lds (32 KB) and
+    // p_lds are dead by now and are reused as the merge buffers.
+    float* mo = reinterpret_cast<float*>(&vt_lds[0][0]);    // [4][G][128]
+    float* mml = reinterpret_cast<float*>(&p_lds[0][0][0]); // [4][G][2]
+    __syncthreads();
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      const int row = crow4 + r;
+      if (row < G) {
+        if (frow == 0) {
+          mml[(wid * G + row) * 2 + 0] = m_r[r];
+          mml[(wid * G + row) * 2 + 1] = l_r[r];
+        }
+#pragma unroll
+        for (int n = 0; n < 8; n++)
+          mo[((wid * G + row) * PA_D) + n * 16 + frow] = o_acc[n][r];
+      }
+    }
+    __syncthreads();
+    if (wid == 0) {
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        const int row = crow4 + r;
+        if (row >= G) continue;
+        float m_star = -INFINITY;
+#pragma unroll
+        for (int w = 0; w < 4; w++)
+          m_star = fmaxf(m_star, mml[(w * G + row) * 2]);
+        float wgt[4], l_tot = 0.0f;
+#pragma unroll
+        for (int w = 0; w < 4; w++) {
+          const float mw = mml[(w * G + row) * 2];
+          wgt[w] = (mw == -INFINITY) ? 0.0f : __expf(mw - m_star);
+          l_tot += wgt[w] * mml[(w * G + row) * 2 + 1];
+        }
+        const float inv = (l_tot > 0.0f) ? 1.0f / l_tot : 0.0f;
+#pragma unroll
+        for (int n = 0; n < 8; n++) {
+          float o = 0.0f;
+#pragma unroll
+          for (int w = 0; w < 4; w++)
+            o += wgt[w] * mo[((w * G + row) * PA_D) + n * 16 + frow];
+          out[(long)seq * out_stride + (long)(kvh * G + row) * PA_D +
+              n * 16 + frow] = f32_to_bf16(o * inv);
+        }
+      }
+    }
+    return;
+  }
+
   // ---- write this wave's partial (no cross-wave sync: the tiny merge
   // kernel below combines the P*4 partials per (seq, kvh) unit) ------------
   const long part = ((long)sk * nwaves + wgp * 4 + wid);
@@ -450,10 +505,11 @@ __global__ __launch_bounds__(256) void paged_attn_decode_small_kernel(
       hipLaunchKernelGGL((paged_attn_decode_kernel<GV>), grid, block, 0,       \
                          stream, ws_ml, ws_o, q, k_cache, v_cache,             \
                          block_tables, seq_lens, scale, n_kv_heads,            \
-                         max_blocks_per_seq, q_stride, P);                     \
-      hipLaunchKernelGGL(paged_attn_decode_merge_kernel, mgrid, block, 0,      \
-                         stream, out, ws_ml, ws_o, GV, n_kv_heads, P * 4,      \
-                         out_stride, total);                                   \
+                         max_blocks_per_seq, q_stride, P, out, out_stride);    \
+      if (P > 1)                                                               \
+        hipLaunchKernelGGL(paged_attn_decode_merge_kernel, mgrid, block, 0,    \
+                           stream, out, ws_ml, ws_o, GV, n_kv_heads, P * 4,    \
+                           out_stride, total);                                 \
     } else {                                                                   \
       hipLaunchKernelGGL((paged_attn_decode_small_kernel<GV>), grid, block, 0, \
                          stream, out, q, k_cache, v_cache, block_tables,       \
