@@ -63,8 +63,10 @@ def main():
         x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
         W = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
         t = evt_time(lambda: xops.skinny_gemm(x, W), flush=flush)
+        tw = evt_time(lambda: xops.skinny_gemm(x, W))
         mb = N * K * 2 / 1e6
-        print(f"skinny {tag:8s}: {t:7.1f}us ({mb/t*1e3/1e3:.2f} TB/s)")
+        print(f"skinny {tag:8s}: warm {tw:7.1f}us  cold {t:7.1f}us "
+              f"({mb/t*1e3/1e3:.2f} TB/s cold)")
 
     print("== paged attention decode (batch 64, seq 1024, 8 kv heads, G=4) ==")
     from xllm_service_amd import ops

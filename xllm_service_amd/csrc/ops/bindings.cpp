@@ -48,6 +48,7 @@ void launch_greedy_sample(long*, const unsigned short*, int, int, hipStream_t);
 void launch_mfma_gemm(unsigned short*, const unsigned short*,
                       const unsigned short*, const unsigned short*, int, int,
                       int, hipStream_t);
+int skinny_gemm_splitk(int M, int N, int K);
 void launch_skinny_gemm(unsigned short*, const unsigned short*,
                         const unsigned short*, const unsigned short*, float*,
                         int, int, int, hipStream_t);
@@ -276,15 +277,20 @@ torch::Tensor mfma_gemm(torch::Tensor a, torch::Tensor b,
 }
 
 torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor b,
-                          c10::optional<torch::Tensor> bias,
-                          torch::Tensor ws) {
-  // a: [M, K]; b: [N, K]; ws: fp32 [SK, ceil(M/16)*16, N]; returns [M, N]
+                          c10::optional<torch::Tensor> bias) {
+  // a: [M, K]; b: [N, K]; returns [M, N]
   CHECK_BF16_CUDA(a); CHECK_BF16_CUDA(b);
-  TORCH_CHECK(ws.dtype() == torch::kFloat && ws.is_cuda());
   const int M = a.size(0), K = a.size(1), N = b.size(0);
-  TORCH_CHECK(a.size(1) == b.size(1) && M <= 128);
+  TORCH_CHECK(a.size(1) == b.size(1) && M <= 256);
   TORCH_CHECK(K % 64 == 0 && N % 4 == 0);
   auto c = torch::empty({M, N}, a.options());
+  const int SK = xllm::skinny_gemm_splitk(M, N, K);
+  const long mpad = ((M + 15) / 16) * 16;
+  torch::Tensor ws;
+  if (SK > 1)
+    ws = torch::empty({(long)SK * mpad * N}, a.options().dtype(torch::kFloat));
+  else
+    ws = torch::empty({1}, a.options().dtype(torch::kFloat));
   const unsigned short* bp = nullptr;
   if (bias.has_value()) bp = u16c(bias.value());
   xllm::launch_skinny_gemm(u16(c), u16c(a), u16c(b), bp,

@@ -14,11 +14,12 @@ from . import parallel_state as ps
 
 
 def _maybe_skinny(x, weight, bias):
-    """Decode batches (rows <= 128) route to the split-K weight-streaming
-    kernel when the layer opted in (ops/skinny_gemm.hip); else hipBLASLt."""
+    """Decode batches (rows <= 64, where the kernel measures at or ahead
+    of hipBLASLt) route to the split-K weight-streaming kernel when the
+    layer opted in (ops/skinny_gemm.hip); else hipBLASLt."""
     from xllm_service_amd import ops as xops
-    if (x.is_cuda and 0 < x.shape[0] <= 128 and xops.HAS_EXT
-            and weight.shape[1] % 32 == 0 and weight.shape[0] % 4 == 0):
+    if (x.is_cuda and 0 < x.shape[0] <= 64 and xops.HAS_EXT
+            and weight.shape[1] % 64 == 0 and weight.shape[0] % 4 == 0):
         return xops.skinny_gemm(x, weight, bias)
     return torch.nn.functional.linear(x, weight, bias)
 

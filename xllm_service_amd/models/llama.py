@@ -36,11 +36,12 @@ class LlamaAttention(nn.Module):
             [cfg.q_size, cfg.kv_size, cfg.kv_size],
             dtype=dtype)
         self.o_proj = RowParallelLinear(cfg.q_size, cfg.hidden_size, dtype=dtype)
-        # NOTE: an experimental split-K weight-streaming GEMM
-        # (csrc/ops/skinny_gemm.hip) was measured against hipBLASLt for the
-        # decode qkv/o shapes and LOST (1.1-1.6 TB/s vs the library's ~3.5
-        # TB/s aggregate in-graph) — library GEMMs stay on this path;
-        # use_skinny remains available per-layer for future experiments.
+        # NOTE: the split-K weight-streaming GEMM (csrc/ops/skinny_gemm.hip
+        # v2) beats hipBLASLt warm on the o-projection microbench (23.1us vs
+        # 30.4us at M=64) but REGRESSES the end-to-end decode bench in-graph
+        # (9.75k -> 9.23k tok/s: the partial-reduce second launch and ws
+        # round-trip cost more than the GEMM saves), so library GEMMs keep
+        # the model path; use_skinny stays available per-layer.
 
     def forward(self, x, positions, kv_cache, meta: AttnMetadata, cos_sin):
         qkv = self.qkv_proj(x)

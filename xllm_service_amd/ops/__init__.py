@@ -200,14 +200,10 @@ def mfma_gemm(a, b, bias=None):
 
 
 def skinny_gemm(a, b, bias=None):
-    """Split-K weight-streaming GEMM for decode batches (M <= 128).
+    """Split-K weight-streaming GEMM for decode batches (M <= 64).
     GPU-only; callers fall back to F.linear elsewhere."""
     _require_ext()
-    sk = (a.shape[1] + 1023) // 1024
-    mpad = (a.shape[0] + 15) // 16 * 16
-    ws = torch.empty(sk, mpad, b.shape[0], dtype=torch.float32,
-                     device=a.device)
-    return _ops.skinny_gemm(a, b, bias, ws)
+    return _ops.skinny_gemm(a, b, bias)
 
 
 def mfma_probe_16x16x32(a, b):
