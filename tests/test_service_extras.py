@@ -40,6 +40,40 @@ async def test_logprobs_and_tracer_and_reload(tmp_path):
         # top-N dict may collapse; presence is what matters here
         assert all(1 <= len(t) <= 3 for t in lp["top_logprobs"])
 
+        # streaming completion carries per-chunk logprobs
+        async with client.stream("POST", "/v1/completions", json={
+                "model": "llama-tiny", "prompt": [3, 4, 5], "max_tokens": 4,
+                "temperature": 0.0, "ignore_eos": True, "logprobs": 2,
+                "stream": True}) as resp:
+            assert resp.status_code == 200
+            chunks = []
+            async for line in resp.aiter_lines():
+                if line.startswith("data: ") and line != "data: [DONE]":
+                    chunks.append(json.loads(line[6:]))
+        stream_lps = [v for c in chunks
+                      for v in (c["choices"][0].get("logprobs") or {})
+                      .get("token_logprobs", [])]
+        assert len(stream_lps) == 4
+        assert all(isinstance(v, float) and v <= 0 for v in stream_lps)
+
+        # streaming chat carries OpenAI chat-style logprobs content blocks
+        async with client.stream("POST", "/v1/chat/completions", json={
+                "model": "llama-tiny",
+                "messages": [{"role": "user", "content": "hi"}],
+                "max_tokens": 3, "temperature": 0.0, "ignore_eos": True,
+                "logprobs": True, "top_logprobs": 2,
+                "stream": True}) as resp:
+            assert resp.status_code == 200
+            items = []
+            async for line in resp.aiter_lines():
+                if line.startswith("data: ") and line != "data: [DONE]":
+                    c = json.loads(line[6:])
+                    if c.get("choices"):
+                        items.extend((c["choices"][0].get("logprobs") or {})
+                                     .get("content", []))
+        assert len(items) == 3
+        assert all("logprob" in it and "top_logprobs" in it for it in items)
+
         # tracer wrote request records
         assert os.path.exists(trace_path)
         lines = [json.loads(x) for x in open(trace_path)]

@@ -56,6 +56,8 @@ class ChatBody(BaseModel):
     stream_options: Optional[Dict[str, Any]] = None
     stop: Union[str, List[str], None] = None
     seed: Optional[int] = None
+    logprobs: bool = False          # OpenAI chat: bool + top_logprobs int
+    top_logprobs: Optional[int] = None
     ignore_eos: bool = False
     offline: bool = False
     min_tokens: int = 0
@@ -320,11 +322,13 @@ def build_app(master) -> FastAPI:
         return await _run(req, request, body.stream, chat=True)
 
     def _sampling_dict(body) -> Dict[str, Any]:
+        lp = getattr(body, "logprobs", None)
+        if isinstance(lp, bool):    # chat API: logprobs=true + top_logprobs=N
+            lp = (getattr(body, "top_logprobs", None) or 0) if lp else None
         d = dict(temperature=body.temperature, top_p=body.top_p,
                  top_k=body.top_k, max_tokens=body.max_tokens or 16,
                  min_tokens=body.min_tokens, seed=body.seed,
-                 ignore_eos=body.ignore_eos,
-                 logprobs=getattr(body, "logprobs", None))
+                 ignore_eos=body.ignore_eos, logprobs=lp)
         if body.stream_options:
             d["include_usage"] = bool(
                 body.stream_options.get("include_usage"))

@@ -39,12 +39,24 @@ class ResponseHandler:
         emitted_tool = False
         usage = None
 
-        def chunk(delta: dict, finish: Optional[str] = None):
+        def chunk(delta: dict, finish: Optional[str] = None, lps=None):
+            choice = {"index": 0, "delta": delta, "finish_reason": finish}
+            if lps:
+                choice["logprobs"] = {"content": lps}
             return _sse({
                 "id": rid, "object": "chat.completion.chunk",
                 "created": created, "model": model,
-                "choices": [{"index": 0, "delta": delta,
-                             "finish_reason": finish}]})
+                "choices": [choice]})
+
+        def chat_lps(gen):
+            if not gen.logprobs:
+                return None
+            return [{"token": self.tokenizer.decode([t]),
+                     "logprob": lp["token_logprob"] if lp else None,
+                     "top_logprobs": [
+                         {"token": self.tokenizer.decode([tt]), "logprob": vv}
+                         for tt, vv in (lp.get("top") or {}).items()]}
+                    for t, lp in zip(gen.token_ids, gen.logprobs)]
 
         try:
             while True:
@@ -73,7 +85,11 @@ class ResponseHandler:
                         tool_idx += 1
                         emitted_tool = True
                 if content:
-                    yield chunk({"content": content})
+                    yield chunk({"content": content}, lps=chat_lps(gen))
+                elif gen.logprobs:
+                    # tokens arrived but the incremental decoder is holding
+                    # partial UTF-8: ship the logprobs with an empty delta
+                    yield chunk({}, lps=chat_lps(gen))
                 if gen.finished:
                     if tp is not None:
                         for tc in tp.flush():
@@ -161,11 +177,15 @@ class ResponseHandler:
                                           "type": "server_error"}})
                     break
                 text = dec.push(gen.token_ids) if gen.token_ids else ""
-                if text:
+                if text or gen.logprobs:
+                    choice = {"index": 0, "text": text, "finish_reason": None}
+                    lp = self._completion_logprobs(gen.token_ids,
+                                                   gen.logprobs or [])
+                    if lp is not None:
+                        choice["logprobs"] = lp
                     yield _sse({"id": rid, "object": "text_completion",
                                 "created": created, "model": req.model,
-                                "choices": [{"index": 0, "text": text,
-                                             "finish_reason": None}]})
+                                "choices": [choice]})
                 if gen.finished:
                     yield _sse({"id": rid, "object": "text_completion",
                                 "created": created, "model": req.model,
