@@ -106,3 +106,60 @@ async def test_epd_three_stage_matches_colocated():
                 await w.stop()
             await master.stop()
     assert outputs["epd"] == outputs["colocated"]
+
+
+# ---------------------------------------------------------------- M-RoPE
+def test_mrope_positions_grid():
+    """Hand-checked example of Qwen2-VL 3-D position assignment."""
+    from xllm_service_amd.models.qwen2_vl import mrope_positions
+    ids = [5, 6] + [9] * 6 + [7, 8]          # text, (1,2,3) image, text
+    pos3, delta = mrope_positions(ids, 9, [(1, 2, 3)])
+    assert pos3[:, 0].tolist() == [0, 0, 0]
+    assert pos3[0, 2:8].tolist() == [2] * 6              # temporal
+    assert pos3[1, 2:8].tolist() == [2, 2, 2, 3, 3, 3]   # height
+    assert pos3[2, 2:8].tolist() == [2, 3, 4, 2, 3, 4]   # width
+    assert pos3[:, 8].tolist() == [5, 5, 5]              # resumes at max+1
+    assert delta == 7 - len(ids)
+
+
+def test_mrope_engine_active_and_chunk_invariant():
+    """3-D ids must change the output vs 1-D rope, and chunked prefill must
+    slice per-chunk 3-D ids identically to single-shot prefill."""
+    cfg = get_config(MODEL)
+
+    def run(max_bt, with_grids):
+        eng = LLMEngine(MODEL, device="cpu", max_kv_blocks=128, seed=0,
+                        max_batched_tokens=max_bt)
+        torch.manual_seed(11)
+        mm = torch.randn(8, cfg.hidden_size)
+        ph = cfg.image_pad_token_id
+        prompt = [3, 4] + [ph] * 8 + [5, 6, 7]
+        outs = eng.generate([prompt],
+                            SamplingParams(max_tokens=8, ignore_eos=True),
+                            mm_embeds=[mm],
+                            mm_grids=[[(1, 2, 4)]] if with_grids else None)
+        return outs[0]
+
+    base = run(16384, True)
+    assert run(16384, False) != base          # M-RoPE actually in effect
+    assert run(4, True) == base               # chunked == single-shot
+
+
+def test_mrope_delta_continues_decode():
+    """Decode positions continue at max(pos)+1: a seq with an image must
+    keep mrope_delta < 0 and still generate deterministically."""
+    cfg = get_config(MODEL)
+    eng = LLMEngine(MODEL, device="cpu", max_kv_blocks=128, seed=0)
+    torch.manual_seed(11)
+    mm = torch.randn(8, cfg.hidden_size)
+    ph = cfg.image_pad_token_id
+    prompt = [3, 4] + [ph] * 8 + [5, 6, 7]
+    eng.add_request("m", prompt,
+                    SamplingParams(max_tokens=4, ignore_eos=True),
+                    mm_embeds=mm, mm_grids=[(1, 2, 4)])
+    seq = eng.seqs["m"]
+    assert seq.mrope_pos is not None
+    # image of (1,2,4) occupies 4 slots instead of 8 -> delta = -4
+    assert seq.mrope_delta == -4
+    while not all(o.finished for o in eng.step()):
+        pass

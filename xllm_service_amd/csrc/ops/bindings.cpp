@@ -20,7 +20,7 @@ void launch_fused_rope_cache(unsigned short*, const unsigned short*,
                              const unsigned short*, unsigned short*,
                              unsigned short*, const long*, const long*,
                              const float*, int, int, int, int, int, long,
-                             long, int, hipStream_t);
+                             long, int, int, int, hipStream_t);
 void launch_silu_and_mul(unsigned short*, const unsigned short*, int, int,
                          hipStream_t);
 void launch_gelu_and_mul(unsigned short*, const unsigned short*, int, int,
@@ -118,14 +118,19 @@ void fused_rope_cache(torch::Tensor positions, torch::Tensor q,
                       torch::Tensor k, torch::Tensor v,
                       torch::Tensor k_cache, torch::Tensor v_cache,
                       torch::Tensor slot_mapping, torch::Tensor cos_sin,
-                      long rot_dim) {
+                      long rot_dim, long ms0, long ms1) {
+  // ms0/ms1: cumulative M-RoPE frequency-section bounds; 0 = 1-D rope.
+  // With ms0 > 0, positions is [3, T] (temporal / height / width rows).
   CHECK_BF16_ROWVIEW(q); CHECK_BF16_ROWVIEW(k); CHECK_BF16_ROWVIEW(v);
   CHECK_BF16_CUDA(k_cache); CHECK_BF16_CUDA(v_cache);
-  TORCH_CHECK(positions.dtype() == torch::kLong && positions.is_cuda());
+  TORCH_CHECK(positions.dtype() == torch::kLong && positions.is_cuda() &&
+              positions.is_contiguous());
+  TORCH_CHECK(ms0 == 0 ? positions.dim() == 1
+                       : (positions.dim() == 2 && positions.size(0) == 3));
   TORCH_CHECK(slot_mapping.dtype() == torch::kLong && slot_mapping.is_cuda());
   TORCH_CHECK(cos_sin.dtype() == torch::kFloat && cos_sin.is_cuda());
   TORCH_CHECK(k.stride(0) == v.stride(0));
-  const int T = positions.size(0);
+  const int T = positions.size(positions.dim() - 1);
   const int n_kv = k_cache.size(1);
   const int bs = k_cache.size(2);
   const int D = k_cache.size(3);
@@ -134,7 +139,7 @@ void fused_rope_cache(torch::Tensor positions, torch::Tensor q,
       u16(q), u16c(k), u16c(v), u16(k_cache), u16(v_cache),
       positions.data_ptr<long>(), slot_mapping.data_ptr<long>(),
       cos_sin.data_ptr<float>(), T, n_q, n_kv, D, (int)rot_dim, q.stride(0),
-      k.stride(0), bs, cur_stream());
+      k.stride(0), bs, (int)ms0, (int)ms1, cur_stream());
 }
 
 void silu_and_mul(torch::Tensor out, torch::Tensor x) {

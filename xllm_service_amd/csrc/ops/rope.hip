@@ -70,12 +70,16 @@ __global__ void fused_rope_cache_kernel(
     const unsigned short* __restrict__ v,      // [T, n_kv, D] strided rows
     unsigned short* __restrict__ k_cache,      // [blocks, n_kv, bs, D]
     unsigned short* __restrict__ v_cache,
-    const long* __restrict__ positions,        // [T]
+    const long* __restrict__ positions,  // [T], or [3, T] when ms0 > 0
     const long* __restrict__ slot_mapping,     // [T] (-1 = skip cache write)
     const float* __restrict__ cos_sin,         // [max_pos, rot]
     const int n_q, const int n_kv, const int D, const int rot,
     const long q_stride, const long kv_stride, const int block_size,
-    const int T) {
+    const int T,
+    // M-RoPE (Qwen2-VL): cumulative frequency-section bounds; freq i takes
+    // its position from row 0 (i < ms0: temporal), 1 (i < ms1: height) or
+    // 2 (width) of the [3, T] positions. ms0 == 0 disables (1-D rope).
+    const int ms0, const int ms1) {
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int wpb = blockDim.x >> 6;
@@ -89,10 +93,16 @@ __global__ void fused_rope_cache_kernel(
     const int h = (int)(wk % total_heads);
     const long slot = slot_mapping[token];
     const float* cs = cos_sin + positions[token] * rot;
+    auto cs_at = [&](int i) -> const float* {
+      if (ms0 == 0) return cs;
+      const int sec = i < ms0 ? 0 : (i < ms1 ? 1 : 2);
+      return cos_sin + positions[(long)sec * T + token] * rot;
+    };
     if (h < n_q) {                       // rope q in place
       unsigned short* base = q + token * q_stride + (long)h * D;
       for (int i = lane; i < half; i += 64) {
-        const float c = cs[i], s = cs[half + i];
+        const float* cf = cs_at(i);
+        const float c = cf[i], s = cf[half + i];
         const float x1 = bf16_to_f32(base[i]);
         const float x2 = bf16_to_f32(base[i + half]);
         base[i] = f32_to_bf16(x1 * c - x2 * s);
@@ -106,7 +116,8 @@ __global__ void fused_rope_cache_kernel(
           k_cache + (((slot / block_size) * n_kv + kh) * block_size +
                      slot % block_size) * (long)D;
       for (int i = lane; i < half; i += 64) {
-        const float c = cs[i], s = cs[half + i];
+        const float* cf = cs_at(i);
+        const float c = cf[i], s = cf[half + i];
         const float x1 = bf16_to_f32(src[i]);
         const float x2 = bf16_to_f32(src[i + half]);
         dst[i] = f32_to_bf16(x1 * c - x2 * s);
@@ -133,7 +144,7 @@ void launch_fused_rope_cache(unsigned short* q, const unsigned short* k,
                              const long* slot_mapping, const float* cos_sin,
                              int T, int n_q, int n_kv, int D, int rot,
                              long q_stride, long kv_stride, int block_size,
-                             hipStream_t stream) {
+                             int ms0, int ms1, hipStream_t stream) {
   const long waves = (long)T * (n_q + 2 * n_kv);
   long g = (waves + 3) / 4;
   if (g > 2048) g = 2048;
@@ -141,7 +152,7 @@ void launch_fused_rope_cache(unsigned short* q, const unsigned short* k,
   hipLaunchKernelGGL(fused_rope_cache_kernel, dim3((unsigned)g), dim3(256), 0,
                      stream, q, k, v, k_cache, v_cache, positions,
                      slot_mapping, cos_sin, n_q, n_kv, D, rot, q_stride,
-                     kv_stride, block_size, T);
+                     kv_stride, block_size, T, ms0, ms1);
 }
 
 }  // namespace xllm

@@ -156,7 +156,7 @@ class LLMEngine:
                     priority: int = 0,
                     eos_token_id: Optional[int] = None,
                     hold_blocks: bool = False,
-                    mm_embeds=None) -> None:
+                    mm_embeds=None, mm_grids=None) -> None:
         if request_id in self.seqs:
             raise ValueError(f"duplicate request_id {request_id}")
         seq = Sequence(request_id=request_id,
@@ -170,6 +170,11 @@ class LLMEngine:
             seq.mm_embeds = mm_embeds
             seq.mm_placeholder = getattr(self.model, "image_pad_token_id",
                                          None)
+            # Qwen2-VL M-RoPE: 3-D position ids over the image grid spans
+            if mm_grids and getattr(self.cfg, "mrope_section", ()):
+                from xllm_service_amd.models.qwen2_vl import mrope_positions
+                seq.mrope_pos, seq.mrope_delta = mrope_positions(
+                    seq.prompt_token_ids, seq.mm_placeholder, mm_grids)
         self.seqs[request_id] = seq
         self.scheduler.add(seq)
 
@@ -228,22 +233,29 @@ class LLMEngine:
         tmp.block_table = list(block_ids)
         self.block_manager.free(tmp)
 
+    def held_mrope_delta(self, request_id: str) -> int:
+        seq = self.held.get(request_id) or self.seqs.get(request_id)
+        return seq.mrope_delta if seq is not None else 0
+
     def activate_migrated_request(self, request_id: str,
                                   prompt_token_ids: List[int],
                                   first_token_ids: List[int],
                                   block_ids: List[int],
                                   params: Optional[SamplingParams] = None,
                                   eos_token_id: Optional[int] = None,
-                                  priority: int = 0) -> None:
+                                  priority: int = 0,
+                                  mrope_delta: int = 0) -> None:
         """Resume a request whose prompt KV was migrated into block_ids
         (decode side). No recompute: decode continues from the first
-        prefill-produced token."""
+        prefill-produced token. mrope_delta carries the prefill side's
+        M-RoPE text-position offset (Qwen2-VL)."""
         seq = Sequence(request_id=request_id,
                        prompt_token_ids=list(prompt_token_ids),
                        params=params or SamplingParams(),
                        eos_token_id=eos_token_id if eos_token_id is not None
                        else self.eos_token_id,
                        priority=priority)
+        seq.mrope_delta = mrope_delta
         seq.block_table = list(block_ids)
         seq.num_computed_tokens = seq.prompt_len
         seq.output_token_ids = list(first_token_ids)
@@ -345,9 +357,12 @@ class LLMEngine:
     # ---- convenience (tests, smoke) ----------------------------------------
     def generate(self, prompts: List[List[int]],
                  params: Optional[SamplingParams] = None,
-                 timeout_s: float = 600.0) -> List[List[int]]:
+                 timeout_s: float = 600.0,
+                 mm_embeds=None, mm_grids=None) -> List[List[int]]:
         for i, p in enumerate(prompts):
-            self.add_request(f"gen-{i}", p, params)
+            self.add_request(f"gen-{i}", p, params,
+                             mm_embeds=mm_embeds[i] if mm_embeds else None,
+                             mm_grids=mm_grids[i] if mm_grids else None)
         results: Dict[str, List[int]] = {}
         t0 = time.monotonic()
         while self.has_work():

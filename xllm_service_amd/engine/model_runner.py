@@ -90,6 +90,8 @@ class ModelRunner:
         tokens: List[int] = []
         positions: List[int] = []
         slots: List[int] = []
+        # M-RoPE: (row_range, seq) pairs whose prefill chunks need 3-D ids
+        mrope_spans = []
 
         cu_q = [0]
         p_seq_lens: List[int] = []
@@ -98,6 +100,8 @@ class ModelRunner:
             seq = sp.seq
             chunk_toks = seq.prompt_token_ids[sp.chunk_start:
                                               sp.chunk_start + sp.chunk_len]
+            if seq.mrope_pos is not None:
+                mrope_spans.append((len(tokens), sp))
             tokens.extend(chunk_toks)
             positions.extend(range(sp.chunk_start, sp.chunk_start + sp.chunk_len))
             for pos in range(sp.chunk_start, sp.chunk_start + sp.chunk_len):
@@ -112,12 +116,25 @@ class ModelRunner:
         for seq in plan.decodes:
             last_tok = (seq.output_token_ids[-1] if seq.output_token_ids
                         else seq.prompt_token_ids[-1])
-            pos = seq.total_len - 1
+            # decode rope positions shift by the M-RoPE delta (text position
+            # after images compresses below total_len); attn metadata keeps
+            # the un-shifted lengths
+            pos = seq.total_len - 1 + seq.mrope_delta
             tokens.append(last_tok)
             positions.append(pos)
             slots.append(bm.append_slot(seq))
             d_seq_lens.append(seq.total_len)
             d_tables.append(seq.block_table)
+
+        import numpy as _np
+        if mrope_spans:
+            pos_np = _np.tile(_np.asarray(positions, dtype=_np.int64), (3, 1))
+            for row0, sp in mrope_spans:
+                pos_np[:, row0:row0 + sp.chunk_len] = \
+                    sp.seq.mrope_pos[:, sp.chunk_start:
+                                     sp.chunk_start + sp.chunk_len]
+        else:
+            pos_np = _np.asarray(positions, dtype=_np.int64)
 
         def pad_tables(tables: List[List[int]]) -> torch.Tensor:
             if not tables:
@@ -132,7 +149,6 @@ class ModelRunner:
         np_ = cu_q[-1]
         from xllm_service_amd.distributed import parallel_state as ps
         if ps.tp_size() > 1 and ps.tp_rank() == 0:
-            import numpy as _np
             def tables_np(tables):
                 if not tables:
                     return None
@@ -144,7 +160,7 @@ class ModelRunner:
             self._tp_batch = dict(
                 kind="eager", np=np_, nd=len(plan.decodes),
                 input_ids=_np.asarray(tokens, dtype=_np.int64),
-                positions=_np.asarray(positions, dtype=_np.int64),
+                positions=pos_np,
                 slots=_np.asarray(slots, dtype=_np.int64),
                 cu_q=_np.asarray(cu_q, dtype=_np.int32)
                 if plan.prefills else None,
@@ -169,7 +185,7 @@ class ModelRunner:
             decode_block_tables=pad_tables(d_tables) if plan.decodes else None,
         )
         input_ids = torch.tensor(tokens, dtype=torch.long, device=dev)
-        pos_t = torch.tensor(positions, dtype=torch.long, device=dev)
+        pos_t = torch.from_numpy(pos_np).to(dev)
         return input_ids, pos_t, meta
 
     # ---- sampling -----------------------------------------------------------
@@ -370,7 +386,7 @@ class ModelRunner:
         for i, seq in enumerate(seqs):
             input_ids[i] = (seq.output_token_ids[-1] if seq.output_token_ids
                             else seq.prompt_token_ids[-1])
-            positions[i] = seq.total_len - 1
+            positions[i] = seq.total_len - 1 + seq.mrope_delta
             slots[i] = bm.append_slot(seq)
             seq_lens[i] = seq.total_len
             bt_rows.append(gr.block_row(seq))

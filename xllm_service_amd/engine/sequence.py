@@ -43,6 +43,10 @@ class Sequence:        # token lists element-wise (O(len) per membership test)
     # token positions during prefill (EPD E->P handoff)
     mm_embeds: Optional["object"] = None      # torch.Tensor [n, hidden]
     mm_placeholder: Optional[int] = None
+    # Qwen2-VL M-RoPE: [3, prompt_len] numpy position ids (None = 1-D rope)
+    # and the text-position offset applied to every decode step
+    mrope_pos: Optional["object"] = None
+    mrope_delta: int = 0
     # PD-disaggregation: set on a decode instance receiving a migrated prefill
     migrated_in: bool = False
     # PD-disaggregation: keep KV blocks alive after finish (prefill side

@@ -40,6 +40,14 @@ from .engine import LLMEngine
 from .sampling import SamplingParams
 
 
+def _image_grids(cfg, images):
+    """LM-token grid (t, h, w) per image after the 2x2 spatial merge —
+    the spans M-RoPE assigns 3-D position ids over."""
+    m = int(cfg.vision.get("spatial_merge_size", 2)) if cfg.vision else 2
+    return [[1, int(img["grid_h"]) // m, int(img["grid_w"]) // m]
+            for img in images]
+
+
 class VisionEncoder:
     """Stage-E vision tower runner (ENCODE instances; also used in-process
     by colocated DEFAULT instances serving multimodal models)."""
@@ -353,6 +361,7 @@ class Worker:
             params=params, routing=routing, prompt_len=len(token_ids),
             offline=offline, multimodal=multimodal)
         mm_embeds = None
+        mm_grids = None
         if multimodal:
             if multimodal.get("embeds_b") is not None:
                 import numpy as np
@@ -360,6 +369,7 @@ class Worker:
                 mm_embeds = torch.from_numpy(raw.copy()).view(
                     torch.bfloat16 if multimodal.get("dtype") == "bfloat16"
                     else torch.float32).reshape(multimodal["embeds_shape"])
+                mm_grids = multimodal.get("grids")
             elif multimodal.get("images"):
                 # colocated multimodal: run the vision tower in-process
                 if self.encoder is None:
@@ -367,6 +377,7 @@ class Worker:
                         None, lambda: VisionEncoder(self.model, self.device))
                 mm_embeds = await self._loop.run_in_executor(
                     None, lambda: self.encoder.encode(multimodal["images"]))
+                mm_grids = _image_grids(self.encoder.cfg, multimodal["images"])
         decode_name = routing.get("decode_name")
         do_migrate = (self.itype == InstanceType.PREFILL
                       and decode_name and decode_name != self.name)
@@ -382,12 +393,13 @@ class Worker:
                 lambda: self.engine.add_request(
                     service_request_id, token_ids, first_sp,
                     priority=1 if offline else 0, hold_blocks=True,
-                    mm_embeds=mm_embeds))
+                    mm_embeds=mm_embeds, mm_grids=mm_grids))
         else:
             self._post_to_engine(
                 lambda: self.engine.add_request(
                     service_request_id, token_ids, sp,
-                    priority=1 if offline else 0, mm_embeds=mm_embeds))
+                    priority=1 if offline else 0, mm_embeds=mm_embeds,
+                    mm_grids=mm_grids))
 
     async def _encode_and_forward(self, rid, token_ids, params, routing,
                                   offline, multimodal):
@@ -404,7 +416,9 @@ class Worker:
                 if embeds.dtype != torch.float32
                 else embeds.cpu().numpy().tobytes(),
                 dtype=str(embeds.dtype).split(".")[-1],
-                embeds_shape=list(embeds.shape))
+                embeds_shape=list(embeds.shape),
+                grids=_image_grids(self.encoder.cfg,
+                                   (multimodal or {}).get("images", [])))
             await conn.notify("execute_request",
                               service_request_id=rid, token_ids=token_ids,
                               params=params, routing=routing,
@@ -433,7 +447,8 @@ class Worker:
                              data: Optional[bytes] = None,
                              src_name: Optional[str] = None,
                              src_blocks: Optional[List[int]] = None,
-                             offline: bool = False) -> bool:
+                             offline: bool = False,
+                             mrope_delta: int = 0) -> bool:
         sp = params_from_dict(params)
         # account the token(s) the prefill already produced
         sp.max_tokens = max(sp.max_tokens, 1)
@@ -467,7 +482,8 @@ class Worker:
                 raise
             self.engine.activate_migrated_request(
                 service_request_id, prompt_token_ids, first_token_ids,
-                blocks, sp, priority=1 if offline else 0)
+                blocks, sp, priority=1 if offline else 0,
+                mrope_delta=mrope_delta)
 
         await self._run_on_engine(_recv)
         return True
@@ -538,7 +554,9 @@ class Worker:
                 first_token_ids=first_toks,
                 params=mig["params"], n_blocks=len(blocks),
                 transport=transport, src_name=self.name,
-                offline=mig.get("offline", False))
+                offline=mig.get("offline", False),
+                mrope_delta=await self._run_on_engine(
+                    lambda: self.engine.held_mrope_delta(rid)))
             if transport == "bytes":
                 kwargs["data"] = await self._run_on_engine(
                     lambda: self.engine.export_block_bytes(blocks))

@@ -23,6 +23,10 @@ class ModelConfig:
     # vision tower (qwen2_vl only)
     vision: dict = field(default_factory=dict)
     image_pad_token_id: int = -1
+    # M-RoPE frequency split (temporal, height, width) in frequency units,
+    # summing to rot_dim // 2. Empty = plain 1-D RoPE.
+    # (reference: Qwen2-VL 3-D rotary position embedding)
+    mrope_section: tuple = ()
 
     @property
     def q_size(self) -> int:
@@ -59,6 +63,7 @@ PRESETS = {
         hidden_size=3584, intermediate_size=18944, num_layers=28,
         num_heads=28, num_kv_heads=4, head_dim=128, rope_theta=1000000.0,
         max_position=8192, image_pad_token_id=151655,
+        mrope_section=(16, 24, 24),
         vision=dict(depth=32, embed_dim=1280, num_heads=16, patch_size=14,
                     spatial_merge_size=2, out_hidden_size=3584)),
     # small multimodal config with head_dim 128 for GPU tests
@@ -66,7 +71,7 @@ PRESETS = {
         name="qwen2-vl-debug", architecture="qwen2_vl", vocab_size=1024,
         hidden_size=512, intermediate_size=1024, num_layers=2, num_heads=4,
         num_kv_heads=2, head_dim=128, rope_theta=10000.0, max_position=2048,
-        image_pad_token_id=9,
+        image_pad_token_id=9, mrope_section=(16, 24, 24),
         vision=dict(depth=2, embed_dim=256, num_heads=4, patch_size=14,
                     spatial_merge_size=2, out_hidden_size=512)),
     # tiny multimodal config for CPU EPD tests
@@ -74,7 +79,7 @@ PRESETS = {
         name="qwen2-vl-tiny", architecture="qwen2_vl", vocab_size=1024,
         hidden_size=256, intermediate_size=512, num_layers=2, num_heads=4,
         num_kv_heads=2, head_dim=64, rope_theta=10000.0, max_position=2048,
-        image_pad_token_id=9,
+        image_pad_token_id=9, mrope_section=(8, 12, 12),
         vision=dict(depth=2, embed_dim=64, num_heads=4, patch_size=14,
                     spatial_merge_size=2, out_hidden_size=256)),
 }

@@ -31,6 +31,7 @@ class LlamaAttention(nn.Module):
         self.n_kv_heads = max(cfg.num_kv_heads // tp, 1)
         self.head_dim = cfg.head_dim
         self.scale = 1.0 / math.sqrt(cfg.head_dim)
+        self.mrope_sections = tuple(getattr(cfg, "mrope_section", ()) or ())
         self.qkv_proj = MergedColumnParallelLinear(
             cfg.hidden_size,
             [cfg.q_size, cfg.kv_size, cfg.kv_size],
@@ -57,7 +58,8 @@ class LlamaAttention(nn.Module):
             -1, (self.n_kv_heads, self.head_dim))
         k_cache, v_cache = kv_cache
         qh = ops.fused_rope_cache(positions, qh, kh, vh, k_cache, v_cache,
-                                  meta.slot_mapping, cos_sin, self.head_dim)
+                                  meta.slot_mapping, cos_sin, self.head_dim,
+                                  mrope_sections=self.mrope_sections)
 
         np_, nd = meta.num_prefill_tokens, meta.num_decode_tokens
         out = torch.empty(T, q_sz, dtype=qkv.dtype, device=qkv.device)

@@ -163,6 +163,44 @@ class Qwen2VisionTransformer(nn.Module):
                              ).to(p.dtype))
 
 
+def mrope_positions(token_ids, placeholder_id, grids):
+    """3-D (temporal, height, width) M-RoPE position ids for one prompt.
+
+    Text tokens advance all three components together; each image span of
+    LM-grid (t, h, w) gets grid-coordinate positions offset by the running
+    maximum, and text after it resumes at max + 1 — so a (1, 4, 6) image
+    occupies only max(1,4,6)=6 position slots instead of 24.
+    (reference semantics: Qwen2-VL M-RoPE / get_rope_index)
+
+    Returns (pos3 [3, L] int64 numpy, delta) where delta = (max_pos + 1) - L
+    is added to text-style positions for all decode steps.
+    """
+    import numpy as np
+    L = len(token_ids)
+    pos3 = np.empty((3, L), dtype=np.int64)
+    gi = 0
+    cur = 0           # next text position
+    i = 0
+    while i < L:
+        if token_ids[i] == placeholder_id and gi < len(grids):
+            t, h, w = (int(x) for x in grids[gi])
+            n = t * h * w
+            st = cur
+            idx = np.arange(n)
+            pos3[0, i:i + n] = st + idx // (h * w)
+            pos3[1, i:i + n] = st + (idx // w) % h
+            pos3[2, i:i + n] = st + idx % w
+            cur = st + max(t, h, w)
+            gi += 1
+            i += n
+        else:
+            pos3[:, i] = cur
+            cur += 1
+            i += 1
+    delta = int(cur - L)
+    return pos3, delta
+
+
 class Qwen2VLForCausalLM(LlamaForCausalLM):
     """Language tower; the vision tower runs in the ENCODE stage (or
     in-process for colocated DEFAULT instances)."""

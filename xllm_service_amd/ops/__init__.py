@@ -57,16 +57,24 @@ def rope(positions, q, k, cos_sin, head_dim, rot_dim):
 
 
 def fused_rope_cache(positions, q, k, v, k_cache, v_cache, slot_mapping,
-                     cos_sin, rot_dim):
+                     cos_sin, rot_dim, mrope_sections=None):
     """GPU: one kernel ropes q in place and scatters roped-k + v into the
-    paged cache. CPU: composed from the reference ops."""
+    paged cache. CPU: composed from the reference ops. With mrope_sections
+    (Qwen2-VL M-RoPE) positions is [3, T]."""
+    if positions.dim() == 1:
+        mrope_sections = None
     if q.is_cuda:
         _require_ext()
+        ms0 = ms1 = 0
+        if mrope_sections:
+            ms0 = int(mrope_sections[0])
+            ms1 = ms0 + int(mrope_sections[1])
         _ops.fused_rope_cache(positions, q, k, v, k_cache, v_cache,
-                              slot_mapping, cos_sin, rot_dim)
+                              slot_mapping, cos_sin, rot_dim, ms0, ms1)
         return q
     D = k_cache.shape[3]
-    q2, k2 = ref.rope(positions, q, k, cos_sin, D, rot_dim)
+    q2, k2 = ref.rope(positions, q, k, cos_sin, D, rot_dim,
+                      mrope_sections=mrope_sections)
     ref.reshape_and_cache(k2, v, k_cache, v_cache, slot_mapping)
     return q2
 

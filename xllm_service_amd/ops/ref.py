@@ -34,9 +34,33 @@ def rope_table(rot_dim: int, max_pos: int, base: float = 10000.0,
     return torch.cat([freqs.cos(), freqs.sin()], dim=-1)
 
 
-def rope(positions, q, k, cos_sin, head_dim, rot_dim):
-    """NeoX rotate-half RoPE, applied out-of-place (reference)."""
+def rope(positions, q, k, cos_sin, head_dim, rot_dim, mrope_sections=None):
+    """NeoX rotate-half RoPE, applied out-of-place (reference). With
+    mrope_sections (t, h, w summing to rot_dim//2) positions is [3, T] and
+    frequency i takes its position from the section containing i."""
     def _apply(x):
+        if mrope_sections:
+            pos3 = positions
+            T = pos3.shape[1]
+            half = rot_dim // 2
+            sec = torch.empty(half, dtype=torch.long)
+            s0, s1, _ = mrope_sections
+            sec[:s0] = 0
+            sec[s0:s0 + s1] = 1
+            sec[s0 + s1:] = 2
+            # per (token, freq) position -> gather cos/sin rows
+            pf = pos3[sec, :].T                      # [T, half]
+            cos_t = cos_sin[:, :half][pf, torch.arange(half)]   # [T, half]
+            sin_t = cos_sin[:, half:rot_dim][pf, torch.arange(half)]
+            xs = x.reshape(T, -1, head_dim).float()
+            cos = cos_t.unsqueeze(1)
+            sin = sin_t.unsqueeze(1)
+            x1 = xs[..., :half]
+            x2 = xs[..., half:rot_dim]
+            o1 = x1 * cos - x2 * sin
+            o2 = x2 * cos + x1 * sin
+            out = torch.cat([o1, o2, xs[..., rot_dim:]], dim=-1)
+            return out.to(x.dtype).reshape(x.shape)
         T = positions.shape[0]
         xs = x.reshape(T, -1, head_dim).float()
         cs = cos_sin[positions]  # [T, rot_dim]
