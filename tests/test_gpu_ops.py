@@ -105,6 +105,35 @@ def test_fused_rope_cache(dev):
     _assert_close(vc, vc_ref, atol=3e-2, label="fused v_cache")
 
 
+def test_fused_rope_cache_mrope(dev):
+    """Sectioned M-RoPE (Qwen2-VL) path of the fused kernel vs the CPU
+    reference with [3, T] positions."""
+    torch.manual_seed(5)
+    T, Hq, Hkv, D, bs, blocks = 9, 8, 2, 128, 16, 8
+    sections = (16, 24, 24)
+    qkv = torch.randn(T, (Hq + 2 * Hkv) * D, device=dev, dtype=torch.bfloat16)
+    q = qkv[:, :Hq * D].unflatten(-1, (Hq, D))
+    k = qkv[:, Hq * D:(Hq + Hkv) * D].unflatten(-1, (Hkv, D))
+    v = qkv[:, (Hq + Hkv) * D:].unflatten(-1, (Hkv, D))
+    kc = torch.zeros(blocks, Hkv, bs, D, device=dev, dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    pos3 = torch.randint(0, 2048, (3, T), device=dev)
+    slots = torch.randperm(blocks * bs, device=dev)[:T]
+    table = ref.rope_table(D, 4096).to(dev)
+    q_ref, k_ref = ref.rope(pos3.cpu(), q.float().cpu().reshape(T, -1),
+                            k.float().cpu().reshape(T, -1), table.cpu(), D, D,
+                            mrope_sections=sections)
+    kc_ref = torch.zeros(blocks, Hkv, bs, D)
+    vc_ref = torch.zeros_like(kc_ref)
+    ref.reshape_and_cache(k_ref.reshape(T, Hkv, D), v.float().cpu(),
+                          kc_ref, vc_ref, slots.cpu())
+    ops.fused_rope_cache(pos3, q, k, v, kc, vc, slots, table, D,
+                         mrope_sections=sections)
+    _assert_close(q.reshape(T, -1), q_ref, atol=3e-2, label="mrope q")
+    _assert_close(kc, kc_ref, atol=3e-2, label="mrope k_cache")
+    _assert_close(vc, vc_ref, atol=3e-2, label="mrope v_cache")
+
+
 def test_silu_and_mul(dev):
     x = torch.randn(37, 2 * 14336, device=dev, dtype=torch.bfloat16)
     got = ops.silu_and_mul(x)
