@@ -186,3 +186,25 @@ def test_swap_preemption_preserves_kv():
     assert eng.block_manager.num_free == eng.block_manager.num_blocks
     assert (eng.cpu_block_manager.num_free ==
             eng.cpu_block_manager.num_blocks)
+
+
+def test_ssd_swap_tier_preserves_kv(tmp_path):
+    """With no DRAM tier (swap_space_mb=0), preemption spools KV blocks to
+    SSD files and restores them exactly; spool files are cleaned up."""
+    import os
+    ssd = str(tmp_path / "kvspool")
+    eng = LLMEngine("llama-tiny", device="cpu", max_kv_blocks=9, seed=7,
+                    enable_prefix_caching=False, swap_space_mb=0,
+                    ssd_swap_dir=ssd)
+    assert eng.cpu_block_manager is None
+    cfg = get_config("llama-tiny")
+    torch.manual_seed(17)
+    prompts = [torch.randint(0, cfg.vocab_size, (60,)).tolist()
+               for _ in range(2)]
+    got = eng.generate(prompts, SamplingParams(max_tokens=8, ignore_eos=True))
+    assert eng.scheduler.num_swap_outs > 0, "no ssd swap happened"
+    assert eng.scheduler.num_swap_ins > 0
+    for p, g in zip(prompts, got):
+        assert g == dense_greedy(eng.model, cfg, p, 8)
+    assert eng.block_manager.num_free == eng.block_manager.num_blocks
+    assert os.listdir(ssd) == []          # spools deleted after swap-in

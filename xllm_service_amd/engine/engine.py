@@ -42,6 +42,20 @@ class EngineStats:
     generated_tokens: int = 0
 
 
+class SsdSpool:
+    """SSD-tier swap handle: a sequence's KV blocks serialized to a spool
+    file. Stored in seq.cpu_block_table (len() = block count, which is all
+    the scheduler's resume accounting needs)."""
+    __slots__ = ("path", "n")
+
+    def __init__(self, path: str, n: int):
+        self.path = path
+        self.n = n
+
+    def __len__(self) -> int:
+        return self.n
+
+
 class LLMEngine:
     def __init__(self, model_name: str = "llama-3-8b",
                  device: Optional[str] = None,
@@ -54,6 +68,7 @@ class LLMEngine:
                  enable_graphs: bool = True,
                  max_model_len: int = 4096,
                  swap_space_mb: int = 1024,
+                 ssd_swap_dir: Optional[str] = None,
                  tp_size: int = 1,
                  load_state_path: Optional[str] = None,
                  seed: int = 0):
@@ -109,12 +124,19 @@ class LLMEngine:
             BlockManager(num_cpu_blocks, BLOCK_SIZE,
                          enable_prefix_caching=False)
             if num_cpu_blocks > 0 else None)
+        # SSD tier below DRAM: spool files holding serialized KV blocks
+        # (reference: the hbm/dram/ssd cache hierarchy in the KV index)
+        self.ssd_swap_dir = ssd_swap_dir
+        if ssd_swap_dir:
+            import os
+            os.makedirs(ssd_swap_dir, exist_ok=True)
+        has_swap = bool(self.cpu_block_manager or ssd_swap_dir)
         self.scheduler = EngineScheduler(
             self.block_manager, max_num_seqs=max_num_seqs,
             max_batched_tokens=max_batched_tokens,
-            swap_out=self._swap_out if self.cpu_block_manager else None,
-            swap_in=self._swap_in if self.cpu_block_manager else None)
-        if self.cpu_block_manager:
+            swap_out=self._swap_out if has_swap else None,
+            swap_in=self._swap_in if has_swap else None)
+        if has_swap:
             self.scheduler.free_cpu_blocks = self._free_cpu_blocks
         self.runner = ModelRunner(self.model, self.cfg, self.device,
                                   num_blocks, dtype=dtype,
@@ -178,13 +200,13 @@ class LLMEngine:
         self.seqs[request_id] = seq
         self.scheduler.add(seq)
 
-    # ---- host-DRAM KV tier (swap) ------------------------------------------
+    # ---- host-DRAM + SSD KV tiers (swap) -----------------------------------
     def _swap_out(self, seq):
         from xllm_service_amd import ops as xops
         cbm = self.cpu_block_manager
         n = len(seq.block_table)
-        if cbm.num_free < n:
-            return None  # dram tier full: caller falls back to recompute
+        if cbm is None or cbm.num_free < n:
+            return self._ssd_swap_out(seq)  # dram full -> ssd tier
         cpu_blocks = cbm.allocate_raw(n)
         for (kc, vc), (ck, cv) in zip(self.runner.kv_caches,
                                       self.runner.cpu_kv_caches):
@@ -194,8 +216,29 @@ class LLMEngine:
             torch.cuda.synchronize(self.device)
         return cpu_blocks
 
+    def _ssd_swap_out(self, seq):
+        """Spool the sequence's KV blocks to a file (tier below DRAM);
+        returns an SsdSpool handle or None (-> recompute)."""
+        if not self.ssd_swap_dir:
+            return None
+        import os
+        path = os.path.join(self.ssd_swap_dir,
+                            f"{seq.request_id}.{seq.preempt_count}.kv")
+        data = self.export_block_bytes(seq.block_table)
+        with open(path, "wb") as f:
+            f.write(data)
+        return SsdSpool(path, len(seq.block_table))
+
     def _swap_in(self, seq):
         from xllm_service_amd import ops as xops
+        if isinstance(seq.cpu_block_table, SsdSpool):
+            spool = seq.cpu_block_table
+            gpu_blocks = self.block_manager.allocate_raw(len(spool))
+            with open(spool.path, "rb") as f:
+                self.import_block_bytes(gpu_blocks, f.read())
+            self._free_cpu_blocks(seq)
+            seq.block_table = gpu_blocks
+            return
         n = len(seq.cpu_block_table)
         gpu_blocks = self.block_manager.allocate_raw(n)
         for (kc, vc), (ck, cv) in zip(self.runner.kv_caches,
@@ -208,9 +251,18 @@ class LLMEngine:
         seq.block_table = gpu_blocks
 
     def _free_cpu_blocks(self, seq):
-        tmp = Sequence("_cpu_tmp", [], SamplingParams())
-        tmp.block_table = list(seq.cpu_block_table)
-        self.cpu_block_manager.free(tmp)
+        if isinstance(seq.cpu_block_table, SsdSpool):
+            import os
+            try:
+                os.unlink(seq.cpu_block_table.path)
+            except OSError:
+                pass
+            seq.cpu_block_table = []
+            return
+        if seq.cpu_block_table:
+            tmp = Sequence("_cpu_tmp", [], SamplingParams())
+            tmp.block_table = list(seq.cpu_block_table)
+            self.cpu_block_manager.free(tmp)
         seq.cpu_block_table = []
 
     # ---- PD-disaggregation support -----------------------------------------

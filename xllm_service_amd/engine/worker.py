@@ -658,6 +658,8 @@ def main():
     ap.add_argument("--max-batched-tokens", type=int, default=8192)
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--no-graphs", action="store_true")
+    ap.add_argument("--ssd-swap-dir", default=None,
+                    help="directory for the SSD KV swap tier (below DRAM)")
     ap.add_argument("--tp", type=int, default=1,
                     help="tensor-parallel group size; launch one process "
                          "per rank with RANK/WORLD_SIZE/MASTER_ADDR set "
@@ -690,6 +692,7 @@ def main():
         engine_kwargs=dict(seed=args.seed, max_num_seqs=args.max_num_seqs,
                            max_batched_tokens=args.max_batched_tokens,
                            enable_graphs=not args.no_graphs,
+                           ssd_swap_dir=args.ssd_swap_dir,
                            tp_size=args.tp,
                            load_state_path=args.load_state))
 
