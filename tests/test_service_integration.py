@@ -124,7 +124,7 @@ async def test_pd_disaggregation_matches_colocated():
     (deterministic greedy, identical random-init weights)."""
     prompt = list(range(40, 76))  # 36 tokens
     outputs = {}
-    for mode in ("colocated", "pd"):
+    for mode in ("colocated", "pd", "pd_relay"):
         master = make_master(policy="RR")
         await master.start(serve_http=False)
         workers = []
@@ -132,8 +132,12 @@ async def test_pd_disaggregation_matches_colocated():
             if mode == "colocated":
                 workers = [Worker("w0", "DEFAULT", **worker_kwargs(master))]
             else:
+                # pd_relay: decode routes generations decode->prefill->master
+                # (the reference's second response topology)
+                relay = mode == "pd_relay"
                 workers = [Worker("p0", "PREFILL", **worker_kwargs(master)),
-                           Worker("d0", "DECODE", **worker_kwargs(master))]
+                           Worker("d0", "DECODE", relay_responses=relay,
+                                  **worker_kwargs(master))]
             for w in workers:
                 await w.start()
             await wait_for(
@@ -151,6 +155,7 @@ async def test_pd_disaggregation_matches_colocated():
                 await w.stop()
             await master.stop()
     assert outputs["pd"] == outputs["colocated"]
+    assert outputs["pd_relay"] == outputs["colocated"]
 
 
 @pytest.mark.anyio

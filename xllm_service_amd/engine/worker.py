@@ -105,6 +105,7 @@ class Worker:
                  heartbeat_s: float = 1.0, lease_ttl_s: float = 3.0,
                  eos_token_id: Optional[int] = None,
                  max_kv_blocks: Optional[int] = None,
+                 relay_responses: bool = False,
                  engine_kwargs: Optional[Dict[str, Any]] = None):
         self.name = name
         self.itype = InstanceType(itype)
@@ -117,6 +118,10 @@ class Worker:
         self.lease_ttl_s = lease_ttl_s
         self.eos_token_id = eos_token_id
         self.max_kv_blocks = max_kv_blocks
+        # decode→prefill→service response relay (reference's second
+        # response topology): a DECODE instance routes its generations
+        # through the prefill peer that migrated the request in
+        self.relay_responses = relay_responses
         self.engine_kwargs = engine_kwargs or {}
         self.incarnation = int(time.time() * 1000)
 
@@ -434,6 +439,17 @@ class Worker:
                 except Exception:
                     pass
 
+    async def on_relay_generations(self, conn, gens):
+        """decode→prefill→service relay hop: forward a decode peer's
+        generations to the master over this instance's master link."""
+        if self.master_conn and not self.master_conn.closed.is_set():
+            try:
+                await self.master_conn.notify("generations", gens=gens)
+                return
+            except Exception:
+                pass
+        log.warning("worker %s: relay forward to master failed", self.name)
+
     def on_abort_request(self, conn, service_request_id: str):
         self.pending_migration.pop(service_request_id, None)
         self._post_to_engine(
@@ -453,7 +469,8 @@ class Worker:
         # account the token(s) the prefill already produced
         sp.max_tokens = max(sp.max_tokens, 1)
         meta = self.req_meta.setdefault(service_request_id, {})
-        meta.update(prompt_len=len(prompt_token_ids), params=params)
+        meta.update(prompt_len=len(prompt_token_ids), params=params,
+                    relay_via=src_name)
 
         if transport == "xgmi":
             # open the peer's cache over IPC once, then pull over xGMI
@@ -524,6 +541,27 @@ class Worker:
                     completion_tokens=o.num_output_tokens))
                 if o.finished:
                     self.req_meta.pop(rid, None)
+            relayed: Dict[str, list] = {}
+            if self.relay_responses:
+                direct = []
+                for g in gens:
+                    via = (self.req_meta.get(g["service_request_id"]) or
+                           {}).get("relay_via")
+                    if via:
+                        relayed.setdefault(via, []).append(g)
+                    else:
+                        direct.append(g)
+                gens = direct
+            for via, batch in relayed.items():
+                conn = await self._peer_conn(via)
+                try:
+                    if conn is None:
+                        raise RuntimeError(f"relay peer {via} unreachable")
+                    await conn.notify("relay_generations", gens=batch)
+                except Exception:
+                    log.warning("worker %s: relay via %s failed, pushing "
+                                "direct", self.name, via)
+                    gens.extend(batch)
             if gens and self.master_conn and not self.master_conn.closed.is_set():
                 try:
                     await self.master_conn.notify("generations", gens=gens)
@@ -660,6 +698,9 @@ def main():
     ap.add_argument("--no-graphs", action="store_true")
     ap.add_argument("--ssd-swap-dir", default=None,
                     help="directory for the SSD KV swap tier (below DRAM)")
+    ap.add_argument("--relay-responses", action="store_true",
+                    help="DECODE instances route generations through their "
+                         "prefill peer (decode->prefill->service topology)")
     ap.add_argument("--tp", type=int, default=1,
                     help="tensor-parallel group size; launch one process "
                          "per rank with RANK/WORLD_SIZE/MASTER_ADDR set "
@@ -689,6 +730,7 @@ def main():
         registry_host=args.registry_host, registry_port=args.registry_port,
         rpc_host=args.rpc_host, rpc_port=args.rpc_port,
         max_kv_blocks=args.max_kv_blocks,
+        relay_responses=args.relay_responses,
         engine_kwargs=dict(seed=args.seed, max_num_seqs=args.max_num_seqs,
                            max_batched_tokens=args.max_batched_tokens,
                            enable_graphs=not args.no_graphs,
