@@ -50,6 +50,8 @@ async def run(args):
             if device:
                 cmd += ["--device", device]
             cmd += ["--max-batched-tokens", str(args.chunk_tokens)]
+            if args.max_kv_blocks:
+                cmd += ["--max-kv-blocks", str(args.max_kv_blocks)]
             procs.append(subprocess.Popen(cmd, cwd=ROOT,
                                           stdout=subprocess.PIPE,
                                           stderr=subprocess.STDOUT))
@@ -176,6 +178,8 @@ def main():
     ap.add_argument("--chunk-tokens", type=int, default=2048,
                     help="worker max batched tokens per step (chunked "
                          "prefill interleave granularity)")
+    ap.add_argument("--max-kv-blocks", type=int, default=None,
+                    help="per-worker KV pool cap (multi-worker single GPU)")
     ap.add_argument("--startup-timeout", type=float, default=600.0)
     args = ap.parse_args()
     asyncio.run(run(args))
