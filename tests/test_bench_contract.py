@@ -1,0 +1,57 @@
+"""The driver contract for bench.py: single JSON line on rank 0, whole-job
+aggregate value, works under torch.distributed.run with N ranks (the driver
+launches N=2,4,8 for the scaling bench)."""
+import json
+import os
+import socket
+import subprocess
+import sys
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _free_port():
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _last_json(out: str) -> dict:
+    for line in reversed(out.strip().splitlines()):
+        if line.startswith("{"):
+            return json.loads(line)
+    raise AssertionError(f"no JSON line in output:\n{out[-2000:]}")
+
+
+def _check(rec: dict, n: int):
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in rec, f"missing {key}"
+    assert rec["n_gpus"] == n
+    assert rec["value"] > 0 and rec["ms_per_step"] > 0
+    assert rec["config"]["parallelism"] == f"dp{n}"
+    assert rec["data"] == "synthetic"
+
+
+def test_bench_single_rank_cpu():
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+         "--device", "cpu", "--model", "llama-tiny"],
+        cwd=ROOT, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    _check(_last_json(out.stdout), 1)
+
+
+def test_bench_torchrun_world2_cpu():
+    port = _free_port()
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--device", "cpu",
+         "--model", "llama-tiny"],
+        cwd=ROOT, capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-2000:]
+    rec = _last_json(out.stdout)
+    _check(rec, 2)
