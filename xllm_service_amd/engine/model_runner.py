@@ -6,7 +6,6 @@ next tokens (greedy via the HIP argmax kernel; stochastic via torch GPU ops).
 """
 from __future__ import annotations
 
-import math
 from typing import Dict, List, Optional, Tuple
 
 import torch
