@@ -208,3 +208,22 @@ def test_ssd_swap_tier_preserves_kv(tmp_path):
         assert g == dense_greedy(eng.model, cfg, p, 8)
     assert eng.block_manager.num_free == eng.block_manager.num_blocks
     assert os.listdir(ssd) == []          # spools deleted after swap-in
+
+
+def test_multi_token_stop_sequence():
+    """A stop sequence longer than one token halts generation at the match
+    (token-suffix matching)."""
+    eng = LLMEngine("llama-tiny", device="cpu", max_kv_blocks=64, seed=3)
+    cfg = get_config("llama-tiny")
+    torch.manual_seed(5)
+    prompt = torch.randint(0, cfg.vocab_size, (12,)).tolist()
+    # discover what the model would generate unconstrained
+    free = eng.generate([prompt], SamplingParams(max_tokens=8,
+                                                 ignore_eos=True))[0]
+    assert len(free) == 8
+    # now stop at the 2-token sequence (free[2], free[3])
+    eng2 = LLMEngine("llama-tiny", device="cpu", max_kv_blocks=64, seed=3)
+    got = eng2.generate([prompt], SamplingParams(
+        max_tokens=8, ignore_eos=True,
+        stop_sequences=[[free[2], free[3]]]))[0]
+    assert got == free[:4]          # stops right after the match

@@ -74,6 +74,13 @@ async def test_logprobs_and_tracer_and_reload(tmp_path):
         assert len(items) == 3
         assert all("logprob" in it and "top_logprobs" in it for it in items)
 
+        # multi-token stop sequences + echo (OpenAI completions params)
+        r = await client.post("/v1/completions", json={
+            "model": "llama-tiny", "prompt": "ab", "max_tokens": 4,
+            "temperature": 0.0, "ignore_eos": True, "echo": True})
+        assert r.status_code == 200, r.text
+        assert r.json()["choices"][0]["text"].startswith("ab")
+
         # tracer wrote request records
         assert os.path.exists(trace_path)
         lines = [json.loads(x) for x in open(trace_path)]

@@ -13,6 +13,8 @@ class SamplingParams:
     max_tokens: int = 128
     min_tokens: int = 0
     stop_token_ids: List[int] = field(default_factory=list)
+    # multi-token stop sequences (token-suffix match on the output)
+    stop_sequences: List[List[int]] = field(default_factory=list)
     ignore_eos: bool = False
     seed: Optional[int] = None
     logprobs: Optional[int] = None  # return top-N logprobs per token

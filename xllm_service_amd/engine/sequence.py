@@ -89,4 +89,9 @@ class Sequence:        # token lists element-wise (O(len) per membership test)
             if last in self.params.stop_token_ids:
                 self.status = SeqStatus.FINISHED_STOP
                 return True
+            for seq in self.params.stop_sequences:
+                n = len(seq)
+                if n and len(out) >= n and out[-n:] == seq:
+                    self.status = SeqStatus.FINISHED_STOP
+                    return True
         return False

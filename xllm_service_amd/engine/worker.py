@@ -91,6 +91,7 @@ def params_from_dict(d: Dict[str, Any]) -> SamplingParams:
         max_tokens=int(d.get("max_tokens", 128) or 128),
         min_tokens=int(d.get("min_tokens", 0) or 0),
         stop_token_ids=list(d.get("stop_token_ids") or []),
+        stop_sequences=[list(x) for x in (d.get("stop_sequences") or [])],
         ignore_eos=bool(d.get("ignore_eos", False)),
         seed=d.get("seed"),
         logprobs=d.get("logprobs"),

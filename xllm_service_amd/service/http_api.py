@@ -335,11 +335,17 @@ def build_app(master) -> FastAPI:
         stops = body.stop if isinstance(body.stop, list) else (
             [body.stop] if body.stop else [])
         stop_ids = []
+        stop_seqs = []
         for s in stops:
             ids = scheduler().tokenizer.encode(s)
             if len(ids) == 1:
                 stop_ids.append(ids[0])
+            elif ids:
+                stop_seqs.append(ids)
         d["stop_token_ids"] = stop_ids
+        d["stop_sequences"] = stop_seqs
+        if getattr(body, "echo", False):
+            d["echo"] = True
         return d
 
     async def _run(req: ServiceRequest, http_request: Request, stream: bool,
@@ -362,12 +368,18 @@ def build_app(master) -> FastAPI:
             await sch.cancel_request(r, reason="client disconnected")
 
         rh = master.response_handler
+        echo = bool(req.params.get("echo")) and not chat
         if stream:
             gen = (rh.stream_chat(req, on_cancel) if chat
-                   else rh.stream_completion(req, on_cancel))
+                   else rh.stream_completion(req, on_cancel,
+                                             echo_text=req.prompt_text
+                                             if echo else None))
             return StreamingResponse(gen, media_type="text/event-stream")
         result = await (rh.collect_chat(req) if chat
                         else rh.collect_completion(req))
+        if echo and "choices" in result:
+            for c in result["choices"]:
+                c["text"] = req.prompt_text + c["text"]
         status = 500 if "error" in result else 200
         master.tracer.trace(req.service_request_id, "response_out", result)
         return JSONResponse(result, status_code=status)

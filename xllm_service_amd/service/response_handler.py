@@ -163,11 +163,17 @@ class ResponseHandler:
             "usage": usage}
 
     # ------------------------------------------------------ completion paths
-    async def stream_completion(self, req: ServiceRequest,
-                                on_cancel) -> AsyncIterator[str]:
+    async def stream_completion(self, req: ServiceRequest, on_cancel,
+                                echo_text: Optional[str] = None
+                                ) -> AsyncIterator[str]:
         rid = req.service_request_id
         created = int(req.created)
         dec = IncrementalDecoder(self.tokenizer)
+        if echo_text:   # OpenAI completions echo=true: prompt leads the text
+            yield _sse({"id": rid, "object": "text_completion",
+                        "created": created, "model": req.model,
+                        "choices": [{"index": 0, "text": echo_text,
+                                     "finish_reason": None}]})
         try:
             while True:
                 gen: GenerationDelta = await asyncio.wait_for(
