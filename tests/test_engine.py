@@ -35,7 +35,8 @@ def dense_greedy(model, cfg, prompt, n_tokens):
             else:
                 h, residual = op_ref.fused_add_rmsnorm(
                     x, residual, layer.input_norm, cfg.rms_eps)
-            qkv = F.linear(h, layer.attn.qkv_proj.weight)
+            qkv = F.linear(h, layer.attn.qkv_proj.weight,
+                           layer.attn.qkv_proj.bias)  # qwen2 has qkv bias
             q, k, v = torch.split(
                 qkv, [cfg.q_size, cfg.kv_size, cfg.kv_size], dim=-1)
             q, k = op_ref.rope(pos, q.contiguous(), k.contiguous(),
@@ -227,3 +228,15 @@ def test_multi_token_stop_sequence():
         max_tokens=8, ignore_eos=True,
         stop_sequences=[[free[2], free[3]]]))[0]
     assert got == free[:4]          # stops right after the match
+
+
+def test_qwen2_text_model():
+    """Plain Qwen2 (llama stack + qkv bias) runs on the paged engine and
+    matches the dense oracle."""
+    eng = LLMEngine("qwen2-tiny", device="cpu", max_kv_blocks=64, seed=2)
+    cfg = get_config("qwen2-tiny")
+    torch.manual_seed(9)
+    prompt = torch.randint(0, cfg.vocab_size, (20,)).tolist()
+    got = eng.generate([prompt], SamplingParams(max_tokens=6,
+                                                ignore_eos=True))[0]
+    assert got == dense_greedy(eng.model, cfg, prompt, 6)

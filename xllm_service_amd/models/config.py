@@ -7,7 +7,7 @@ from dataclasses import dataclass, field
 @dataclass
 class ModelConfig:
     name: str = "llama"
-    architecture: str = "llama"          # llama | opt | qwen2_vl
+    architecture: str = "llama"          # llama | opt | qwen2 | qwen2_vl
     vocab_size: int = 128256
     hidden_size: int = 4096
     intermediate_size: int = 14336
@@ -57,6 +57,16 @@ PRESETS = {
         name="opt-125m", architecture="opt", vocab_size=50272,
         hidden_size=768, intermediate_size=3072, num_layers=12, num_heads=12,
         num_kv_heads=12, head_dim=64, max_position=2048, hidden_act="gelu"),
+    # Qwen2-7B (text-only: the VL language tower without a vision stage)
+    "qwen2-7b": ModelConfig(
+        name="qwen2-7b", architecture="qwen2", vocab_size=152064,
+        hidden_size=3584, intermediate_size=18944, num_layers=28,
+        num_heads=28, num_kv_heads=4, head_dim=128, rope_theta=1000000.0,
+        max_position=8192),
+    "qwen2-tiny": ModelConfig(
+        name="qwen2-tiny", architecture="qwen2", vocab_size=1024,
+        hidden_size=256, intermediate_size=512, num_layers=2, num_heads=4,
+        num_kv_heads=2, head_dim=64, rope_theta=10000.0, max_position=2048),
     # Qwen2-VL-7B — language tower dims
     "qwen2-vl-7b": ModelConfig(
         name="qwen2-vl-7b", architecture="qwen2_vl", vocab_size=152064,
