@@ -206,3 +206,22 @@ def test_tiktoken_tokenizer(tmp_path):
     assert len(ids) < len("hello world")  # merges happened
     factory_tk = TokenizerFactory.create(str(tmp_path))
     assert type(factory_tk).__name__ == "TiktokenTokenizer"
+
+
+def test_slo_policy_seeded_from_registration_profiles():
+    """TTFT/TPOT profiling samples in InstanceMetaInfo pre-seed the SLO
+    predictors via InstanceMgr.profile_seed_cb before any runtime
+    observations accrue."""
+    from xllm_service_amd.service.policies import SloAwarePolicy
+
+    class _Mgr:
+        pass
+
+    pol = SloAwarePolicy(_Mgr(), None)
+    ttft = [[256 * i, 40.0 + 0.5 * (256 * i)] for i in range(1, 9)]
+    tpot = [[b, b, 8.0 + 2.0 * b] for b in range(1, 9)]
+    pol.seed_from_meta("w0", ttft, tpot)
+    assert "w0" in pol.ttft and "w0" in pol.tpot
+    # predictors fit: roughly linear TTFT and affine TPOT
+    assert abs(pol.ttft["w0"].predict(1024) - (40 + 512)) < 60
+    assert abs(pol.tpot["w0"].predict(4, 4) - 16.0) < 4.0

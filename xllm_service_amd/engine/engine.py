@@ -40,6 +40,10 @@ class EngineStats:
     kv_usage: float = 0.0
     steps: int = 0
     generated_tokens: int = 0
+    # shape of the most recent step (for the worker's TTFT/TPOT profiling
+    # samples shipped in InstanceMetaInfo — reference common/types.h)
+    last_prefill_tokens: int = 0
+    last_decodes: int = 0
 
 
 class SsdSpool:
@@ -334,6 +338,9 @@ class LLMEngine:
         plan = self.scheduler.schedule()
         if plan.empty:
             return []
+        self.stats.last_prefill_tokens = sum(sp.chunk_len
+                                             for sp in plan.prefills)
+        self.stats.last_decodes = len(plan.decodes)
         new_tokens = self.runner.execute(plan, self.block_manager)
 
         outputs: List[StepOutput] = []

@@ -139,6 +139,12 @@ class Worker:
         self.pending_migration: Dict[str, Dict[str, Any]] = {}
         # remember request params for usage accounting
         self.req_meta: Dict[str, Dict[str, Any]] = {}
+        # TTFT/TPOT profiling samples measured from real steps, shipped in
+        # InstanceMetaInfo so the master's SLO predictors start seeded
+        # (reference: profiling samples in common/types.h registration)
+        self.prof_ttft: List[List[float]] = []   # [num_tokens, ms]
+        self.prof_tpot: List[List[float]] = []   # [batch, tokens, ms]
+        self._prof_shipped = False
 
         self._cmd_q: "queue.Queue" = queue.Queue()
         self._out_q: "queue.Queue" = queue.Queue()
@@ -211,6 +217,8 @@ class Worker:
             if self.engine else [],
             v_cache_ids=list(range(self.engine.cfg.num_layers))
             if self.engine else [],
+            ttft_profile=list(self.prof_ttft),
+            tpot_profile=list(self.prof_tpot),
         )
 
     async def _connect_master(self):
@@ -274,6 +282,7 @@ class Worker:
             except queue.Empty:
                 pass
             if self.engine.has_work():
+                t_step = time.monotonic()
                 try:
                     outs = self.engine.step()
                 except Exception:
@@ -292,6 +301,17 @@ class Worker:
                         outs.append(StepOutput(
                             request_id=rid, new_token_ids=[], finished=True,
                             finish_reason="abort"))
+                else:
+                    ms = (time.monotonic() - t_step) * 1000.0
+                    st = self.engine.stats
+                    if st.last_prefill_tokens > 0 and len(self.prof_ttft) < 64:
+                        self.prof_ttft.append(
+                            [float(st.last_prefill_tokens), ms])
+                    elif (st.last_prefill_tokens == 0 and st.last_decodes > 0
+                          and len(self.prof_tpot) < 64):
+                        self.prof_tpot.append(
+                            [float(st.last_decodes),
+                             float(st.last_decodes), ms])
                 if outs:
                     self._out_q.put(outs)
                 did = True
@@ -675,6 +695,15 @@ class Worker:
                     kv_removed=[bytes(h) for h in ev.removed])
                 self._ttft_samples.clear()
                 self._tbt_samples.clear()
+                # ship the first batch of profiling samples by re-PUTting
+                # the registration meta (same incarnation: a refresh)
+                if (not self._prof_shipped and self.registry is not None
+                        and (len(self.prof_ttft) >= 8
+                             or len(self.prof_tpot) >= 8)):
+                    await self.registry.put_json(
+                        self._regkey(), self.meta().to_dict(),
+                        lease_id=self._lease_id)
+                    self._prof_shipped = True
             except Exception:
                 pass
 

@@ -86,6 +86,9 @@ class InstanceMgr:
         self.lease_lost_heartbeat_timeout_s = lease_lost_heartbeat_timeout_s
         self.suspect_eviction_s = suspect_eviction_s
         self.is_master = is_master
+        # optional: seeds the SLO predictors from registration-time
+        # TTFT/TPOT profiling samples — cb(name, ttft_profile, tpot_profile)
+        self.profile_seed_cb = None
 
         self.instances: Dict[str, Instance] = {}
         self.prefill_index: List[str] = []   # PREFILL + DEFAULT (+ MIX overflow)
@@ -129,9 +132,12 @@ class InstanceMgr:
                                        reason="replaced")
                 await self._register(meta)
             elif meta.incarnation_id == existing.meta.incarnation_id:
-                # re-PUT refreshes a degraded instance back to ACTIVE
+                # re-PUT refreshes a degraded instance back to ACTIVE and
+                # may carry newly-measured TTFT/TPOT profiling samples
                 existing.set_status(InstanceStatus.ACTIVE)
                 existing.last_heartbeat = time.monotonic()
+                existing.meta = meta
+                self._seed_profiles(meta)
             # stale older incarnation: ignore
         elif ev.type == "delete":
             prev = json.loads(ev.prev_value) if ev.prev_value else {}
@@ -154,6 +160,17 @@ class InstanceMgr:
             inst.load = LoadMetrics.from_dict(json.loads(ev.value))
 
     # ---- register / deregister ----------------------------------------------
+    def _seed_profiles(self, meta: InstanceMetaInfo):
+        """Feed registration-time TTFT/TPOT profiling samples to the SLO
+        predictors (reference: profiling fields of InstanceMetaInfo)."""
+        cb = self.profile_seed_cb
+        if cb is None or not (meta.ttft_profile or meta.tpot_profile):
+            return
+        try:
+            cb(meta.name, meta.ttft_profile, meta.tpot_profile)
+        except Exception:
+            log.exception("profile seeding failed for %s", meta.name)
+
     async def _register(self, meta: InstanceMetaInfo):
         async with self._lock:
             if meta.name in self.instances:
@@ -165,6 +182,7 @@ class InstanceMgr:
                 log.warning("register %s: cannot connect rpc (%s)", meta.name, e)
                 inst.conn = None
             self.instances[meta.name] = inst
+            self._seed_profiles(meta)
             # link fan-out: P<->D peers exchange cluster info
             links: List[Tuple[Instance, InstanceMetaInfo]] = []
             it = inst.itype

@@ -189,6 +189,8 @@ class Master:
                                  target_tpot_ms=opts.target_tpot_ms)
         policy = create_policy(opts.load_balance_policy, self.instance_mgr,
                                self.kv_mgr, **policy_kwargs)
+        if hasattr(policy, "seed_from_meta"):
+            self.instance_mgr.profile_seed_cb = policy.seed_from_meta
         self.scheduler = ServiceScheduler(self.instance_mgr, self.kv_mgr,
                                           policy, self.tokenizer,
                                           chat_template, self.tracer)

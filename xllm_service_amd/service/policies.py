@@ -99,6 +99,18 @@ class SloAwarePolicy(LoadBalancePolicy):
         self.tpot: Dict[str, TPOTPredictor] = {}
 
     # ---- observation ingestion ---------------------------------------------
+    def seed_from_meta(self, instance: str, ttft_profile, tpot_profile):
+        """Registration-time profiling samples (InstanceMetaInfo
+        ttft_profile/tpot_profile) pre-seed the predictors so SLO-aware
+        selection works before any runtime observations accrue."""
+        for row in ttft_profile or []:
+            if len(row) == 2:
+                self.observe_ttft(instance, int(row[0]), float(row[1]))
+        for row in tpot_profile or []:
+            if len(row) == 3:
+                self.observe_tpot(instance, int(row[0]), int(row[1]),
+                                  float(row[2]))
+
     def observe_ttft(self, instance: str, num_tokens: int, ttft_ms: float):
         self.ttft.setdefault(instance, TTFTPredictor()).add_sample(
             num_tokens, ttft_ms)
