@@ -240,3 +240,18 @@ def test_qwen2_text_model():
     got = eng.generate([prompt], SamplingParams(max_tokens=6,
                                                 ignore_eos=True))[0]
     assert got == dense_greedy(eng.model, cfg, prompt, 6)
+
+
+def test_max_model_len_enforced():
+    """Prompts past max_model_len are rejected; max_tokens is clamped so
+    total length never exceeds the graph/rope capacity."""
+    eng = LLMEngine("llama-tiny", device="cpu", max_kv_blocks=512,
+                    max_model_len=256)
+    torch.manual_seed(0)
+    with pytest.raises(ValueError):
+        eng.add_request("too-long", list(range(30)) * 10,
+                        SamplingParams(max_tokens=4))   # 300 tokens
+    prompt = torch.randint(0, 100, (250,)).tolist()
+    out = eng.generate([prompt], SamplingParams(max_tokens=64,
+                                                ignore_eos=True))[0]
+    assert len(out) == 256 - 250                        # clamped to capacity

@@ -6,7 +6,7 @@ engine/worker.py (RPC); bench.py and tests drive it directly.
 from __future__ import annotations
 
 import time
-from dataclasses import dataclass
+from dataclasses import dataclass, replace
 from typing import Dict, List, Optional
 
 import torch
@@ -142,6 +142,8 @@ class LLMEngine:
             swap_in=self._swap_in if has_swap else None)
         if has_swap:
             self.scheduler.free_cpu_blocks = self._free_cpu_blocks
+        self.max_model_len = min(max_model_len, self.cfg.max_position)
+        max_model_len = self.max_model_len
         self.runner = ModelRunner(self.model, self.cfg, self.device,
                                   num_blocks, dtype=dtype,
                                   enable_graphs=enable_graphs,
@@ -185,9 +187,19 @@ class LLMEngine:
                     mm_embeds=None, mm_grids=None) -> None:
         if request_id in self.seqs:
             raise ValueError(f"duplicate request_id {request_id}")
+        if len(prompt_token_ids) >= self.max_model_len:
+            raise ValueError(
+                f"prompt length {len(prompt_token_ids)} exceeds "
+                f"max_model_len {self.max_model_len}")
+        params = params or SamplingParams()
+        # clamp so total_len never exceeds the graph block tables / rope
+        # table (the request finishes with finish_reason="length")
+        budget = self.max_model_len - len(prompt_token_ids)
+        if params.max_tokens > budget:
+            params = replace(params, max_tokens=budget)
         seq = Sequence(request_id=request_id,
                        prompt_token_ids=list(prompt_token_ids),
-                       params=params or SamplingParams(),
+                       params=params,
                        eos_token_id=eos_token_id if eos_token_id is not None
                        else self.eos_token_id,
                        priority=priority,

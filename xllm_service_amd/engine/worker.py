@@ -383,6 +383,18 @@ class Worker:
                                            multimodal)
             return
         sp = params_from_dict(params)
+        if (self.engine is not None
+                and len(token_ids) >= self.engine.max_model_len):
+            if self.master_conn:
+                try:
+                    await self.master_conn.notify("generations", gens=[dict(
+                        service_request_id=service_request_id, token_ids=[],
+                        finished=True, finish_reason="abort",
+                        error=f"prompt length {len(token_ids)} exceeds "
+                              f"max_model_len {self.engine.max_model_len}")])
+                except Exception:
+                    pass
+            return
         self.req_meta[service_request_id] = dict(
             params=params, routing=routing, prompt_len=len(token_ids),
             offline=offline, multimodal=multimodal)
