@@ -251,6 +251,8 @@ def test_max_model_len_enforced():
     with pytest.raises(ValueError):
         eng.add_request("too-long", list(range(30)) * 10,
                         SamplingParams(max_tokens=4))   # 300 tokens
+    with pytest.raises(ValueError):
+        eng.add_request("empty", [], SamplingParams(max_tokens=4))
     prompt = torch.randint(0, 100, (250,)).tolist()
     out = eng.generate([prompt], SamplingParams(max_tokens=64,
                                                 ignore_eos=True))[0]

@@ -187,6 +187,8 @@ class LLMEngine:
                     mm_embeds=None, mm_grids=None) -> None:
         if request_id in self.seqs:
             raise ValueError(f"duplicate request_id {request_id}")
+        if not prompt_token_ids:
+            raise ValueError("empty prompt")
         if len(prompt_token_ids) >= self.max_model_len:
             raise ValueError(
                 f"prompt length {len(prompt_token_ids)} exceeds "

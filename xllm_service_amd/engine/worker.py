@@ -383,15 +383,17 @@ class Worker:
                                            multimodal)
             return
         sp = params_from_dict(params)
-        if (self.engine is not None
-                and len(token_ids) >= self.engine.max_model_len):
+        if self.engine is not None and (
+                not token_ids
+                or len(token_ids) >= self.engine.max_model_len):
             if self.master_conn:
                 try:
                     await self.master_conn.notify("generations", gens=[dict(
                         service_request_id=service_request_id, token_ids=[],
                         finished=True, finish_reason="abort",
-                        error=f"prompt length {len(token_ids)} exceeds "
-                              f"max_model_len {self.engine.max_model_len}")])
+                        error=("empty prompt" if not token_ids else
+                               f"prompt length {len(token_ids)} exceeds "
+                               f"max_model_len {self.engine.max_model_len}"))])
                 except Exception:
                     pass
             return
