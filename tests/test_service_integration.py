@@ -231,3 +231,44 @@ async def test_heartbeat_feeds_global_kv_index():
     finally:
         await worker.stop()
         await master.stop()
+
+
+@pytest.mark.anyio
+async def test_multi_worker_concurrent_load():
+    """24 concurrent requests over two DEFAULT workers (CAR policy): all
+    complete, outputs are deterministic per prompt (identical random-init
+    engines), and both workers actually serve traffic."""
+    master = make_master(policy="CAR")
+    await master.start(serve_http=False)
+    workers = [Worker(f"w{i}", "DEFAULT", **worker_kwargs(master))
+               for i in range(2)]
+    try:
+        for w in workers:
+            await w.start()
+        await wait_for(lambda: len(master.instance_mgr.instances) == 2)
+        client = await http_client(master)
+
+        async def one(i):
+            prompt = [40 + (i % 5), 41, 42 + (i % 3)]  # 5x3 distinct prompts
+            r = await client.post("/v1/completions", json={
+                "model": "llama-tiny", "prompt": prompt, "max_tokens": 6,
+                "temperature": 0.0, "ignore_eos": True})
+            assert r.status_code == 200, r.text
+            body = r.json()
+            assert body["usage"]["completion_tokens"] == 6
+            return (tuple(prompt), body["choices"][0]["text"])
+
+        results = await asyncio.gather(*[one(i) for i in range(24)])
+        by_prompt = {}
+        for prompt, text in results:
+            assert by_prompt.setdefault(prompt, text) == text, \
+                "same prompt gave different outputs across workers"
+        # both workers saw requests (CAR balances on load)
+        served = [w for w in workers
+                  if w.engine.stats.generated_tokens > 0]
+        assert len(served) == 2, "traffic was not distributed"
+        await client.aclose()
+    finally:
+        for w in workers:
+            await w.stop()
+        await master.stop()

@@ -51,7 +51,13 @@ class CacheAwarePolicy(LoadBalancePolicy):
     """score = matched/max_blocks - gpu_cache_usage - waiting/max_waiting,
     evaluated independently for the prefill and decode pools; falls back to
     least-loaded when no prefix overlaps.
-    (reference: cache_aware_routing.cpp:22-85)"""
+    (reference: cache_aware_routing.cpp:22-85)
+
+    On top of the reference's heartbeat-reported load, the score includes
+    the scheduler's OPTIMISTIC in-flight counters (num_decoding /
+    pending_prefill_tokens, incremented at dispatch): heartbeats arrive
+    every ~1-3 s, so within a request burst the reported load is stale and
+    score ties would send the whole burst to one instance."""
     name = "CAR"
     MAX_WAITING = 64.0
 
@@ -61,7 +67,9 @@ class CacheAwarePolicy(LoadBalancePolicy):
         if max_blocks > 0:
             s += overlap.get(inst.name, 0.0) / max_blocks
         s -= inst.load.gpu_cache_usage_perc
-        s -= inst.load.waiting_requests_num / self.MAX_WAITING
+        inflight = (inst.num_decoding +
+                    inst.pending_prefill_tokens / 1024.0)
+        s -= (inst.load.waiting_requests_num + inflight) / self.MAX_WAITING
         return s
 
     def select_instances_pair(self, token_ids):
