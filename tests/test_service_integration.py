@@ -272,3 +272,36 @@ async def test_multi_worker_concurrent_load():
         for w in workers:
             await w.stop()
         await master.stop()
+
+
+@pytest.mark.anyio
+async def test_stream_disconnect_cancels_on_worker():
+    """Closing an SSE stream mid-generation aborts the request on the
+    worker (client-disconnect cancellation, reference scheduler.cpp
+    handle_generation cancel path)."""
+    master = make_master(policy="RR")
+    await master.start(serve_http=False)
+    worker = Worker("w0", "DEFAULT", **worker_kwargs(master))
+    try:
+        await worker.start()
+        await wait_for(lambda: master.scheduler.has_available_instances())
+        client = await http_client(master)
+        got_chunks = 0
+        async with client.stream("POST", "/v1/completions", json={
+                "model": "llama-tiny", "prompt": [3, 4, 5],
+                "max_tokens": 4096, "temperature": 0.0, "ignore_eos": True,
+                "stream": True}) as resp:
+            assert resp.status_code == 200
+            async for line in resp.aiter_lines():
+                if line.startswith("data: "):
+                    got_chunks += 1
+                    if got_chunks >= 3:
+                        break   # client walks away mid-stream
+        assert got_chunks >= 3
+        # the abort must reach the worker's engine
+        await wait_for(lambda: not worker.engine.has_work(), timeout=10.0)
+        assert len(master.scheduler.requests) == 0
+        await client.aclose()
+    finally:
+        await worker.stop()
+        await master.stop()
