@@ -164,6 +164,8 @@ class LLMEngine:
         (configs/tunableop_gfx950.csv, produced by PYTORCH_TUNABLEOP_TUNING
         on an MI355X) so library GEMMs use the fastest algo per shape."""
         import os
+        if os.environ.get("PYTORCH_TUNABLEOP_TUNING") == "1":
+            return  # explicit tuning session manages its own state/file
         csv = os.path.join(os.path.dirname(os.path.dirname(
             os.path.dirname(os.path.abspath(__file__)))),
             "configs", "tunableop_gfx950.csv")
