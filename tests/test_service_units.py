@@ -225,3 +225,55 @@ def test_slo_policy_seeded_from_registration_profiles():
     # predictors fit: roughly linear TTFT and affine TPOT
     assert abs(pol.ttft["w0"].predict(1024) - (40 + 512)) < 60
     assert abs(pol.tpot["w0"].predict(4, 4) - 16.0) < 4.0
+
+
+@pytest.mark.parametrize("model,family,rtag,ttags", [
+    ("DeepSeek-V3-0324", "deepseek_v3", "<think>",
+     ("<｜tool▁call▁begin｜>", "<｜tool▁call▁end｜>")),
+    ("Kimi-K2-Instruct", "kimi_k2", "<think>",
+     ("<|tool_call_begin|>", "<|tool_call_end|>")),
+    ("GLM-4.5-Air", "glm4_moe", "<think>", ("<tool_call>", "</tool_call>")),
+    ("step3-32k", "step3", "<think>", ("<tool_call>", "</tool_call>")),
+    ("Qwen2.5-72B", "qwen2", None, ("<tool_call>", "</tool_call>")),
+    ("Llama-3.1-70B", "llama", None, ("<|python_tag|>", "<|eom_id|>")),
+])
+def test_parser_families(model, family, rtag, ttags):
+    """Every supported model family's tag set parses reasoning and tool
+    calls, non-stream and streaming (reference parser bridge tag sets)."""
+    from xllm_service_amd.service.parsers import infer_model_family
+    assert infer_model_family(model) == family
+    rp, tp = make_parsers(model)
+    payload = '{"name": "f", "arguments": {"x": 1}}'
+    text = ""
+    if rtag:
+        text += f"{rtag}thinking...{rtag.replace('<', '</', 1)}"
+    text += f"hello {ttags[0]}{payload}{ttags[1]} bye"
+    if rp is not None:
+        reasoning, text2 = rp.extract(text)
+        assert reasoning == "thinking..."
+    else:
+        assert rtag is None
+        text2 = text
+    assert tp is not None
+    rest, calls = tp.extract(text2)
+    assert len(calls) == 1 and calls[0].name == "f"
+    assert "hello" in rest and "bye" in rest
+
+    # streaming variant fed in awkward 3-char pieces
+    srp, stp = make_stream_parsers(model)
+    out_text, out_calls, reason_acc = "", [], ""
+    buf = text
+    for i in range(0, len(buf), 3):
+        piece = buf[i:i + 3]
+        if srp is not None:
+            rdelta, piece = srp.feed(piece)
+            if rdelta:
+                reason_acc += rdelta
+        if stp is not None and piece:
+            piece, cs = stp.feed(piece)
+            out_calls.extend(cs)
+    if stp is not None:
+        out_calls.extend(stp.flush())
+    if srp is not None:
+        assert reason_acc == "thinking..."
+    assert len(out_calls) == 1 and out_calls[0].name == "f"
