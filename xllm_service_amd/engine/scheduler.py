@@ -57,6 +57,7 @@ class EngineScheduler:
         self.running: List[Sequence] = []
         self.num_swap_outs = 0
         self.num_swap_ins = 0
+        self.num_preempts = 0
 
     # ---- API ----------------------------------------------------------------
     def add(self, seq: Sequence):
@@ -112,6 +113,7 @@ class EngineScheduler:
         request and restore it)."""
         self.running.remove(victim)
         victim.preempt_count += 1
+        self.num_preempts += 1
         if self.swap_out is not None and victim.prefill_done:
             cpu_blocks = self.swap_out(victim)
             if cpu_blocks is not None:
