@@ -31,6 +31,8 @@ class StepOutput:
     num_prompt_tokens: int = 0
     num_output_tokens: int = 0
     first_token: bool = False
+    ttft_ms: Optional[float] = None   # set on the first emitted token
+    tbt_ms: Optional[float] = None    # inter-token gap (decode steps)
 
 
 @dataclass
@@ -316,14 +318,26 @@ class LLMEngine:
                                   params: Optional[SamplingParams] = None,
                                   eos_token_id: Optional[int] = None,
                                   priority: int = 0,
-                                  mrope_delta: int = 0) -> None:
+                                  mrope_delta: int = 0) -> Optional[str]:
         """Resume a request whose prompt KV was migrated into block_ids
         (decode side). No recompute: decode continues from the first
         prefill-produced token. mrope_delta carries the prefill side's
-        M-RoPE text-position offset (Qwen2-VL)."""
+        M-RoPE text-position offset (Qwen2-VL).
+
+        Returns None when the sequence is live; otherwise the finish reason
+        ("length"/"stop": the prefill token(s) already completed the request
+        — e.g. a near-limit prompt, or an EOS first token) — the caller must
+        free block_ids and report the finish itself."""
+        params = params or SamplingParams()
+        # re-apply the add_request max_tokens clamp: the prefill side's
+        # params are un-clamped, and total_len must never exceed the rope /
+        # graph-table capacity of THIS engine
+        budget = self.max_model_len - len(prompt_token_ids)
+        if params.max_tokens > budget:
+            params = replace(params, max_tokens=max(budget, 0))
         seq = Sequence(request_id=request_id,
                        prompt_token_ids=list(prompt_token_ids),
-                       params=params or SamplingParams(),
+                       params=params,
                        eos_token_id=eos_token_id if eos_token_id is not None
                        else self.eos_token_id,
                        priority=priority)
@@ -332,12 +346,16 @@ class LLMEngine:
         seq.num_computed_tokens = seq.prompt_len
         seq.output_token_ids = list(first_token_ids)
         seq.migrated_in = True
+        if seq.check_finish():
+            return ("stop" if seq.status == SeqStatus.FINISHED_STOP
+                    else "length")
         seq.status = SeqStatus.RUNNING
         if seq.first_token_time is None:
             import time as _t
             seq.first_token_time = _t.monotonic()
         self.seqs[request_id] = seq
         self.scheduler.running.append(seq)
+        return None
 
     def abort_request(self, request_id: str) -> bool:
         seq = self.scheduler.abort(request_id)
@@ -364,7 +382,7 @@ class LLMEngine:
             seq = self.seqs.get(rid)
             if seq is None:
                 continue
-            first = not seq.output_token_ids
+            first = seq.num_emitted == 0
             seq.append_token(tok)
             if lp is not None:
                 seq.cumulative_logprob += lp["token_logprob"]
@@ -378,9 +396,12 @@ class LLMEngine:
                     "stop" if seq.status == SeqStatus.FINISHED_STOP else
                     "length" if seq.status == SeqStatus.FINISHED_LENGTH else
                     None),
-                num_prompt_tokens=seq.prompt_len,
-                num_output_tokens=len(seq.output_token_ids),
+                num_prompt_tokens=seq.orig_prompt_len,
+                num_output_tokens=seq.num_emitted,
                 first_token=first,
+                ttft_ms=((seq.first_token_time - seq.arrival_time) * 1000.0
+                         if first else None),
+                tbt_ms=seq.last_tbt_ms if not first else None,
             ))
             self.stats.generated_tokens += 1
         # advance prefill progress + retire finished sequences

@@ -126,8 +126,24 @@ class EngineScheduler:
         self.bm.free(victim)
         victim.status = SeqStatus.PREEMPTED
         victim.num_computed_tokens = 0
-        victim.prompt_token_ids = victim.all_token_ids()
-        victim.output_token_ids = []
+        n_out = len(victim.output_token_ids)
+        if n_out:
+            # fold emitted tokens into the recompute prompt, keeping the
+            # emitted-token count so max_tokens/min_tokens/usage stay exact
+            if victim.mrope_pos is not None:
+                # extend the 3-D M-RoPE table for the folded decode tokens:
+                # a text token at overall index i decodes at position
+                # i + mrope_delta on all three rows (model_runner decode path)
+                import numpy as np
+                cur = victim.mrope_pos.shape[1]
+                cols = (np.arange(cur, cur + n_out,
+                                  dtype=victim.mrope_pos.dtype)
+                        + victim.mrope_delta)
+                victim.mrope_pos = np.concatenate(
+                    [victim.mrope_pos, np.tile(cols, (3, 1))], axis=1)
+            victim.prompt_token_ids = victim.all_token_ids()
+            victim.output_token_ids = []
+            victim.num_folded_output_tokens += n_out
         self.waiting.appendleft(victim)
 
     def _preempt_one(self) -> bool:

@@ -35,6 +35,10 @@ class Sequence:        # token lists element-wise (O(len) per membership test)
 
     status: SeqStatus = SeqStatus.WAITING
     output_token_ids: List[int] = field(default_factory=list)
+    # recompute preemption folds already-emitted output tokens into
+    # prompt_token_ids; this counter keeps max_tokens/min_tokens/usage
+    # accounting correct across the fold
+    num_folded_output_tokens: int = 0
     block_table: List[int] = field(default_factory=list)
     cpu_block_table: List[int] = field(default_factory=list)  # dram tier
     num_computed_tokens: int = 0          # prompt tokens already prefilled
@@ -53,11 +57,24 @@ class Sequence:        # token lists element-wise (O(len) per membership test)
     # holds them until the decode instance has pulled the blocks)
     hold_blocks: bool = False
     first_token_time: Optional[float] = None
+    last_token_time: Optional[float] = None
+    last_tbt_ms: float = 0.0     # inter-token gap of the latest decode step
     cumulative_logprob: float = 0.0
 
     @property
     def prompt_len(self) -> int:
         return len(self.prompt_token_ids)
+
+    @property
+    def num_emitted(self) -> int:
+        """Total output tokens produced so far, including ones folded into
+        the prompt by a recompute preemption."""
+        return len(self.output_token_ids) + self.num_folded_output_tokens
+
+    @property
+    def orig_prompt_len(self) -> int:
+        """Prompt length as submitted (excludes folded output tokens)."""
+        return len(self.prompt_token_ids) - self.num_folded_output_tokens
 
     @property
     def total_len(self) -> int:
@@ -72,15 +89,20 @@ class Sequence:        # token lists element-wise (O(len) per membership test)
 
     def append_token(self, token_id: int):
         self.output_token_ids.append(token_id)
+        now = time.monotonic()
         if self.first_token_time is None:
-            self.first_token_time = time.monotonic()
+            self.first_token_time = now
+        elif self.last_token_time is not None:
+            self.last_tbt_ms = (now - self.last_token_time) * 1000.0
+        self.last_token_time = now
 
     def check_finish(self) -> bool:
         out = self.output_token_ids
-        if len(out) >= self.params.max_tokens:
+        emitted = self.num_emitted
+        if emitted >= self.params.max_tokens:
             self.status = SeqStatus.FINISHED_LENGTH
             return True
-        if len(out) >= self.params.min_tokens and out:
+        if emitted >= self.params.min_tokens and out:
             last = out[-1]
             if (not self.params.ignore_eos and self.eos_token_id is not None
                     and last == self.eos_token_id):
@@ -91,7 +113,12 @@ class Sequence:        # token lists element-wise (O(len) per membership test)
                 return True
             for seq in self.params.stop_sequences:
                 n = len(seq)
-                if n and len(out) >= n and out[-n:] == seq:
-                    self.status = SeqStatus.FINISHED_STOP
-                    return True
+                if n and emitted >= n:
+                    # a recompute fold may have moved part of the tail into
+                    # prompt_token_ids; match across the fold boundary
+                    tail = (out[-n:] if len(out) >= n
+                            else self.all_token_ids()[-n:])
+                    if tail == seq:
+                        self.status = SeqStatus.FINISHED_STOP
+                        return True
         return False

@@ -532,12 +532,25 @@ class Worker:
             except Exception:
                 self.engine.free_blocks(blocks)
                 raise
-            self.engine.activate_migrated_request(
+            fin = self.engine.activate_migrated_request(
                 service_request_id, prompt_token_ids, first_token_ids,
                 blocks, sp, priority=1 if offline else 0,
                 mrope_delta=mrope_delta)
+            if fin is not None:
+                self.engine.free_blocks(blocks)
+            return fin
 
-        await self._run_on_engine(_recv)
+        fin = await self._run_on_engine(_recv)
+        if fin is not None:
+            # the prefill token(s) already exhausted the request's budget
+            # (or ended on EOS): report the finish through the normal push
+            # path (handles the relay topology too)
+            from .engine import StepOutput
+            self._out_q.put([StepOutput(
+                service_request_id, new_token_ids=[], finished=True,
+                finish_reason=fin,
+                num_prompt_tokens=len(prompt_token_ids),
+                num_output_tokens=len(first_token_ids))])
         return True
 
     # ------------------------------------------------------------- push loop
@@ -550,6 +563,12 @@ class Worker:
             migrations = []
             for o in outs:
                 rid = o.request_id
+                # heartbeat LatencyMetrics: per-request TTFT / inter-token
+                # gaps measured at emission (recent_max_* in the reference)
+                if o.ttft_ms is not None and len(self._ttft_samples) < 4096:
+                    self._ttft_samples.append(o.ttft_ms)
+                elif o.tbt_ms and len(self._tbt_samples) < 4096:
+                    self._tbt_samples.append(o.tbt_ms)
                 mig = self.pending_migration.get(rid)
                 if mig is not None and o.finished:
                     # first token produced by prefill: announce + migrate

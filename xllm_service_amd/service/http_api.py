@@ -124,7 +124,8 @@ def build_app(master) -> FastAPI:
             service_request_id=make_request_id("msg"),
             kind="chat", model=body.get("model") or master.model_id,
             stream=bool(body.get("stream")), token_ids=token_ids,
-            prompt_text=prompt_text, params=params)
+            prompt_text=prompt_text, params=params,
+            stop_texts=[s for s in (body.get("stop_sequences") or []) if s])
         sch = scheduler()
         try:
             sch.schedule(req)
@@ -183,7 +184,7 @@ def build_app(master) -> FastAPI:
                     raise
             return StreamingResponse(gen(), media_type="text/event-stream")
 
-        token_ids_out, usage, finish, err = \
+        token_ids_out, text_out, usage, finish, err = \
             await master.response_handler._collect(req)
         if err:
             return JSONResponse({"type": "error", "error": {
@@ -191,8 +192,7 @@ def build_app(master) -> FastAPI:
         return {
             "id": req.service_request_id, "type": "message",
             "role": "assistant", "model": req.model,
-            "content": [{"type": "text",
-                         "text": master.tokenizer.decode(token_ids_out)}],
+            "content": [{"type": "text", "text": text_out}],
             "stop_reason": ("max_tokens" if finish == "length"
                             else "end_turn"),
             "usage": {"input_tokens": usage["prompt_tokens"],
@@ -251,11 +251,13 @@ def build_app(master) -> FastAPI:
             prompt_text = body.prompt if isinstance(body.prompt, str) \
                 else "".join(body.prompt or [])
             token_ids = scheduler().tokenizer.encode(prompt_text)
+        params = _sampling_dict(body)
+        stop_texts = params.pop("stop_texts", [])
         req = ServiceRequest(
             service_request_id=make_request_id("cmpl"),
             kind="completion", model=body.model or master.model_id,
             stream=body.stream, token_ids=token_ids, prompt_text=prompt_text,
-            params=_sampling_dict(body), offline=body.offline)
+            params=params, offline=body.offline, stop_texts=stop_texts)
         return await _run(req, request, body.stream, chat=False)
 
     @app.post("/v1/chat/completions")
@@ -310,6 +312,7 @@ def build_app(master) -> FastAPI:
             token_ids = [pad_id] * n_img_tokens + token_ids
             multimodal = {"images": images}
         params = _sampling_dict(body)
+        stop_texts = params.pop("stop_texts", [])
         if body.max_completion_tokens is not None:
             params["max_tokens"] = body.max_completion_tokens
         elif body.max_tokens is None:
@@ -318,7 +321,8 @@ def build_app(master) -> FastAPI:
             service_request_id=make_request_id("chatcmpl"),
             kind="chat", model=body.model or master.model_id,
             stream=body.stream, token_ids=token_ids, prompt_text=prompt_text,
-            params=params, offline=body.offline, multimodal=multimodal)
+            params=params, offline=body.offline, multimodal=multimodal,
+            stop_texts=stop_texts)
         return await _run(req, request, body.stream, chat=True)
 
     def _sampling_dict(body) -> Dict[str, Any]:
@@ -344,6 +348,9 @@ def build_app(master) -> FastAPI:
                 stop_seqs.append(ids)
         d["stop_token_ids"] = stop_ids
         d["stop_sequences"] = stop_seqs
+        # raw stop strings for the service-layer text scanner (popped off
+        # before the params dict travels to the worker)
+        d["stop_texts"] = [s for s in stops if s]
         if getattr(body, "echo", False):
             d["echo"] = True
         return d
@@ -351,7 +358,9 @@ def build_app(master) -> FastAPI:
     async def _run(req: ServiceRequest, http_request: Request, stream: bool,
                    chat: bool):
         sch = scheduler()
-        req.is_disconnected = lambda: False  # updated below for streams
+        # non-stream disconnects are detected by polling this (streams
+        # cancel via GeneratorExit) — see ServiceScheduler.handle_generation
+        req.http_request = http_request
         master.tracer.trace(req.service_request_id, "request_in",
                             {"kind": req.kind, "tokens": len(req.token_ids)})
         try:

@@ -309,13 +309,25 @@ class InstanceMgr:
                         inst.name)
 
     async def _probe_health(self, inst: Instance) -> bool:
-        if inst.conn is None or inst.conn.closed.is_set():
+        """Probe over a FRESH connection (reference: a fresh HTTP GET
+        /health, instance_mgr.cpp:500-539). Probing the existing RPC
+        connection would report a wedged-but-connected worker healthy."""
+        try:
+            conn = await asyncio.wait_for(
+                msgrpc.connect(inst.meta.rpc_host, inst.meta.rpc_port),
+                self.probe_timeout_s)
+        except Exception:
             return False
         try:
-            r = await inst.conn.call("health", timeout=self.probe_timeout_s)
+            r = await conn.call("health", timeout=self.probe_timeout_s)
             return bool(r)
         except Exception:
             return False
+        finally:
+            try:
+                await conn.close()
+            except Exception:
+                pass
 
     async def _reconcile_loop(self):
         while True:

@@ -27,6 +27,9 @@ class GenerationDelta:
     usage_completion_tokens: int = 0
     logprobs: Optional[List[float]] = None
     error: Optional[str] = None
+    # text-level stop trimming: when set, this is the delta's text and the
+    # response handler must use it instead of decoding token_ids
+    text: Optional[str] = None
 
 
 @dataclass
@@ -58,4 +61,12 @@ class ServiceRequest:
     output_queue: "asyncio.Queue[GenerationDelta]" = field(
         default_factory=asyncio.Queue)
     is_disconnected: Callable[[], bool] = lambda: False
+    # non-stream disconnect detection: the live HTTP request object
+    # (awaited .is_disconnected(), rate-limited in handle_generation)
+    http_request: Optional[Any] = None
+    _last_disc_check: float = 0.0
+    # OpenAI text-level stop strings (engine token-match is the fast path;
+    # the scanner is authoritative — see service/stop_scanner.py)
+    stop_texts: List[str] = field(default_factory=list)
+    stop_scanner: Optional[Any] = None
     trace_cb: Optional[Callable[[str, Any], None]] = None

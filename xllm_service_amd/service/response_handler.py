@@ -66,7 +66,8 @@ class ResponseHandler:
                     yield _sse({"error": {"message": gen.error,
                                           "type": "server_error"}})
                     break
-                text = dec.push(gen.token_ids) if gen.token_ids else ""
+                text = (gen.text if gen.text is not None
+                        else dec.push(gen.token_ids) if gen.token_ids else "")
                 if not sent_role and (text or gen.finished):
                     yield chunk({"role": "assistant", "content": ""})
                     sent_role = True
@@ -125,10 +126,9 @@ class ResponseHandler:
 
     # -------------------------------------------------------- chat non-stream
     async def collect_chat(self, req: ServiceRequest) -> dict:
-        token_ids, usage, finish, err = await self._collect(req)
+        token_ids, text, usage, finish, err = await self._collect(req)
         if err:
             return {"error": {"message": err, "type": "server_error"}}
-        text = self.tokenizer.decode(token_ids)
         rp, tp = make_parsers(req.model, self.parser_mode)
         reasoning = None
         tool_calls = []
@@ -182,7 +182,8 @@ class ResponseHandler:
                     yield _sse({"error": {"message": gen.error,
                                           "type": "server_error"}})
                     break
-                text = dec.push(gen.token_ids) if gen.token_ids else ""
+                text = (gen.text if gen.text is not None
+                        else dec.push(gen.token_ids) if gen.token_ids else "")
                 if text or gen.logprobs:
                     choice = {"index": 0, "text": text, "finish_reason": None}
                     lp = self._completion_logprobs(gen.token_ids,
@@ -215,11 +216,10 @@ class ResponseHandler:
             await on_cancel(req)
 
     async def collect_completion(self, req: ServiceRequest) -> dict:
-        token_ids, usage, finish, err = await self._collect(req)
+        token_ids, text, usage, finish, err = await self._collect(req)
         if err:
             return {"error": {"message": err, "type": "server_error"}}
-        choice = {"index": 0, "text": self.tokenizer.decode(token_ids),
-                  "finish_reason": finish}
+        choice = {"index": 0, "text": text, "finish_reason": finish}
         lp = self._completion_logprobs(
             token_ids, getattr(req, "_collected_logprobs", []))
         if lp is not None:
@@ -232,7 +232,12 @@ class ResponseHandler:
 
     # ---------------------------------------------------------------- common
     async def _collect(self, req: ServiceRequest):
+        """Drain the request to completion. Returns
+        (token_ids, text, usage, finish_reason, error); text honours
+        per-delta text overrides (text-level stop trimming)."""
         token_ids = []
+        text_parts = []
+        dec = IncrementalDecoder(self.tokenizer)
         logprobs = []
         usage = {"prompt_tokens": 0, "completion_tokens": 0, "total_tokens": 0}
         finish = "stop"
@@ -240,8 +245,10 @@ class ResponseHandler:
             gen: GenerationDelta = await asyncio.wait_for(
                 req.output_queue.get(), STREAM_TIMEOUT_S)
             if gen.error:
-                return token_ids, usage, finish, gen.error
+                return token_ids, "".join(text_parts), usage, finish, gen.error
             token_ids.extend(gen.token_ids)
+            text_parts.append(gen.text if gen.text is not None
+                              else dec.push(gen.token_ids))
             if gen.logprobs:
                 logprobs.extend(gen.logprobs)
             if gen.finished:
@@ -251,7 +258,7 @@ class ResponseHandler:
                          "total_tokens": gen.usage_prompt_tokens +
                          gen.usage_completion_tokens}
                 req._collected_logprobs = logprobs
-                return token_ids, usage, finish, None
+                return token_ids, "".join(text_parts), usage, finish, None
 
     def _completion_logprobs(self, token_ids, logprobs):
         """OpenAI completion-style logprobs block."""

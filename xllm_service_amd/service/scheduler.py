@@ -82,6 +82,9 @@ class ServiceScheduler:
             pair.prefill.num_scheduled += 1
             pair.prefill.num_prefill_unfinished += 1
             pair.prefill.pending_prefill_tokens += len(req.token_ids)
+        if req.stop_texts:
+            from .stop_scanner import StopTextScanner
+            req.stop_scanner = StopTextScanner(self.tokenizer, req.stop_texts)
         if pair.decode is not None:
             req.decode_name = pair.decode.name
             req.decode_incarnation = pair.decode.meta.incarnation_id
@@ -135,6 +138,20 @@ class ServiceScheduler:
         if req.is_disconnected():
             await self.cancel_request(req, reason="client disconnected")
             return False
+        if req.http_request is not None and not req.stream:
+            # non-stream clients have no generator to cancel; poll the HTTP
+            # connection (rate-limited so the hot path stays cheap). Stream
+            # paths cancel via GeneratorExit already.
+            now0 = time.monotonic()
+            if now0 - req._last_disc_check > 0.5:
+                req._last_disc_check = now0
+                try:
+                    if await req.http_request.is_disconnected():
+                        await self.cancel_request(
+                            req, reason="client disconnected")
+                        return False
+                except Exception:
+                    pass
 
         now = time.monotonic()
         toks = list(gen.get("token_ids") or [])
@@ -168,10 +185,35 @@ class ServiceScheduler:
                 inst.pending_prefill_tokens = max(
                     0, inst.pending_prefill_tokens - len(req.token_ids))
 
+        # text-level stop strings (OpenAI semantics): scan decoded text,
+        # trim at the match; engine token-level matching is the fast path
+        finished = bool(gen.get("finished"))
+        finish_reason = gen.get("finish_reason")
+        text_override = None
+        scan = req.stop_scanner
+        if scan is not None and not gen.get("error"):
+            stop_hit = False
+            if toks:
+                toks, text_override, stop_hit = scan.feed(toks)
+            if stop_hit:
+                finished, finish_reason = True, "stop"
+                # the engine may still be generating: stop it
+                for name in {req.prefill_name, req.decode_name} - {None}:
+                    inst = self.mgr.get(name)
+                    if inst and inst.conn:
+                        try:
+                            await inst.conn.notify(
+                                "abort_request", service_request_id=rid)
+                        except Exception:
+                            pass
+            elif finished:
+                toks = toks + scan.flush()
+
         delta = GenerationDelta(
             token_ids=toks,
-            finished=bool(gen.get("finished")),
-            finish_reason=gen.get("finish_reason"),
+            finished=finished,
+            finish_reason=finish_reason,
+            text=text_override,
             finished_on_prefill=bool(gen.get("finished_on_prefill")),
             usage_prompt_tokens=gen.get("prompt_tokens", len(req.token_ids)),
             usage_completion_tokens=gen.get("completion_tokens",
@@ -194,6 +236,13 @@ class ServiceScheduler:
     async def cancel_request(self, req: ServiceRequest, reason: str):
         metrics.REQUEST_CANCEL_TOTAL.labels(reason=reason).inc()
         self.finish_request(req)
+        # unblock any pending collector (non-stream handler task)
+        try:
+            req.output_queue.put_nowait(GenerationDelta(
+                token_ids=[], finished=True, finish_reason="abort",
+                error=f"cancelled: {reason}"))
+        except Exception:
+            pass
         # tell the instance(s) to stop computing
         for name in {req.prefill_name, req.decode_name} - {None}:
             inst = self.mgr.get(name)
