@@ -1,29 +1,48 @@
 #!/usr/bin/env python3
-"""Flagship serving benchmark — the driver contract (see repo instructions).
+"""Flagship benchmark — the driver contract (see repo instructions).
 
-Measures the BASELINE.json metric: output tok/s of continuous-batching
-serving of Llama-3-8B (bf16, synthetic prompts, random-init weights) on
-N MI355X GPUs. A "step" is one engine step (one decode iteration of the
-running batch, plus any admitted prefill chunks). For N>1 ranks each run an
-independent engine replica (data parallel / weak scaling: this mirrors the
-instance pool of the serving deployment); the reported value is the
-whole-job aggregate output tokens per second.
+Default mode measures the BASELINE.json headline the honest way: SLO-goodput
+(output tok/s from requests whose TTFT meets the 1 s SLO) through the FULL
+serving stack — HTTP (uvicorn, real TCP) -> master scheduler -> msgrpc ->
+per-GPU worker processes -> SSE streaming back — under open-loop Poisson
+arrivals whose rate is auto-calibrated to ~0.9x the measured capacity.
 
-  python bench.py --gpus 1 --steps 64 --warmup 16
+Topology scales with --gpus N (one worker process per GPU):
+  N=1            one DEFAULT (colocated prefill+decode) worker
+  N>=2           PD-disaggregated: max(1, N//4) PREFILL + rest DECODE
+                 (8 GPUs -> 2P+6D, BASELINE config 3), KV blocks migrating
+                 prefill->decode over xGMI (hipIpc + peer copies)
+
+A "step" is one COMPLETED request in steady state: after --warmup completed
+requests under the Poisson load, the timed region spans exactly --steps
+further completions (barrier + torch.cuda.synchronize on both sides, MAX
+elapsed over ranks). value = sum of output tokens of SLO-passing requests
+completed in the window / elapsed — the whole-job aggregate.
+
+  python bench.py --gpus 1 --steps 20 --warmup 5
   python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
-      --master-addr 127.0.0.1 bench.py --gpus 8 --steps 64 --warmup 16
+      --master-addr 127.0.0.1 bench.py --gpus 8 --steps 120 --warmup 10
+
+--mode engine keeps the round-1 closed-loop engine-step microbench (kernel
+iteration tool; its metric string says so).
 """
 from __future__ import annotations
 
 import argparse
+import asyncio
 import json
 import os
+import random
+import socket
+import subprocess
+import sys
 import time
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
 
 # hipBLASLt/rocBLAS algo selection tuned offline on MI355X (TunableOp);
 # must be configured before the first torch import in the process.
-_TUNABLE = os.path.join(os.path.dirname(os.path.abspath(__file__)),
-                        "configs", "tunableop_gfx950.csv")
+_TUNABLE = os.path.join(ROOT, "configs", "tunableop_gfx950.csv")
 if os.path.exists(_TUNABLE):
     os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
     os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
@@ -35,16 +54,408 @@ import torch
 def parse_args():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=64)
-    ap.add_argument("--warmup", type=int, default=16)
+    ap.add_argument("--steps", type=int, default=20,
+                    help="serving: timed completed requests; "
+                         "engine: timed engine steps")
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--mode", choices=("serving", "engine"),
+                    default="serving")
     ap.add_argument("--model", default="llama-3-8b")
     ap.add_argument("--concurrency", type=int, default=64,
-                    help="concurrent requests per GPU")
+                    help="ramp/calibration concurrency per decode GPU")
     ap.add_argument("--input-len", type=int, default=1024)
     ap.add_argument("--output-len", type=int, default=1024)
-    ap.add_argument("--max-batched-tokens", type=int, default=16384)
+    ap.add_argument("--max-batched-tokens", type=int, default=8192,
+                    help="worker chunked-prefill budget per step")
     ap.add_argument("--device", default=None)
+    ap.add_argument("--policy", default="CAR")
+    ap.add_argument("--slo-ttft-ms", type=float, default=1000.0)
+    ap.add_argument("--pace", type=float, default=0.90,
+                    help="Poisson arrival rate as a fraction of the "
+                         "ramp-measured capacity")
+    ap.add_argument("--arrival-rate", type=float, default=0.0,
+                    help="req/s override (0 = auto-calibrate)")
+    ap.add_argument("--ramp-s", type=float, default=0.0,
+                    help="capacity-calibration window (0 = auto)")
+    ap.add_argument("--push-interval-ms", type=float, default=25.0,
+                    help="worker->master token push coalescing window")
+    ap.add_argument("--max-kv-blocks", type=int, default=0)
+    ap.add_argument("--startup-timeout", type=float, default=900.0)
     return ap.parse_args()
+
+
+# --------------------------------------------------------------------------
+# engine mode (round-1 closed-loop engine-step microbench)
+# --------------------------------------------------------------------------
+def run_engine_mode(args, rank, world, local_rank, use_gpu, dist, device,
+                    model_name):
+    from xllm_service_amd.engine.engine import LLMEngine
+    from xllm_service_amd.engine.sampling import SamplingParams
+    from xllm_service_amd.models.config import get_config
+
+    cfg = get_config(model_name)
+    eng = LLMEngine(model_name, device=device,
+                    max_num_seqs=args.concurrency * 2,
+                    max_batched_tokens=args.max_batched_tokens,
+                    max_kv_blocks=None if use_gpu else 4096,
+                    seed=0)
+    torch.manual_seed(1234 + rank)
+    rid = 0
+
+    def feed(n):
+        nonlocal rid
+        for _ in range(n):
+            prompt = torch.randint(0, cfg.vocab_size,
+                                   (args.input_len,)).tolist()
+            eng.add_request(f"r{rank}-{rid}", prompt,
+                            SamplingParams(max_tokens=args.output_len,
+                                           ignore_eos=True))
+            rid += 1
+
+    feed(args.concurrency)
+
+    def one_step():
+        outs = eng.step()
+        done = sum(1 for o in outs if o.finished)
+        if done:
+            feed(done)
+        return sum(len(o.new_token_ids) for o in outs)
+
+    for _ in range(args.warmup):
+        one_step()
+    if dist:
+        dist.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    generated = 0
+    for _ in range(args.steps):
+        generated += one_step()
+    if use_gpu:
+        torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+    if dist:
+        te = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        elapsed = float(te.item())
+        tg = torch.tensor([float(generated)], dtype=torch.float64)
+        dist.all_reduce(tg, op=dist.ReduceOp.SUM)
+        generated = int(tg.item())
+    n_gpus = world if world > 1 else (1 if use_gpu else args.gpus)
+    if rank == 0:
+        print(json.dumps({
+            "metric": "engine-step throughput (out tok/s, closed loop, "
+                      "no serving stack), Llama-3-8B",
+            "value": round(generated / elapsed, 2),
+            "unit": "tok/s", "n_gpus": n_gpus,
+            "steps": args.steps, "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True, "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32", "data": "synthetic",
+            "config": {"model": model_name,
+                       "global_batch": args.concurrency * n_gpus,
+                       "seq_len": args.input_len + args.output_len,
+                       "input_len": args.input_len,
+                       "output_len": args.output_len,
+                       "parallelism": f"dp{n_gpus}"},
+        }))
+
+
+# --------------------------------------------------------------------------
+# serving mode helpers
+# --------------------------------------------------------------------------
+def free_port():
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def topology(n: int):
+    """Instance role per rank."""
+    if n <= 1:
+        return ["DEFAULT"]
+    n_p = max(1, n // 4)
+    return ["PREFILL"] * n_p + ["DECODE"] * (n - n_p)
+
+
+def spawn_worker(args, name, itype, device, registry_port, log_path,
+                 model_name):
+    cmd = [sys.executable, "-m", "xllm_service_amd.engine.worker",
+           "--name", name, "--type", itype, "--model", model_name,
+           "--registry-host", "127.0.0.1",
+           "--registry-port", str(registry_port),
+           "--max-batched-tokens", str(args.max_batched_tokens),
+           "--push-interval-ms", str(args.push_interval_ms),
+           "--seed", "0"]
+    if device:
+        cmd += ["--device", device]
+    if args.max_kv_blocks:
+        cmd += ["--max-kv-blocks", str(args.max_kv_blocks)]
+    logf = open(log_path, "wb")
+    return subprocess.Popen(cmd, cwd=ROOT, stdout=logf,
+                            stderr=subprocess.STDOUT), logf
+
+
+class Completions:
+    """Completion log + exact K-step timing."""
+
+    def __init__(self):
+        self.records = []          # (t_end, ttft_s, out_tokens)
+        self.phase_count = 0
+        self.k = None
+        self.t1 = None
+        self.event = asyncio.Event()
+
+    def arm(self, k):
+        self.phase_count, self.k, self.t1 = 0, k, None
+        self.event = asyncio.Event()
+        if k <= 0:
+            self.t1 = time.monotonic()
+            self.event.set()
+
+    def on_complete(self, ttft, ntok):
+        t = time.monotonic()
+        self.records.append((t, ttft, ntok))
+        if self.k is not None and self.t1 is None:
+            self.phase_count += 1
+            if self.phase_count >= self.k:
+                self.t1 = t
+                self.event.set()
+
+
+async def run_serving_rank0(args, world, dist, use_gpu, model_name,
+                            barrier, my_worker_device):
+    import httpx
+
+    from xllm_service_amd.service import metrics
+    from xllm_service_amd.service.master import Master, MasterOptions
+
+    http_port = free_port()
+    master = Master(MasterOptions(load_balance_policy=args.policy,
+                                  model_id=model_name, host_registry=True,
+                                  registry_port=0, rpc_port=0,
+                                  http_port=http_port))
+    await master.start(serve_http=True)
+    roles = topology(max(world, 1))
+    n_total = len(roles)
+    if world > 1:
+        # ship the registry address to the other ranks (they spawn their
+        # own worker subprocess against it)
+        await asyncio.get_event_loop().run_in_executor(
+            None, dist.broadcast_object_list,
+            [("127.0.0.1", master.opts.registry_port)], 0)
+
+    os.makedirs("gpurun_out", exist_ok=True)
+    proc, logf = spawn_worker(
+        args, f"{roles[0].lower()}-0", roles[0], my_worker_device,
+        master.opts.registry_port, "gpurun_out/bench_worker_r0.log",
+        model_name)
+    try:
+        deadline = time.monotonic() + args.startup_timeout
+        while len(master.instance_mgr.instances) < n_total or \
+                not master.scheduler.has_available_instances():
+            if proc.poll() is not None:
+                raise RuntimeError(
+                    "worker died; tail:\n" + open(
+                        "gpurun_out/bench_worker_r0.log",
+                        "rb").read()[-4000:].decode(errors="replace"))
+            if time.monotonic() > deadline:
+                raise TimeoutError(
+                    f"{len(master.instance_mgr.instances)}/{n_total} "
+                    "workers registered before timeout")
+            await asyncio.sleep(0.5)
+        await asyncio.sleep(1.0)   # let P<->D links settle
+        await barrier()            # B1: cluster up
+
+        client = httpx.AsyncClient(
+            base_url=f"http://127.0.0.1:{http_port}", timeout=600.0,
+            limits=httpx.Limits(max_connections=4096,
+                                max_keepalive_connections=4096))
+        rnd = random.Random(0)
+        comp = Completions()
+        vocab_hi = 120000 if "8b" in model_name else 400
+
+        async def one_request():
+            prompt = [rnd.randrange(10, vocab_hi)
+                      for _ in range(args.input_len)]
+            t0 = time.monotonic()
+            ttft = None
+            try:
+                async with client.stream("POST", "/v1/completions", json={
+                        "model": model_name, "prompt": prompt,
+                        "max_tokens": args.output_len, "temperature": 0.0,
+                        "ignore_eos": True, "stream": True}) as resp:
+                    if resp.status_code != 200:
+                        return
+                    async for line in resp.aiter_lines():
+                        if ttft is None and line.startswith("data: ") \
+                                and "[DONE]" not in line:
+                            ttft = time.monotonic() - t0
+            except (httpx.HTTPError, OSError):
+                return
+            if ttft is not None:
+                comp.on_complete(ttft, args.output_len)
+
+        # ---- phase A: closed-loop ramp + capacity calibration ----------
+        ramp_s = args.ramp_s or (8.0 if use_gpu else 2.0)
+        ramping = True
+        n_decode = sum(1 for r in roles if r != "PREFILL") or 1
+        conc = args.concurrency * n_decode
+
+        async def ramp_client():
+            while ramping:
+                await one_request()
+
+        ramp_tasks = [asyncio.create_task(ramp_client())
+                      for _ in range(conc)]
+        tok_counter = metrics.GENERATED_TOKENS._value
+        base = tok_counter.get()
+        while tok_counter.get() - base < conc:   # first tokens flowing
+            await asyncio.sleep(0.2)
+            if time.monotonic() > deadline:
+                raise TimeoutError("no tokens during ramp")
+        c0, tA = tok_counter.get(), time.monotonic()
+        await asyncio.sleep(ramp_s)
+        cap_tok_s = (tok_counter.get() - c0) / (time.monotonic() - tA)
+        ramping = False                 # in-flight ramp requests drain
+
+        rate = args.arrival_rate or max(
+            args.pace * cap_tok_s / args.output_len, 0.2)
+
+        # ---- phase B: open-loop Poisson ---------------------------------
+        stop_poisson = False
+
+        async def poisson():
+            tasks = []
+            while not stop_poisson:
+                tasks.append(asyncio.create_task(one_request()))
+                await asyncio.sleep(rnd.expovariate(rate))
+            for t in tasks:
+                if not t.done():
+                    t.cancel()
+
+        comp.arm(args.warmup)
+        ptask = asyncio.create_task(poisson())
+        await comp.event.wait()          # W completions under Poisson load
+
+        await barrier()                  # B2
+        if use_gpu:
+            torch.cuda.synchronize()
+        comp.arm(args.steps)
+        t0 = time.monotonic()
+        await comp.event.wait()
+        t1 = comp.t1
+        if use_gpu:
+            torch.cuda.synchronize()
+        await barrier()                  # B3
+
+        stop_poisson = True
+        elapsed = t1 - t0
+        if world > 1:
+            te = torch.tensor([elapsed], dtype=torch.float64)
+            await asyncio.get_event_loop().run_in_executor(
+                None, dist.all_reduce, te, dist.ReduceOp.MAX)
+            elapsed = float(te.item())
+
+        window = [r for r in comp.records if t0 <= r[0] <= t1]
+        slo_s = args.slo_ttft_ms / 1000.0
+        good_tokens = sum(r[2] for r in window if r[1] <= slo_s)
+        all_tokens = sum(r[2] for r in window)
+        ttfts = sorted(r[1] for r in window)
+        p50 = ttfts[len(ttfts) // 2] * 1000 if ttfts else 0.0
+        p99 = ttfts[int(len(ttfts) * 0.99)] * 1000 if ttfts else 0.0
+
+        n_p = sum(1 for r in roles if r == "PREFILL")
+        n_d = len(roles) - n_p
+        par = ("colocated-1gpu" if n_total == 1
+               else f"pd-{n_p}p{n_d}d")
+        print(json.dumps({
+            "metric": "SLO-goodput (out tok/s under p50 TTFT SLO), "
+                      "Llama-3-8B PD-disagg on 8 MI355X",
+            "value": round(good_tokens / elapsed, 2),
+            "unit": "tok/s",
+            "n_gpus": n_total,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "total_tok_per_s": round(all_tokens / elapsed, 2),
+            "p50_ttft_ms": round(p50, 1),
+            "p99_ttft_ms": round(p99, 1),
+            "slo_ttft_ms": args.slo_ttft_ms,
+            "arrival_rate_req_s": round(rate, 2),
+            "calibrated_capacity_tok_s": round(cap_tok_s, 1),
+            "requests_timed": len(window),
+            "config": {
+                "model": model_name,
+                "global_batch": conc,
+                "seq_len": args.input_len + args.output_len,
+                "input_len": args.input_len,
+                "output_len": args.output_len,
+                "parallelism": par,
+                "stack": "http(uvicorn tcp)+sse -> master -> msgrpc -> "
+                         f"{n_total} worker proc(s)",
+                "policy": args.policy,
+                "push_interval_ms": args.push_interval_ms,
+            },
+        }))
+        ptask.cancel()
+        for t in ramp_tasks:
+            t.cancel()
+        await asyncio.gather(ptask, *ramp_tasks, return_exceptions=True)
+        await client.aclose()
+        await barrier()                  # B4: teardown
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+        logf.close()
+        await master.stop()
+
+
+def run_serving_follower(args, rank, dist, use_gpu, model_name):
+    """Ranks > 0: host one worker subprocess on this rank's GPU and follow
+    the barrier protocol (B1 cluster-up, B2/B3 timing, B4 teardown)."""
+    box = [None]
+    dist.broadcast_object_list(box, src=0)
+    host, registry_port = box[0]
+    device = f"cuda:{int(os.environ.get('LOCAL_RANK', rank))}" if use_gpu \
+        else "cpu"
+    roles = topology(int(os.environ["WORLD_SIZE"]))
+    os.makedirs("gpurun_out", exist_ok=True)
+    proc, logf = spawn_worker(
+        args, f"{roles[rank].lower()}-{rank}", roles[rank], device,
+        registry_port, f"gpurun_out/bench_worker_r{rank}.log", model_name)
+    try:
+        dist.barrier()                   # B1
+        dist.barrier()                   # B2
+        if use_gpu:
+            torch.cuda.synchronize()
+        t0 = time.monotonic()
+        dist.barrier()                   # B3
+        t1 = time.monotonic()
+        if use_gpu:
+            torch.cuda.synchronize()
+        te = torch.tensor([t1 - t0], dtype=torch.float64)
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        dist.barrier()                   # B4
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+        logf.close()
 
 
 def main():
@@ -60,109 +471,43 @@ def main():
         import torch.distributed as dist_mod
         dist = dist_mod
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        backend = "nccl" if use_gpu else "gloo"
-        dist.init_process_group(backend=backend)
+        # gloo for the benchmark's own barriers/reductions (host-side);
+        # the workers' data plane (RCCL / xGMI) is their own
+        dist.init_process_group(backend="gloo")
         if use_gpu:
             torch.cuda.set_device(local_rank)
 
-    device = args.device or (f"cuda:{local_rank}" if use_gpu else "cpu")
     model_name = args.model
-    if not use_gpu and model_name == "llama-3-8b":
+    if not use_gpu:
         # CPU fallback so `python bench.py` runs anywhere; the measured
         # config on GPU is the flagship model.
-        model_name = "llama-tiny"
+        if model_name == "llama-3-8b":
+            model_name = "llama-tiny"
+        if args.input_len == 1024:
+            args.input_len = 48
+        if args.output_len == 1024:
+            args.output_len = 16
+        if args.concurrency == 64:
+            args.concurrency = 8
 
-    from xllm_service_amd.engine.engine import LLMEngine
-    from xllm_service_amd.engine.sampling import SamplingParams
-    from xllm_service_amd.models.config import get_config
+    if args.mode == "engine":
+        device = args.device or (f"cuda:{local_rank}" if use_gpu else "cpu")
+        run_engine_mode(args, rank, world, local_rank, use_gpu, dist,
+                        device, model_name)
+    elif rank == 0:
+        my_dev = (args.device or
+                  (f"cuda:{local_rank}" if use_gpu else "cpu"))
 
-    cfg = get_config(model_name)
-    eng = LLMEngine(model_name, device=device,
-                    max_num_seqs=args.concurrency * 2,
-                    max_batched_tokens=args.max_batched_tokens,
-                    max_kv_blocks=None if use_gpu else 4096,
-                    seed=0)
+        async def barrier():
+            if dist:
+                await asyncio.get_event_loop().run_in_executor(
+                    None, dist.barrier)
 
-    torch.manual_seed(1234 + rank)
-    rid = 0
+        asyncio.run(run_serving_rank0(args, world, dist, use_gpu,
+                                      model_name, barrier, my_dev))
+    else:
+        run_serving_follower(args, rank, dist, use_gpu, model_name)
 
-    def feed(n):
-        nonlocal rid
-        for _ in range(n):
-            prompt = torch.randint(0, cfg.vocab_size,
-                                   (args.input_len,)).tolist()
-            eng.add_request(f"r{rank}-{rid}", prompt,
-                            SamplingParams(max_tokens=args.output_len,
-                                           ignore_eos=True))
-            rid += 1
-
-    # keep the pipe full: closed-loop client at fixed concurrency
-    feed(args.concurrency)
-
-    def one_step():
-        outs = eng.step()
-        done = sum(1 for o in outs if o.finished)
-        if done:
-            feed(done)
-        return sum(len(o.new_token_ids) for o in outs)
-
-    # ---- warmup -----------------------------------------------------------
-    for _ in range(args.warmup):
-        one_step()
-
-    # ---- timed region -----------------------------------------------------
-    if dist:
-        dist.barrier()
-    if use_gpu:
-        torch.cuda.synchronize()
-    t0 = time.perf_counter()
-    generated = 0
-    for _ in range(args.steps):
-        generated += one_step()
-    if use_gpu:
-        torch.cuda.synchronize()
-    if dist:
-        dist.barrier()
-    t1 = time.perf_counter()
-
-    elapsed = t1 - t0
-    # max elapsed over ranks + total tokens over ranks
-    if dist:
-        te = torch.tensor([elapsed], dtype=torch.float64,
-                          device=device if use_gpu else "cpu")
-        dist.all_reduce(te, op=dist.ReduceOp.MAX)
-        elapsed = float(te.item())
-        tg = torch.tensor([generated], dtype=torch.float64,
-                          device=device if use_gpu else "cpu")
-        dist.all_reduce(tg, op=dist.ReduceOp.SUM)
-        generated = int(tg.item())
-
-    n_gpus = world if world > 1 else (1 if use_gpu else args.gpus)
-    value = generated / elapsed
-    if rank == 0:
-        print(json.dumps({
-            "metric": "SLO-goodput (out tok/s under p50 TTFT SLO), "
-                      "Llama-3-8B PD-disagg on 8 MI355X",
-            "value": round(value, 2),
-            "unit": "tok/s",
-            "n_gpus": n_gpus,
-            "steps": args.steps,
-            "warmup": args.warmup,
-            "ms_per_step": round(elapsed / args.steps * 1000, 3),
-            "higher_is_better": True,
-            "scaling": "weak",
-            "vs_baseline": None,
-            "dtype": "bf16" if use_gpu else "fp32",
-            "data": "synthetic",
-            "config": {
-                "model": model_name,
-                "global_batch": args.concurrency * n_gpus,
-                "seq_len": args.input_len + args.output_len,
-                "input_len": args.input_len,
-                "output_len": args.output_len,
-                "parallelism": f"dp{n_gpus}",
-            },
-        }))
     if dist:
         dist.destroy_process_group()
 

@@ -107,6 +107,7 @@ class Worker:
                  eos_token_id: Optional[int] = None,
                  max_kv_blocks: Optional[int] = None,
                  relay_responses: bool = False,
+                 push_interval_ms: float = 0.0,
                  engine_kwargs: Optional[Dict[str, Any]] = None):
         self.name = name
         self.itype = InstanceType(itype)
@@ -123,6 +124,12 @@ class Worker:
         # response topology): a DECODE instance routes its generations
         # through the prefill peer that migrated the request in
         self.relay_responses = relay_responses
+        # push coalescing: hold non-urgent token pushes for up to this long
+        # so the master sees one batched Generations per window instead of
+        # one per engine step (the reference pushes batched
+        # DisaggStreamGenerations too). First tokens (TTFT) and finishes
+        # always flush immediately.
+        self.push_interval_ms = push_interval_ms
         self.engine_kwargs = engine_kwargs or {}
         self.incarnation = int(time.time() * 1000)
 
@@ -555,13 +562,29 @@ class Worker:
 
     # ------------------------------------------------------------- push loop
     async def _push_loop(self):
+        """Drain engine outputs and push batched Generations to the master.
+
+        With push_interval_ms > 0, plain decode tokens coalesce per request
+        for up to one interval (one notify carries many tokens per request),
+        which keeps the master's per-token work O(tokens/interval) instead
+        of O(engine steps x batch). First tokens (TTFT), finishes and
+        migrations always flush immediately."""
+        pending: Dict[str, dict] = {}
+        urgent = False
+        interval = self.push_interval_ms / 1000.0
+        last_flush = time.monotonic()
+        get_timeout = 0.1 if interval <= 0 else min(0.1, interval / 2)
         while not self._stop.is_set():
-            outs = await self._loop.run_in_executor(None, self._out_q_get)
-            if outs is None:
-                continue
-            gens = []
+            outs = await self._loop.run_in_executor(
+                None, self._out_q_get, get_timeout)
             migrations = []
-            for o in outs:
+            batches = [outs] if outs else []
+            while True:
+                try:
+                    batches.append(self._out_q.get_nowait())
+                except queue.Empty:
+                    break
+            for o in (o for b in batches for o in b):
                 rid = o.request_id
                 # heartbeat LatencyMetrics: per-request TTFT / inter-token
                 # gaps measured at emission (recent_max_* in the reference)
@@ -572,61 +595,85 @@ class Worker:
                 mig = self.pending_migration.get(rid)
                 if mig is not None and o.finished:
                     # first token produced by prefill: announce + migrate
-                    gens.append(dict(
+                    pending[rid] = dict(
                         service_request_id=rid,
                         token_ids=o.new_token_ids,
                         finished=False, finished_on_prefill=True,
                         prompt_tokens=o.num_prompt_tokens,
-                        completion_tokens=o.num_output_tokens))
+                        completion_tokens=o.num_output_tokens)
                     migrations.append((rid, self.pending_migration.pop(rid),
                                        o.new_token_ids))
+                    urgent = True
                     continue
-                gens.append(dict(
-                    service_request_id=rid,
-                    token_ids=o.new_token_ids,
-                    finished=o.finished,
-                    logprobs=o.logprobs,
-                    finish_reason=o.finish_reason,
-                    finished_on_prefill=(
-                        o.first_token and
-                        self.itype in (InstanceType.DEFAULT, InstanceType.MIX,
-                                       InstanceType.PREFILL)),
-                    prompt_tokens=o.num_prompt_tokens,
-                    completion_tokens=o.num_output_tokens))
+                g = pending.get(rid)
+                if g is None:
+                    pending[rid] = g = dict(
+                        service_request_id=rid, token_ids=[],
+                        finished=False, logprobs=None, finish_reason=None,
+                        finished_on_prefill=False, prompt_tokens=0,
+                        completion_tokens=0)
+                    if self.relay_responses:
+                        g["_via"] = (self.req_meta.get(rid) or
+                                     {}).get("relay_via")
+                g["token_ids"] = g["token_ids"] + o.new_token_ids
+                if o.logprobs:
+                    g["logprobs"] = (g["logprobs"] or []) + o.logprobs
+                g["finished"] = o.finished
+                g["finish_reason"] = o.finish_reason
+                g["finished_on_prefill"] = g["finished_on_prefill"] or (
+                    o.first_token and
+                    self.itype in (InstanceType.DEFAULT, InstanceType.MIX,
+                                   InstanceType.PREFILL))
+                g["prompt_tokens"] = o.num_prompt_tokens
+                g["completion_tokens"] = o.num_output_tokens
                 if o.finished:
                     self.req_meta.pop(rid, None)
-            relayed: Dict[str, list] = {}
-            if self.relay_responses:
-                direct = []
-                for g in gens:
-                    via = (self.req_meta.get(g["service_request_id"]) or
-                           {}).get("relay_via")
-                    if via:
-                        relayed.setdefault(via, []).append(g)
-                    else:
-                        direct.append(g)
-                gens = direct
-            for via, batch in relayed.items():
-                conn = await self._peer_conn(via)
-                try:
-                    if conn is None:
-                        raise RuntimeError(f"relay peer {via} unreachable")
-                    await conn.notify("relay_generations", gens=batch)
-                except Exception:
-                    log.warning("worker %s: relay via %s failed, pushing "
-                                "direct", self.name, via)
-                    gens.extend(batch)
-            if gens and self.master_conn and not self.master_conn.closed.is_set():
-                try:
-                    await self.master_conn.notify("generations", gens=gens)
-                except Exception:
-                    log.warning("worker %s: generations push failed", self.name)
+                    urgent = True
+                if o.first_token:
+                    urgent = True
+            now = time.monotonic()
+            if pending and (urgent or interval <= 0
+                            or now - last_flush >= interval):
+                gens = list(pending.values())
+                pending.clear()
+                urgent = False
+                last_flush = now
+                await self._push_gens(gens)
             for rid, mig, first_toks in migrations:
                 asyncio.create_task(self._do_migration(rid, mig, first_toks))
 
-    def _out_q_get(self):
+    async def _push_gens(self, gens: List[dict]):
+        relayed: Dict[str, list] = {}
+        if self.relay_responses:
+            direct = []
+            for g in gens:
+                via = g.pop("_via", None) or (
+                    self.req_meta.get(g["service_request_id"]) or
+                    {}).get("relay_via")
+                if via:
+                    relayed.setdefault(via, []).append(g)
+                else:
+                    direct.append(g)
+            gens = direct
+        for via, batch in relayed.items():
+            conn = await self._peer_conn(via)
+            try:
+                if conn is None:
+                    raise RuntimeError(f"relay peer {via} unreachable")
+                await conn.notify("relay_generations", gens=batch)
+            except Exception:
+                log.warning("worker %s: relay via %s failed, pushing "
+                            "direct", self.name, via)
+                gens.extend(batch)
+        if gens and self.master_conn and not self.master_conn.closed.is_set():
+            try:
+                await self.master_conn.notify("generations", gens=gens)
+            except Exception:
+                log.warning("worker %s: generations push failed", self.name)
+
+    def _out_q_get(self, timeout: float = 0.1):
         try:
-            return self._out_q.get(timeout=0.1)
+            return self._out_q.get(timeout=timeout)
         except queue.Empty:
             return None
 
@@ -761,6 +808,10 @@ def main():
     ap.add_argument("--no-graphs", action="store_true")
     ap.add_argument("--ssd-swap-dir", default=None,
                     help="directory for the SSD KV swap tier (below DRAM)")
+    ap.add_argument("--push-interval-ms", type=float, default=0.0,
+                    help="coalesce plain token pushes to the master for up "
+                         "to this long (first tokens/finishes always flush "
+                         "immediately); 0 = push every engine step")
     ap.add_argument("--relay-responses", action="store_true",
                     help="DECODE instances route generations through their "
                          "prefill peer (decode->prefill->service topology)")
@@ -794,6 +845,7 @@ def main():
         rpc_host=args.rpc_host, rpc_port=args.rpc_port,
         max_kv_blocks=args.max_kv_blocks,
         relay_responses=args.relay_responses,
+        push_interval_ms=args.push_interval_ms,
         engine_kwargs=dict(seed=args.seed, max_num_seqs=args.max_num_seqs,
                            max_batched_tokens=args.max_batched_tokens,
                            enable_graphs=not args.no_graphs,

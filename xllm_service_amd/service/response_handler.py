@@ -20,6 +20,33 @@ def _sse(obj) -> str:
     return f"data: {json.dumps(obj, ensure_ascii=False)}\n\n"
 
 
+async def _next_delta(queue) -> GenerationDelta:
+    """Await one delta, then greedily coalesce any backlog of plain token
+    deltas into it (one SSE write then carries the merged text). Deltas
+    carrying errors, text overrides or logprobs are never merged across."""
+    gen: GenerationDelta = await asyncio.wait_for(queue.get(),
+                                                  STREAM_TIMEOUT_S)
+    if gen.error or gen.finished or gen.text is not None or gen.logprobs:
+        return gen
+    while True:
+        try:
+            nxt: GenerationDelta = queue.get_nowait()
+        except asyncio.QueueEmpty:
+            break
+        if nxt.error or nxt.text is not None or nxt.logprobs:
+            # cannot merge: re-deliver it right after this one
+            queue._queue.appendleft(nxt)  # asyncio.Queue: deque internally
+            break
+        gen.token_ids = list(gen.token_ids) + list(nxt.token_ids)
+        if nxt.finished:
+            gen.finished = True
+            gen.finish_reason = nxt.finish_reason
+            gen.usage_prompt_tokens = nxt.usage_prompt_tokens
+            gen.usage_completion_tokens = nxt.usage_completion_tokens
+            break
+    return gen
+
+
 class ResponseHandler:
     def __init__(self, tokenizer: Tokenizer, parser_mode: str = "auto"):
         self.tokenizer = tokenizer
@@ -60,8 +87,7 @@ class ResponseHandler:
 
         try:
             while True:
-                gen: GenerationDelta = await asyncio.wait_for(
-                    req.output_queue.get(), STREAM_TIMEOUT_S)
+                gen = await _next_delta(req.output_queue)
                 if gen.error:
                     yield _sse({"error": {"message": gen.error,
                                           "type": "server_error"}})
@@ -176,8 +202,7 @@ class ResponseHandler:
                                      "finish_reason": None}]})
         try:
             while True:
-                gen: GenerationDelta = await asyncio.wait_for(
-                    req.output_queue.get(), STREAM_TIMEOUT_S)
+                gen = await _next_delta(req.output_queue)
                 if gen.error:
                     yield _sse({"error": {"message": gen.error,
                                           "type": "server_error"}})
