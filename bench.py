@@ -201,10 +201,12 @@ def spawn_worker(args, name, itype, device, registry_port, log_path,
 
 
 class Completions:
-    """Completion log + exact K-step timing."""
+    """Completion log + exact K-step timing. Only open-loop ("poisson")
+    requests count as steps — the closed-loop ramp burst has queued-up
+    TTFTs by construction and its stragglers complete in bursts."""
 
     def __init__(self):
-        self.records = []          # (t_end, ttft_s, out_tokens)
+        self.records = []          # (t_end, ttft_s, out_tokens, tag)
         self.phase_count = 0
         self.k = None
         self.t1 = None
@@ -217,10 +219,10 @@ class Completions:
             self.t1 = time.monotonic()
             self.event.set()
 
-    def on_complete(self, ttft, ntok):
+    def on_complete(self, ttft, ntok, tag):
         t = time.monotonic()
-        self.records.append((t, ttft, ntok))
-        if self.k is not None and self.t1 is None:
+        self.records.append((t, ttft, ntok, tag))
+        if self.k is not None and self.t1 is None and tag == "poisson":
             self.phase_count += 1
             if self.phase_count >= self.k:
                 self.t1 = t
@@ -279,7 +281,7 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
         comp = Completions()
         vocab_hi = 120000 if "8b" in model_name else 400
 
-        async def one_request():
+        async def one_request(tag):
             prompt = [rnd.randrange(10, vocab_hi)
                       for _ in range(args.input_len)]
             t0 = time.monotonic()
@@ -298,7 +300,7 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
             except (httpx.HTTPError, OSError):
                 return
             if ttft is not None:
-                comp.on_complete(ttft, args.output_len)
+                comp.on_complete(ttft, args.output_len, tag)
 
         # ---- phase A: closed-loop ramp + capacity calibration ----------
         ramp_s = args.ramp_s or (8.0 if use_gpu else 2.0)
@@ -308,7 +310,7 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
 
         async def ramp_client():
             while ramping:
-                await one_request()
+                await one_request("ramp")
 
         ramp_tasks = [asyncio.create_task(ramp_client())
                       for _ in range(conc)]
@@ -332,7 +334,7 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
         async def poisson():
             tasks = []
             while not stop_poisson:
-                tasks.append(asyncio.create_task(one_request()))
+                tasks.append(asyncio.create_task(one_request("poisson")))
                 await asyncio.sleep(rnd.expovariate(rate))
             for t in tasks:
                 if not t.done():
@@ -361,7 +363,8 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
                 None, dist.all_reduce, te, dist.ReduceOp.MAX)
             elapsed = float(te.item())
 
-        window = [r for r in comp.records if t0 <= r[0] <= t1]
+        window = [r for r in comp.records
+                  if t0 <= r[0] <= t1 and r[3] == "poisson"]
         slo_s = args.slo_ttft_ms / 1000.0
         good_tokens = sum(r[2] for r in window if r[1] <= slo_s)
         all_tokens = sum(r[2] for r in window)
