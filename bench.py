@@ -229,19 +229,37 @@ class Completions:
                 self.event.set()
 
 
+def _parse_prom(text: str) -> dict:
+    vals = {}
+    for line in text.splitlines():
+        if line.startswith("#") or " " not in line:
+            continue
+        name, _, v = line.rpartition(" ")
+        try:
+            vals[name] = float(v)
+        except ValueError:
+            pass
+    return vals
+
+
 async def run_serving_rank0(args, world, dist, use_gpu, model_name,
                             barrier, my_worker_device):
     import httpx
 
-    from xllm_service_amd.service import metrics
-    from xllm_service_amd.service.master import Master, MasterOptions
-
     http_port = free_port()
-    master = Master(MasterOptions(load_balance_policy=args.policy,
-                                  model_id=model_name, host_registry=True,
-                                  registry_port=0, rpc_port=0,
-                                  http_port=http_port))
-    await master.start(serve_http=True)
+    rpc_port = free_port()
+    registry_port = free_port()
+    os.makedirs("gpurun_out", exist_ok=True)
+    # the master runs in its own process (real deployment shape; keeps the
+    # load generator's event loop out of the serving path)
+    mlog = open("gpurun_out/bench_master.log", "wb")
+    mproc = subprocess.Popen(
+        [sys.executable, "-m", "xllm_service_amd.service.master",
+         "--http-host", "127.0.0.1", "--http-port", str(http_port),
+         "--rpc-port", str(rpc_port), "--registry-port", str(registry_port),
+         "--model-id", model_name, "--policy", args.policy],
+        cwd=ROOT, stdout=mlog, stderr=subprocess.STDOUT)
+
     roles = topology(max(world, 1))
     n_total = len(roles)
     if world > 1:
@@ -249,17 +267,42 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
         # own worker subprocess against it)
         await asyncio.get_event_loop().run_in_executor(
             None, dist.broadcast_object_list,
-            [("127.0.0.1", master.opts.registry_port)], 0)
+            [("127.0.0.1", registry_port)], 0)
 
-    os.makedirs("gpurun_out", exist_ok=True)
-    proc, logf = spawn_worker(
-        args, f"{roles[0].lower()}-0", roles[0], my_worker_device,
-        master.opts.registry_port, "gpurun_out/bench_worker_r0.log",
-        model_name)
+    proc = logf = None
+    client = httpx.AsyncClient(
+        base_url=f"http://127.0.0.1:{http_port}", timeout=600.0,
+        limits=httpx.Limits(max_connections=4096,
+                            max_keepalive_connections=4096))
+
+    async def prom() -> dict:
+        try:
+            r = await client.get("/metrics")
+            return _parse_prom(r.text)
+        except (httpx.HTTPError, OSError):
+            return {}
+
     try:
         deadline = time.monotonic() + args.startup_timeout
-        while len(master.instance_mgr.instances) < n_total or \
-                not master.scheduler.has_available_instances():
+        while not await prom():            # master HTTP up
+            if mproc.poll() is not None or time.monotonic() > deadline:
+                raise RuntimeError("master process did not come up; tail:\n"
+                                   + open("gpurun_out/bench_master.log",
+                                          "rb").read()[-4000:].decode(
+                                              errors="replace"))
+            await asyncio.sleep(0.5)
+        proc, logf = spawn_worker(
+            args, f"{roles[0].lower()}-0", roles[0], my_worker_device,
+            registry_port, "gpurun_out/bench_worker_r0.log", model_name)
+
+        def n_instances(m):
+            return sum(v for k, v in m.items()
+                       if k.startswith("cluster_schedulable_instances"))
+
+        while True:
+            m = await prom()
+            if n_instances(m) >= n_total:
+                break
             if proc.poll() is not None:
                 raise RuntimeError(
                     "worker died; tail:\n" + open(
@@ -267,16 +310,12 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
                         "rb").read()[-4000:].decode(errors="replace"))
             if time.monotonic() > deadline:
                 raise TimeoutError(
-                    f"{len(master.instance_mgr.instances)}/{n_total} "
-                    "workers registered before timeout")
+                    f"{n_instances(m)}/{n_total} workers registered "
+                    "before timeout")
             await asyncio.sleep(0.5)
         await asyncio.sleep(1.0)   # let P<->D links settle
         await barrier()            # B1: cluster up
 
-        client = httpx.AsyncClient(
-            base_url=f"http://127.0.0.1:{http_port}", timeout=600.0,
-            limits=httpx.Limits(max_connections=4096,
-                                max_keepalive_connections=4096))
         rnd = random.Random(0)
         comp = Completions()
         vocab_hi = 120000 if "8b" in model_name else 400
@@ -314,19 +353,30 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
 
         ramp_tasks = [asyncio.create_task(ramp_client())
                       for _ in range(conc)]
-        tok_counter = metrics.GENERATED_TOKENS._value
-        base = tok_counter.get()
-        while tok_counter.get() - base < conc:   # first tokens flowing
-            await asyncio.sleep(0.2)
+
+        async def gen_tokens():
+            return (await prom()).get("generated_tokens_total", 0.0)
+
+        base = await gen_tokens()
+        while await gen_tokens() - base < conc:   # first tokens flowing
+            await asyncio.sleep(0.25)
             if time.monotonic() > deadline:
                 raise TimeoutError("no tokens during ramp")
-        c0, tA = tok_counter.get(), time.monotonic()
+        c0, tA = await gen_tokens(), time.monotonic()
         await asyncio.sleep(ramp_s)
-        cap_tok_s = (tok_counter.get() - c0) / (time.monotonic() - tA)
-        ramping = False                 # in-flight ramp requests drain
+        cap_tok_s = (await gen_tokens() - c0) / (time.monotonic() - tA)
+        ramping = False                 # stop issuing ramp requests
 
         rate = args.arrival_rate or max(
             args.pace * cap_tok_s / args.output_len, 0.2)
+
+        # let the closed-loop burst drain completely so the open-loop
+        # phase builds its own steady state (in-flight reaches steady
+        # ~R x request-duration within one request duration)
+        drain_deadline = time.monotonic() + 300.0
+        while (await prom()).get("server_active_requests", 0.0) > 0 and \
+                time.monotonic() < drain_deadline:
+            await asyncio.sleep(0.25)
 
         # ---- phase B: open-loop Poisson ---------------------------------
         stop_poisson = False
@@ -404,8 +454,8 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
                 "input_len": args.input_len,
                 "output_len": args.output_len,
                 "parallelism": par,
-                "stack": "http(uvicorn tcp)+sse -> master -> msgrpc -> "
-                         f"{n_total} worker proc(s)",
+                "stack": "http(uvicorn tcp)+sse -> master proc -> msgrpc "
+                         f"-> {n_total} worker proc(s)",
                 "policy": args.policy,
                 "push_interval_ms": args.push_interval_ms,
             },
@@ -414,16 +464,21 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
         for t in ramp_tasks:
             t.cancel()
         await asyncio.gather(ptask, *ramp_tasks, return_exceptions=True)
-        await client.aclose()
         await barrier()                  # B4: teardown
     finally:
-        proc.terminate()
-        try:
-            proc.wait(timeout=15)
-        except subprocess.TimeoutExpired:
-            proc.kill()
-        logf.close()
-        await master.stop()
+        await client.aclose()
+        for p in (proc, mproc):
+            if p is not None:
+                p.terminate()
+        for p in (proc, mproc):
+            if p is not None:
+                try:
+                    p.wait(timeout=15)
+                except subprocess.TimeoutExpired:
+                    p.kill()
+        if logf:
+            logf.close()
+        mlog.close()
 
 
 def run_serving_follower(args, rank, dist, use_gpu, model_name):

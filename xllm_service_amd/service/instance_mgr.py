@@ -228,6 +228,15 @@ class InstanceMgr:
             log.info("registered instance %s type=%s inc=%d", meta.name,
                      meta.itype, meta.incarnation_id)
 
+    def _update_index_gauges(self):
+        from . import metrics
+        metrics.AVAILABLE_INSTANCES.labels(side="prefill").set(
+            len(self.prefill_index))
+        metrics.AVAILABLE_INSTANCES.labels(side="decode").set(
+            len(self.decode_index))
+        metrics.AVAILABLE_INSTANCES.labels(side="encode").set(
+            len(self.encode_index))
+
     def _add_to_index(self, inst: Instance):
         it = inst.itype
         if it in (InstanceType.DEFAULT, InstanceType.PREFILL):
@@ -245,11 +254,13 @@ class InstanceMgr:
                 self.prefill_index.append(inst.name)
             else:
                 self.decode_index.append(inst.name)
+        self._update_index_gauges()
 
     def _remove_from_index(self, name: str):
         for idx in (self.prefill_index, self.decode_index, self.encode_index):
             if name in idx:
                 idx.remove(name)
+        self._update_index_gauges()
 
     async def _deregister(self, name: str, incarnation: int, reason: str):
         async with self._lock:
