@@ -464,6 +464,17 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
         for t in ramp_tasks:
             t.cancel()
         await asyncio.gather(ptask, *ramp_tasks, return_exceptions=True)
+        # debugging artifacts: per-request completions + final master metrics
+        try:
+            with open("gpurun_out/bench_requests.csv", "w") as f:
+                f.write("t_end,ttft_s,tokens,tag\n")
+                for r in comp.records:
+                    f.write(f"{r[0]:.3f},{r[1]:.3f},{r[2]},{r[3]}\n")
+            mtxt = (await client.get("/metrics")).text
+            with open("gpurun_out/bench_metrics_after.txt", "w") as f:
+                f.write(mtxt)
+        except Exception:
+            pass
         await barrier()                  # B4: teardown
     finally:
         await client.aclose()

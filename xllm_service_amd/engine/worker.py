@@ -773,6 +773,13 @@ class Worker:
                         recent_max_tbt_ms=max(self._tbt_samples or [0.0])),
                     kv_stored=[bytes(h) for h in ev.stored],
                     kv_removed=[bytes(h) for h in ev.removed])
+                if self._ttft_samples and log.isEnabledFor(logging.INFO):
+                    ss = sorted(self._ttft_samples)
+                    log.info(
+                        "engine ttft ms p50=%.0f max=%.0f n=%d "
+                        "waiting=%d running=%d kv=%.2f",
+                        ss[len(ss) // 2], ss[-1], len(ss),
+                        st.num_waiting, st.num_running, st.kv_usage)
                 self._ttft_samples.clear()
                 self._tbt_samples.clear()
                 # ship the first batch of profiling samples by re-PUTting
