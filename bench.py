@@ -48,7 +48,8 @@ if os.path.exists(_TUNABLE):
     os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
     os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", _TUNABLE)
 
-import torch
+if os.environ.get("BENCH_LOADGEN") != "1":
+    import torch
 
 
 def parse_args():
@@ -219,14 +220,97 @@ class Completions:
             self.t1 = time.monotonic()
             self.event.set()
 
-    def on_complete(self, ttft, ntok, tag):
-        t = time.monotonic()
+    def on_complete(self, t, ttft, ntok, tag):
         self.records.append((t, ttft, ntok, tag))
         if self.k is not None and self.t1 is None and tag == "poisson":
             self.phase_count += 1
             if self.phase_count >= self.k:
                 self.t1 = t
                 self.event.set()
+
+
+
+# --------------------------------------------------------------------------
+# load-generator child process (no torch; driven over stdin/stdout).
+# A single asyncio loop cannot parse tens of thousands of SSE chunks per
+# second AND issue new requests on time — measured client-side TTFT then
+# includes loadgen queueing, not serving latency. So the load is spread
+# over M light processes; completions stream back as DONE lines
+# (CLOCK_MONOTONIC is system-wide on Linux, so child timestamps are
+# directly comparable in the parent).
+# --------------------------------------------------------------------------
+def loadgen_child_main():
+    import httpx
+    cfg = json.loads(os.environ["BENCH_LOADGEN_CFG"])
+
+    async def run():
+        client = httpx.AsyncClient(
+            base_url=f"http://127.0.0.1:{cfg['http_port']}", timeout=600.0,
+            limits=httpx.Limits(max_connections=2048,
+                                max_keepalive_connections=2048))
+        rnd = random.Random(cfg["seed"])
+        ramping = {"on": False}
+        quit_ev = asyncio.Event()
+        tasks = []
+
+        async def one_request(tag):
+            prompt = [rnd.randrange(10, cfg["vocab_hi"])
+                      for _ in range(cfg["input_len"])]
+            t0 = time.monotonic()
+            ttft = None
+            try:
+                async with client.stream("POST", "/v1/completions", json={
+                        "model": cfg["model"], "prompt": prompt,
+                        "max_tokens": cfg["output_len"],
+                        "temperature": 0.0, "ignore_eos": True,
+                        "stream": True}) as resp:
+                    if resp.status_code != 200:
+                        return
+                    async for line in resp.aiter_lines():
+                        if ttft is None and line.startswith("data: ") \
+                                and "[DONE]" not in line:
+                            ttft = time.monotonic() - t0
+            except (httpx.HTTPError, OSError):
+                return
+            if ttft is not None:
+                print(f"DONE {time.monotonic():.4f} {ttft:.4f} "
+                      f"{cfg['output_len']} {tag}", flush=True)
+
+        async def ramp_client():
+            while ramping["on"]:
+                await one_request("ramp")
+
+        async def poisson(rate):
+            while not quit_ev.is_set():
+                tasks.append(asyncio.create_task(one_request("poisson")))
+                await asyncio.sleep(rnd.expovariate(rate))
+
+        loop = asyncio.get_event_loop()
+        while True:
+            line = await loop.run_in_executor(None, sys.stdin.readline)
+            if not line:
+                break
+            parts = line.split()
+            if not parts:
+                continue
+            if parts[0] == "ramp":
+                ramping["on"] = True
+                for _ in range(int(parts[1])):
+                    tasks.append(asyncio.create_task(ramp_client()))
+            elif parts[0] == "stopramp":
+                ramping["on"] = False
+            elif parts[0] == "poisson":
+                tasks.append(asyncio.create_task(poisson(float(parts[1]))))
+            elif parts[0] == "quit":
+                break
+        quit_ev.set()
+        ramping["on"] = False
+        for t in tasks:
+            t.cancel()
+        await asyncio.gather(*tasks, return_exceptions=True)
+        await client.aclose()
+
+    asyncio.run(run())
 
 
 def _parse_prom(text: str) -> dict:
@@ -316,43 +400,55 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
         await asyncio.sleep(1.0)   # let P<->D links settle
         await barrier()            # B1: cluster up
 
-        rnd = random.Random(0)
         comp = Completions()
         vocab_hi = 120000 if "8b" in model_name else 400
-
-        async def one_request(tag):
-            prompt = [rnd.randrange(10, vocab_hi)
-                      for _ in range(args.input_len)]
-            t0 = time.monotonic()
-            ttft = None
-            try:
-                async with client.stream("POST", "/v1/completions", json={
-                        "model": model_name, "prompt": prompt,
-                        "max_tokens": args.output_len, "temperature": 0.0,
-                        "ignore_eos": True, "stream": True}) as resp:
-                    if resp.status_code != 200:
-                        return
-                    async for line in resp.aiter_lines():
-                        if ttft is None and line.startswith("data: ") \
-                                and "[DONE]" not in line:
-                            ttft = time.monotonic() - t0
-            except (httpx.HTTPError, OSError):
-                return
-            if ttft is not None:
-                comp.on_complete(ttft, args.output_len, tag)
-
-        # ---- phase A: closed-loop ramp + capacity calibration ----------
-        ramp_s = args.ramp_s or (8.0 if use_gpu else 2.0)
-        ramping = True
         n_decode = sum(1 for r in roles if r != "PREFILL") or 1
         conc = args.concurrency * n_decode
 
-        async def ramp_client():
-            while ramping:
-                await one_request("ramp")
+        # spawn the load-generator children (see loadgen_child_main)
+        n_lg = max(2, 2 * n_decode)
+        lg_env = dict(os.environ, BENCH_LOADGEN="1")
+        children = []
+        lg_logs = []
+        for i in range(n_lg):
+            lg_env["BENCH_LOADGEN_CFG"] = json.dumps(dict(
+                http_port=http_port, model=model_name,
+                input_len=args.input_len, output_len=args.output_len,
+                vocab_hi=vocab_hi, seed=1000 + i))
+            lf = open(f"gpurun_out/bench_loadgen_{i}.log", "wb")
+            lg_logs.append(lf)
+            children.append(subprocess.Popen(
+                [sys.executable, os.path.abspath(__file__)],
+                cwd=ROOT, env=dict(lg_env), stdin=subprocess.PIPE,
+                stdout=subprocess.PIPE, stderr=lf))
 
-        ramp_tasks = [asyncio.create_task(ramp_client())
-                      for _ in range(conc)]
+        def cmd_all(line_fmt, *per_child):
+            for i, ch in enumerate(children):
+                vals = [pc[i] for pc in per_child]
+                ch.stdin.write((line_fmt.format(*vals) + "\n").encode())
+                ch.stdin.flush()
+
+        async def reader(ch):
+            loop = asyncio.get_event_loop()
+            while True:
+                line = await loop.run_in_executor(None, ch.stdout.readline)
+                if not line:
+                    return
+                if line.startswith(b"DONE"):
+                    _, t, ttft, ntok, tag = line.split()
+                    comp.on_complete(float(t), float(ttft), int(ntok),
+                                     tag.decode())
+
+        from concurrent.futures import ThreadPoolExecutor
+        asyncio.get_event_loop().set_default_executor(
+            ThreadPoolExecutor(max_workers=n_lg + 8))
+        readers = [asyncio.create_task(reader(ch)) for ch in children]
+
+        # ---- phase A: closed-loop ramp + capacity calibration ----------
+        ramp_s = args.ramp_s or (8.0 if use_gpu else 2.0)
+        shares = [conc // n_lg + (1 if i < conc % n_lg else 0)
+                  for i in range(n_lg)]
+        cmd_all("ramp {}", shares)
 
         async def gen_tokens():
             return (await prom()).get("generated_tokens_total", 0.0)
@@ -365,7 +461,7 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
         c0, tA = await gen_tokens(), time.monotonic()
         await asyncio.sleep(ramp_s)
         cap_tok_s = (await gen_tokens() - c0) / (time.monotonic() - tA)
-        ramping = False                 # stop issuing ramp requests
+        cmd_all("stopramp", )
 
         rate = args.arrival_rate or max(
             args.pace * cap_tok_s / args.output_len, 0.2)
@@ -379,19 +475,8 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
             await asyncio.sleep(0.25)
 
         # ---- phase B: open-loop Poisson ---------------------------------
-        stop_poisson = False
-
-        async def poisson():
-            tasks = []
-            while not stop_poisson:
-                tasks.append(asyncio.create_task(one_request("poisson")))
-                await asyncio.sleep(rnd.expovariate(rate))
-            for t in tasks:
-                if not t.done():
-                    t.cancel()
-
         comp.arm(args.warmup)
-        ptask = asyncio.create_task(poisson())
+        cmd_all("poisson {}", [rate / n_lg] * n_lg)
         await comp.event.wait()          # W completions under Poisson load
 
         await barrier()                  # B2
@@ -405,7 +490,6 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
             torch.cuda.synchronize()
         await barrier()                  # B3
 
-        stop_poisson = True
         elapsed = t1 - t0
         if world > 1:
             te = torch.tensor([elapsed], dtype=torch.float64)
@@ -460,10 +544,7 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
                 "push_interval_ms": args.push_interval_ms,
             },
         }))
-        ptask.cancel()
-        for t in ramp_tasks:
-            t.cancel()
-        await asyncio.gather(ptask, *ramp_tasks, return_exceptions=True)
+        cmd_all("quit", )
         # debugging artifacts: per-request completions + final master metrics
         try:
             with open("gpurun_out/bench_requests.csv", "w") as f:
@@ -478,15 +559,18 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
         await barrier()                  # B4: teardown
     finally:
         await client.aclose()
-        for p in (proc, mproc):
+        procs = [proc, mproc] + [c for c in locals().get("children", [])]
+        for p in procs:
             if p is not None:
                 p.terminate()
-        for p in (proc, mproc):
+        for p in procs:
             if p is not None:
                 try:
                     p.wait(timeout=15)
                 except subprocess.TimeoutExpired:
                     p.kill()
+        for lf in locals().get("lg_logs", []):
+            lf.close()
         if logf:
             logf.close()
         mlog.close()
@@ -582,4 +666,7 @@ def main():
 
 
 if __name__ == "__main__":
-    main()
+    if os.environ.get("BENCH_LOADGEN") == "1":
+        loadgen_child_main()
+    else:
+        main()
