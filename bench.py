@@ -220,8 +220,8 @@ class Completions:
             self.t1 = time.monotonic()
             self.event.set()
 
-    def on_complete(self, t, ttft, ntok, tag):
-        self.records.append((t, ttft, ntok, tag))
+    def on_complete(self, t, ttft, ntok, tag, t_hdr=0.0):
+        self.records.append((t, ttft, ntok, tag, t_hdr))
         if self.k is not None and self.t1 is None and tag == "poisson":
             self.phase_count += 1
             if self.phase_count >= self.k:
@@ -258,12 +258,14 @@ def loadgen_child_main():
                       for _ in range(cfg["input_len"])]
             t0 = time.monotonic()
             ttft = None
+            t_hdr = None
             try:
                 async with client.stream("POST", "/v1/completions", json={
                         "model": cfg["model"], "prompt": prompt,
                         "max_tokens": cfg["output_len"],
                         "temperature": 0.0, "ignore_eos": True,
                         "stream": True}) as resp:
+                    t_hdr = time.monotonic() - t0
                     if resp.status_code != 200:
                         return
                     async for line in resp.aiter_lines():
@@ -274,7 +276,7 @@ def loadgen_child_main():
                 return
             if ttft is not None:
                 print(f"DONE {time.monotonic():.4f} {ttft:.4f} "
-                      f"{cfg['output_len']} {tag}", flush=True)
+                      f"{cfg['output_len']} {tag} {t_hdr:.4f}", flush=True)
 
         async def ramp_client():
             while ramping["on"]:
@@ -435,9 +437,11 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
                 if not line:
                     return
                 if line.startswith(b"DONE"):
-                    _, t, ttft, ntok, tag = line.split()
-                    comp.on_complete(float(t), float(ttft), int(ntok),
-                                     tag.decode())
+                    parts = line.split()
+                    comp.on_complete(float(parts[1]), float(parts[2]),
+                                     int(parts[3]), parts[4].decode(),
+                                     float(parts[5]) if len(parts) > 5
+                                     else 0.0)
 
         from concurrent.futures import ThreadPoolExecutor
         asyncio.get_event_loop().set_default_executor(
@@ -529,6 +533,9 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
             "p99_ttft_ms": round(p99, 1),
             "slo_ttft_ms": args.slo_ttft_ms,
             "arrival_rate_req_s": round(rate, 2),
+            "p50_headers_ms": round(sorted(
+                r[4] for r in window)[len(window) // 2] * 1000, 1)
+            if window else 0.0,
             "calibrated_capacity_tok_s": round(cap_tok_s, 1),
             "requests_timed": len(window),
             "config": {
