@@ -55,9 +55,37 @@ class ByteTokenizer(Tokenizer):
             ids = [self.bos_token_id] + ids
         return ids
 
+    _B36 = "0123456789abcdefghijklmnopqrstuvwxyz"
+
     def decode(self, ids, skip_special_tokens: bool = True):
-        data = bytes(i - 2 for i in ids if 2 <= i < 258)
-        return data.decode("utf-8", errors="replace")
+        # ids >= 258 (beyond the byte range) decode to a deterministic
+        # word-like string (" " + base36) so models with large vocabs and
+        # random-init weights still stream realistic text volume — without
+        # this, serving benches emit empty SSE deltas and TTFT is
+        # unmeasurable.
+        parts = []
+        buf = bytearray()
+        for i in ids:
+            if 2 <= i < 258:
+                buf.append(i - 2)
+            elif i >= 258:
+                if buf:
+                    parts.append(buf.decode("utf-8", errors="replace"))
+                    buf = bytearray()
+                n = i
+                s = ""
+                while n:
+                    s = self._B36[n % 36] + s
+                    n //= 36
+                parts.append(" " + s)
+            elif not skip_special_tokens:
+                if buf:
+                    parts.append(buf.decode("utf-8", errors="replace"))
+                    buf = bytearray()
+                parts.append("<eos>" if i == self.eos_token_id else "<bos>")
+        if buf:
+            parts.append(buf.decode("utf-8", errors="replace"))
+        return "".join(parts)
 
     @property
     def vocab_size(self):
@@ -128,22 +156,27 @@ class TokenizerFactory:
 
 
 class IncrementalDecoder:
-    """Streaming detokenizer emitting stable text deltas."""
+    """Streaming detokenizer emitting stable text deltas.
+
+    Decodes only a sliding tail window (prefix/read offsets) so per-push
+    cost is O(new tokens), not O(all tokens so far) — at 1024-token
+    streams the full-redecode scheme is O(n^2) per request and dominates
+    the master's token fan-out path."""
 
     def __init__(self, tokenizer: Tokenizer):
         self.tk = tokenizer
         self.ids: List[int] = []
-        self.emitted = ""
+        self.prefix = 0    # window start (safe decode boundary)
+        self.read = 0      # ids already represented in emitted text
 
     def push(self, new_ids: List[int]) -> str:
         self.ids.extend(new_ids)
-        text = self.tk.decode(self.ids)
-        # hold back a trailing replacement char (partial UTF-8 sequence)
-        if text.endswith("�"):
-            text = text[:-1]
-        if not text.startswith(self.emitted):
-            delta = text  # decoder changed its mind; re-emit
-        else:
-            delta = text[len(self.emitted):]
-        self.emitted = self.emitted + delta if delta else self.emitted
+        prefix_text = self.tk.decode(self.ids[self.prefix:self.read])
+        new_text = self.tk.decode(self.ids[self.prefix:])
+        if new_text.endswith("�"):
+            # partial UTF-8 sequence: hold everything new until it resolves
+            return ""
+        delta = new_text[len(prefix_text):]
+        self.prefix = self.read
+        self.read = len(self.ids)
         return delta
