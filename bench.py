@@ -71,7 +71,7 @@ def parse_args():
     ap.add_argument("--device", default=None)
     ap.add_argument("--policy", default="CAR")
     ap.add_argument("--slo-ttft-ms", type=float, default=1000.0)
-    ap.add_argument("--pace", type=float, default=0.90,
+    ap.add_argument("--pace", type=float, default=0.95,
                     help="Poisson arrival rate as a fraction of the "
                          "ramp-measured capacity")
     ap.add_argument("--arrival-rate", type=float, default=0.0,
