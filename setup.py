@@ -33,6 +33,7 @@ ext_modules = [
                 "sampling.hip",
                 "gemm.hip",
                 "skinny_gemm.hip",
+                "packed_gemm.hip",
             )
         ],
         extra_compile_args={

@@ -239,6 +239,22 @@ def test_skinny_gemm(dev, M, N, K):
                   label=f"skinny {M}x{N}x{K}")
 
 
+@pytest.mark.parametrize("M,N,K", [(1, 6144, 4096), (64, 6144, 4096),
+                                   (64, 4096, 4096), (64, 28672, 4096),
+                                   (64, 4096, 14336), (33, 512, 1536),
+                                   (128, 1024, 4160)])
+def test_packed_gemm(dev, M, N, K):
+    torch.manual_seed(M + 1)
+    a = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+    b = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
+    bias = torch.randn(N, device=dev, dtype=torch.bfloat16)
+    wp = ops.pack_gemm_weight(b)
+    got = ops.packed_gemm(a, wp, N, bias)
+    want = torch.nn.functional.linear(a.float(), b.float(), bias.float())
+    _assert_close(got, want, atol=0.05 + 0.02 * (K / 1024),
+                  label=f"packed {M}x{N}x{K}")
+
+
 def test_greedy_sample(dev):
     torch.manual_seed(0)
     logits = torch.randn(33, 128256, device=dev, dtype=torch.bfloat16)

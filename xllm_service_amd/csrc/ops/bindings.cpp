@@ -52,6 +52,10 @@ int skinny_gemm_splitk(int M, int N, int K);
 void launch_skinny_gemm(unsigned short*, const unsigned short*,
                         const unsigned short*, const unsigned short*, float*,
                         int, int, int, hipStream_t);
+int packed_gemm_splitk(int M, int N, int K);
+void launch_packed_gemm(unsigned short*, const unsigned short*,
+                        const unsigned short*, const unsigned short*, float*,
+                        int, int, int, hipStream_t);
 void launch_mfma_probe(float*, const unsigned short*, const unsigned short*,
                        hipStream_t);
 }  // namespace xllm
@@ -303,6 +307,28 @@ torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor b,
   return c;
 }
 
+torch::Tensor packed_gemm(torch::Tensor a, torch::Tensor wp, int64_t N,
+                          c10::optional<torch::Tensor> bias) {
+  // a: [M, K]; wp: W[N,K] pre-packed by ops.pack_gemm_weight; returns [M, N]
+  CHECK_BF16_CUDA(a); CHECK_BF16_CUDA(wp);
+  const int M = a.size(0), K = a.size(1);
+  TORCH_CHECK(wp.numel() == (long)N * K, "packed weight size mismatch");
+  TORCH_CHECK(K % 64 == 0 && N % 64 == 0 && M <= 128);
+  auto c = torch::empty({M, (long)N}, a.options());
+  const int SK = xllm::packed_gemm_splitk(M, (int)N, K);
+  const long mpad = M <= 16 ? 16 : M <= 32 ? 32 : M <= 64 ? 64 : 128;
+  torch::Tensor ws;
+  if (SK > 1)
+    ws = torch::empty({(long)SK * mpad * N}, a.options().dtype(torch::kFloat));
+  else
+    ws = torch::empty({1}, a.options().dtype(torch::kFloat));
+  const unsigned short* bp = nullptr;
+  if (bias.has_value()) bp = u16c(bias.value());
+  xllm::launch_packed_gemm(u16(c), u16c(a), u16c(wp), bp,
+                           ws.data_ptr<float>(), M, (int)N, K, cur_stream());
+  return c;
+}
+
 torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b) {
   CHECK_BF16_CUDA(a); CHECK_BF16_CUDA(b);
   TORCH_CHECK(a.size(0) == 16 && a.size(1) == 32);
@@ -475,6 +501,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
   m.def("mfma_gemm", &mfma_gemm);
   m.def("skinny_gemm", &skinny_gemm);
+  m.def("packed_gemm", &packed_gemm);
   m.def("migrate_blocks_peer", &migrate_blocks_peer);
   m.def("ipc_get_handle", &ipc_get_handle);
   m.def("ipc_open_handle", &ipc_open_handle);

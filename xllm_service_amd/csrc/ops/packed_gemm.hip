@@ -1,0 +1,181 @@
+// Packed-weight streaming GEMM for decode batches on MI355X (gfx950):
+// C[M,N] = A[M,K] @ W[N,K]^T with W PRE-PACKED into per-wave stream order.
+//
+// Why: decode-layer GEMMs are pure weight streaming (W bytes dominate A/C
+// by 50-400x), and the chip's DRAM streams fastest when each wave reads
+// LONG CONTIGUOUS spans. The round-1 skinny kernel (skinny_gemm.hip) loads
+// MFMA b-fragments straight from the row-major weight: 16 parallel row
+// streams each advancing 64 B per 32-k step — measured 1.7-3.2 TB/s cold.
+// The decode-attention kernel, whose waves read ~KB-contiguous K-cache
+// tiles, streams 3.4-4.8 TB/s cold with the same fragment shapes
+// (docs/TODO_ROUND2.md round-1 PMC analysis). Weights are static at
+// inference, so we simply pre-pack W into the exact fragment order a wave
+// consumes:
+//
+//   P[strip n/16][k/32][kgroup 4][row 16][8 k]   (bf16)
+//
+// i.e. per 16-column strip, the K/32 one-KB fragment blocks lie back to
+// back — a wave assigned (strip, k-range) reads ONE contiguous span of
+// 16*K_range bytes with lane l taking bytes [16l, 16l+16) of each block
+// (identical to the b-fragment the MFMA wants, so zero shuffling).
+//
+//   * one wave per (strip, split-k chunk); a workgroup = 4 adjacent strips
+//   * A[M,K] row-major fragments re-read per strip — A is tiny and L2-hot
+//   * 64-k steps double-buffered exactly like skinny_gemm (that scheme is
+//     measured to give full load/MFMA overlap; see its header notes)
+//   * split-K partials reduce with skinny_reduce_kernel; SK==1 writes
+//     bf16 directly
+//
+// Parity: reference engine decode-path linears.
+#include "common.h"
+
+namespace xllm {
+
+typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bf16x8_p;
+typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4_p;
+
+static inline int packed_gemm_mt(int M) {
+  if (M <= 16) return 1;
+  if (M <= 32) return 2;
+  if (M <= 64) return 4;
+  return 8;   // M <= 128
+}
+
+int packed_gemm_splitk(int M, int N, int K) {
+  const int wgs = N / 64;                  // strip groups (4 strips each)
+  int sk = (768 + wgs - 1) / wgs;          // ~3 WGs per CU
+  const int max_sk = K / 256 > 0 ? K / 256 : 1;
+  if (sk > max_sk) sk = max_sk;
+  if (sk > 32) sk = 32;
+  if (sk < 1) sk = 1;
+  return sk;
+}
+
+template <int MT, bool DIRECT>
+__global__ __launch_bounds__(256) void packed_gemm_kernel(
+    float* __restrict__ ws,                // [SK, MT*16, N] fp32 partials
+    unsigned short* __restrict__ c,        // [M, N] bf16 (DIRECT only)
+    const unsigned short* __restrict__ a,  // [M, K] bf16 row-major
+    const unsigned short* __restrict__ w,  // packed P[N/16][K/32][4][16][8]
+    const unsigned short* __restrict__ bias,  // [N] or nullptr (DIRECT)
+    const int M, const int N, const int K, const int SK) {
+  const int group = blockIdx.x / SK;       // 64-column group
+  const int sk = blockIdx.x % SK;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int strip = group * 4 + wid;       // this wave's 16 columns
+  // ceil-divided k-range, 64-aligned so the unrolled step never straddles
+  const int kper = ((K / 64 + SK - 1) / SK) * 64;
+  const int k_begin = sk * kper;
+  const int k_end = min(k_begin + kper, K);
+
+  const int frow = lane & 15;
+  const int fcol8 = (lane >> 4) * 8;
+  const int crow4 = (lane >> 4) * 4;
+
+  f32x4_p acc[MT];
+#pragma unroll
+  for (int m = 0; m < MT; m++) acc[m] = f32x4_p{0, 0, 0, 0};
+
+  // wave-contiguous W stream: strip slab + k offset + this lane's 16 B
+  // (one 512-element block per 32 k). Uniform SGPR base + static voffset.
+  const unsigned short* wbase =
+      w + (long)strip * K * 16 + (long)k_begin * 16 + lane * 8;
+  int aofs[MT];
+#pragma unroll
+  for (int m = 0; m < MT; m++) {
+    const int row = m * 16 + frow;
+    aofs[m] = (row < M ? row : M - 1) * K + fcol8;
+  }
+
+  auto load_w = [&](bf16x8_p (&bb)[2], int krel) {
+    // krel relative to k_begin; two 32-k blocks = 1 KB contiguous each
+    bb[0] = *reinterpret_cast<const bf16x8_p*>(wbase + (long)krel * 16);
+    bb[1] = *reinterpret_cast<const bf16x8_p*>(wbase + (long)(krel + 32) * 16);
+  };
+  auto load_a = [&](bf16x8_p (&aa)[2][MT], int k0) {
+#pragma unroll
+    for (int m = 0; m < MT; m++) {
+      aa[0][m] = *reinterpret_cast<const bf16x8_p*>(a + k0 + aofs[m]);
+      aa[1][m] = *reinterpret_cast<const bf16x8_p*>(a + k0 + 32 + aofs[m]);
+    }
+  };
+  auto mfmas = [&](bf16x8_p (&aa)[2][MT], bf16x8_p (&bb)[2]) {
+#pragma unroll
+    for (int h = 0; h < 2; h++)
+#pragma unroll
+      for (int m = 0; m < MT; m++)
+        acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aa[h][m], bb[h],
+                                                         acc[m], 0, 0, 0);
+  };
+
+  // 64-k steps, double-buffered a full step ahead (same proven shape as
+  // skinny_gemm: explicit buffer names, clamped prefetch, no phase var)
+  const int k_last = k_end - 64;
+  bf16x8_p b_a[2], b_b[2], a_a[2][MT], a_b[2][MT];
+  if (k_begin < k_end) { load_w(b_a, 0); load_a(a_a, k_begin); }
+  for (int k0 = k_begin; k0 < k_end; k0 += 128) {
+    const int kh2 = k0 + 64;
+    const int kh2c = min(kh2, k_last);
+    load_w(b_b, kh2c - k_begin); load_a(a_b, kh2c);
+    mfmas(a_a, b_a);
+    const int kn = min(k0 + 128, k_last);
+    load_w(b_a, kn - k_begin); load_a(a_a, kn);
+    if (kh2 < k_end) mfmas(a_b, b_b);
+  }
+
+  // epilogue: C row = m*16 + crow4 + r, col = strip*16 + frow
+  const int col = strip * 16 + frow;
+  if (col < N) {
+    if (DIRECT) {
+      const float bv = (bias != nullptr) ? bf16_to_f32(bias[col]) : 0.0f;
+#pragma unroll
+      for (int m = 0; m < MT; m++)
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+          const int row = m * 16 + crow4 + r;
+          if (row < M)
+            c[(long)row * N + col] = f32_to_bf16(acc[m][r] + bv);
+        }
+    } else {
+      float* wsk = ws + (long)sk * (MT * 16) * N;
+#pragma unroll
+      for (int m = 0; m < MT; m++)
+#pragma unroll
+        for (int r = 0; r < 4; r++)
+          wsk[(long)(m * 16 + crow4 + r) * N + col] = acc[m][r];
+    }
+  }
+}
+
+void launch_skinny_reduce(unsigned short* c, const float* ws,
+                          const unsigned short* bias, int M, int N, int MPAD,
+                          int SK, hipStream_t stream);
+
+void launch_packed_gemm(unsigned short* c, const unsigned short* a,
+                        const unsigned short* w, const unsigned short* bias,
+                        float* ws, int M, int N, int K, hipStream_t stream) {
+  const int SK = packed_gemm_splitk(M, N, K);
+  const int MT = packed_gemm_mt(M);
+  dim3 grid((N / 64) * SK), block(256);
+#define PG_LAUNCH(MTV)                                                         \
+  do {                                                                         \
+    if (SK == 1) {                                                             \
+      hipLaunchKernelGGL((packed_gemm_kernel<MTV, true>), grid, block, 0,      \
+                         stream, ws, c, a, w, bias, M, N, K, SK);              \
+      return;                                                                  \
+    }                                                                          \
+    hipLaunchKernelGGL((packed_gemm_kernel<MTV, false>), grid, block, 0,       \
+                       stream, ws, c, a, w, bias, M, N, K, SK);                \
+  } while (0)
+  switch (MT) {
+    case 1: PG_LAUNCH(1); break;
+    case 2: PG_LAUNCH(2); break;
+    case 4: PG_LAUNCH(4); break;
+    default: PG_LAUNCH(8); break;
+  }
+#undef PG_LAUNCH
+  launch_skinny_reduce(c, ws, bias, M, N, MT * 16, SK, stream);
+}
+
+}  // namespace xllm

@@ -198,6 +198,16 @@ __global__ void skinny_reduce_kernel(
   }
 }
 
+void launch_skinny_reduce(unsigned short* c, const float* ws,
+                          const unsigned short* bias, int M, int N, int MPAD,
+                          int SK, hipStream_t stream) {
+  long total4 = (long)M * N / 4;
+  long rg = (total4 + 255) / 256;
+  if (rg > 1024) rg = 1024;
+  hipLaunchKernelGGL(skinny_reduce_kernel, dim3((unsigned)rg), dim3(256), 0,
+                     stream, c, ws, bias, M, N, MPAD, SK);
+}
+
 void launch_skinny_gemm(unsigned short* c, const unsigned short* a,
                         const unsigned short* w, const unsigned short* bias,
                         float* ws, int M, int N, int K, hipStream_t stream) {

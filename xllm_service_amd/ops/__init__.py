@@ -214,6 +214,24 @@ def skinny_gemm(a, b, bias=None):
     return _ops.skinny_gemm(a, b, bias)
 
 
+def pack_gemm_weight(w: torch.Tensor) -> torch.Tensor:
+    """Re-order a [N, K] bf16 weight into the per-wave contiguous stream
+    layout csrc/ops/packed_gemm.hip consumes:
+    P[N/16][K/32][kgroup 4][row 16][8 k]. Pure reshuffle (done once at
+    load time); the original tensor stays for prefill (hipBLASLt)."""
+    N, K = w.shape
+    assert N % 16 == 0 and K % 32 == 0, (N, K)
+    return (w.view(N // 16, 16, K // 32, 4, 8)
+             .permute(0, 2, 3, 1, 4).contiguous().view(N // 16, K * 16))
+
+
+def packed_gemm(a, w_packed, n: int, bias=None):
+    """C[M, n] = a @ W^T with W pre-packed by pack_gemm_weight (decode
+    batches, M <= 128). GPU-only; callers fall back to F.linear."""
+    _require_ext()
+    return _ops.packed_gemm(a, w_packed, n, bias)
+
+
 def mfma_probe_16x16x32(a, b):
     _require_ext()
     return _ops.mfma_probe_16x16x32(a, b)
