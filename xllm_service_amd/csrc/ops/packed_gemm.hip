@@ -109,19 +109,40 @@ __global__ __launch_bounds__(256) void packed_gemm_kernel(
                                                          acc[m], 0, 0, 0);
   };
 
-  // 64-k steps, double-buffered a full step ahead (same proven shape as
-  // skinny_gemm: explicit buffer names, clamped prefetch, no phase var)
+  // 64-k steps with a PF-deep W pipeline (4 KB in flight per wave: the
+  // MFMA tail of one step is ~50 ns while HBM latency is ~1 us, so a
+  // 2-deep buffer leaves waves parked at s_waitcnt — measured 1.1-2.0
+  // TB/s; depth 4 puts enough independent loads in flight). A is L2-hot
+  // (every strip re-reads it) and keeps a 2-deep buffer. All buffer
+  // indices are compile-time (unrolled stage loop): phase variables make
+  // the allocator insert vmcnt drains (see skinny_gemm.hip notes).
+  constexpr int PF = 4;
   const int k_last = k_end - 64;
-  bf16x8_p b_a[2], b_b[2], a_a[2][MT], a_b[2][MT];
-  if (k_begin < k_end) { load_w(b_a, 0); load_a(a_a, k_begin); }
-  for (int k0 = k_begin; k0 < k_end; k0 += 128) {
-    const int kh2 = k0 + 64;
-    const int kh2c = min(kh2, k_last);
-    load_w(b_b, kh2c - k_begin); load_a(a_b, kh2c);
-    mfmas(a_a, b_a);
-    const int kn = min(k0 + 128, k_last);
-    load_w(b_a, kn - k_begin); load_a(a_a, kn);
-    if (kh2 < k_end) mfmas(a_b, b_b);
+  auto clampk = [&](int k) { return k < k_last ? k : k_last; };
+  bf16x8_p wq[PF][2], aq[2][2][MT];
+  if (k_begin < k_end) {
+#pragma unroll
+    for (int s = 0; s < PF; s++)
+      load_w(wq[s], clampk(k_begin + s * 64) - k_begin);
+    load_a(aq[0], k_begin);
+    load_a(aq[1], clampk(k_begin + 64));
+  }
+  for (int k0 = k_begin; k0 < k_end; k0 += PF * 64) {
+#pragma unroll
+    for (int s = 0; s < PF; s++) {
+      const int kcur = k0 + s * 64;
+      if (kcur < k_end) {          // uniform (SGPR) branch, no divergence
+        if (s & 1) {
+          mfmas(aq[1], wq[s]);
+          load_w(wq[s], clampk(kcur + PF * 64) - k_begin);
+          load_a(aq[1], clampk(kcur + 128));
+        } else {
+          mfmas(aq[0], wq[s]);
+          load_w(wq[s], clampk(kcur + PF * 64) - k_begin);
+          load_a(aq[0], clampk(kcur + 128));
+        }
+      }
+    }
   }
 
   // epilogue: C row = m*16 + crow4 + r, col = strip*16 + frow
