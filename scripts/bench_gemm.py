@@ -63,7 +63,21 @@ def main():
         gb = N * K * 2 / 1e9
 
         t_lib = timed(lambda i: torch.nn.functional.linear(a, ws[i % copies]))
-        t_pk = timed(lambda i: ops.packed_gemm(a, wps[i % copies], N))
+        t_s = {}
+        for s_try in (1, 2, 4):
+            if N % (64 * s_try):
+                continue
+            try:
+                t_s[s_try] = timed(
+                    lambda i: ops.packed_gemm(a, wps[i % copies], N,
+                                              s_override=s_try))
+            except Exception:
+                pass
+        s_best = min(t_s, key=t_s.get)
+        t_pk = t_s[s_best]
+        sweep = "/".join(f"S{k}:{v:.0f}" for k, v in t_s.items())
+        from xllm_service_amd.ops import _ops
+        t_pr = timed(lambda i: _ops.packed_gemm_probe(a, wps[i % copies], N))
         try:
             t_sk = timed(lambda i: ops.skinny_gemm(a, ws[i % copies])) \
                 if M <= 64 else float("nan")
@@ -76,7 +90,8 @@ def main():
         err = (got - want).abs().max().item()
         print(f"{name:8s} N={N:6d} K={K:6d} ({gb*1000:6.1f} MB): "
               f"lib {t_lib:7.1f}us ({gb/(t_lib/1e6):5.2f} TB/s) | "
-              f"packed {t_pk:7.1f}us ({gb/(t_pk/1e6):5.2f} TB/s) | "
+              f"packed[{sweep}] {t_pk:7.1f}us ({gb/(t_pk/1e6):5.2f} TB/s) | "
+              f"noA {t_pr:7.1f}us ({gb/(t_pr/1e6):5.2f} TB/s) | "
               f"skinny {t_sk:7.1f}us | maxerr {err:.3f}")
         del ws, wps
         torch.cuda.empty_cache()

@@ -52,10 +52,15 @@ int skinny_gemm_splitk(int M, int N, int K);
 void launch_skinny_gemm(unsigned short*, const unsigned short*,
                         const unsigned short*, const unsigned short*, float*,
                         int, int, int, hipStream_t);
-int packed_gemm_splitk(int M, int N, int K);
+int packed_gemm_splitk(int M, int N, int K, int S);
 void launch_packed_gemm(unsigned short*, const unsigned short*,
-                        const unsigned short*, const unsigned short*, float*,
-                        int, int, int, hipStream_t);
+                        unsigned short*, const unsigned short*,
+                        const unsigned short*, float*, int, int, int, int,
+                        hipStream_t);
+int packed_gemm_pick_s(int N);
+void launch_packed_gemm_probe(unsigned short*, const unsigned short*,
+                              const unsigned short*, float*, int, int, int,
+                              hipStream_t);
 void launch_mfma_probe(float*, const unsigned short*, const unsigned short*,
                        hipStream_t);
 }  // namespace xllm
@@ -308,14 +313,18 @@ torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor b,
 }
 
 torch::Tensor packed_gemm(torch::Tensor a, torch::Tensor wp, int64_t N,
-                          c10::optional<torch::Tensor> bias) {
+                          c10::optional<torch::Tensor> bias,
+                          int64_t s_override) {
   // a: [M, K]; wp: W[N,K] pre-packed by ops.pack_gemm_weight; returns [M, N]
   CHECK_BF16_CUDA(a); CHECK_BF16_CUDA(wp);
   const int M = a.size(0), K = a.size(1);
   TORCH_CHECK(wp.numel() == (long)N * K, "packed weight size mismatch");
-  TORCH_CHECK(K % 64 == 0 && N % 64 == 0 && M <= 128);
+  TORCH_CHECK(K % 256 == 0 && N % 64 == 0 && M <= 128);
   auto c = torch::empty({M, (long)N}, a.options());
-  const int SK = xllm::packed_gemm_splitk(M, (int)N, K);
+  const int S = s_override > 0 ? (int)s_override
+                               : xllm::packed_gemm_pick_s((int)N);
+  TORCH_CHECK(N % (64 * S) == 0, "N not divisible for S");
+  const int SK = xllm::packed_gemm_splitk(M, (int)N, K, S);
   const long mpad = M <= 16 ? 16 : M <= 32 ? 32 : M <= 64 ? 64 : 128;
   torch::Tensor ws;
   if (SK > 1)
@@ -324,8 +333,29 @@ torch::Tensor packed_gemm(torch::Tensor a, torch::Tensor wp, int64_t N,
     ws = torch::empty({1}, a.options().dtype(torch::kFloat));
   const unsigned short* bp = nullptr;
   if (bias.has_value()) bp = u16c(bias.value());
-  xllm::launch_packed_gemm(u16(c), u16c(a), u16c(wp), bp,
-                           ws.data_ptr<float>(), M, (int)N, K, cur_stream());
+  const long mt = mpad / 16;
+  auto ap = torch::empty({(long)(K / 32) * mt * 512}, a.options());
+  xllm::launch_packed_gemm(u16(c), u16c(a), u16(ap), u16c(wp), bp,
+                           ws.data_ptr<float>(), M, (int)N, K, S,
+                           cur_stream());
+  return c;
+}
+
+torch::Tensor packed_gemm_probe(torch::Tensor a, torch::Tensor wp,
+                                int64_t N) {
+  // W-stream timing probe: output is garbage (A loads compiled out)
+  CHECK_BF16_CUDA(a); CHECK_BF16_CUDA(wp);
+  const int M = a.size(0), K = a.size(1);
+  auto c = torch::empty({M, (long)N}, a.options());
+  const int SK = xllm::packed_gemm_splitk(M, (int)N, K, 1);
+  torch::Tensor ws;
+  if (SK > 1)
+    ws = torch::empty({(long)SK * 64 * N}, a.options().dtype(torch::kFloat));
+  else
+    ws = torch::empty({1}, a.options().dtype(torch::kFloat));
+  xllm::launch_packed_gemm_probe(u16(c), u16c(a), u16c(wp),
+                                 ws.data_ptr<float>(), M, (int)N, K,
+                                 cur_stream());
   return c;
 }
 
@@ -502,6 +532,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_gemm", &mfma_gemm);
   m.def("skinny_gemm", &skinny_gemm);
   m.def("packed_gemm", &packed_gemm);
+  m.def("packed_gemm_probe", &packed_gemm_probe);
   m.def("migrate_blocks_peer", &migrate_blocks_peer);
   m.def("ipc_get_handle", &ipc_get_handle);
   m.def("ipc_open_handle", &ipc_open_handle);

@@ -225,11 +225,11 @@ def pack_gemm_weight(w: torch.Tensor) -> torch.Tensor:
              .permute(0, 2, 3, 1, 4).contiguous().view(N // 16, K * 16))
 
 
-def packed_gemm(a, w_packed, n: int, bias=None):
+def packed_gemm(a, w_packed, n: int, bias=None, s_override: int = 0):
     """C[M, n] = a @ W^T with W pre-packed by pack_gemm_weight (decode
     batches, M <= 128). GPU-only; callers fall back to F.linear."""
     _require_ext()
-    return _ops.packed_gemm(a, w_packed, n, bias)
+    return _ops.packed_gemm(a, w_packed, n, bias, s_override)
 
 
 def mfma_probe_16x16x32(a, b):
