@@ -209,7 +209,8 @@ class Completions:
     TTFTs by construction and its stragglers complete in bursts."""
 
     def __init__(self):
-        self.records = []          # (t_end, ttft_s, out_tokens, tag)
+        self.records = []          # (t_end, ttft_s, out_tokens, tag, hdr)
+        self.durations = []        # completed request wall time
         self.phase_count = 0
         self.k = None
         self.t1 = None
@@ -222,8 +223,10 @@ class Completions:
             self.t1 = time.monotonic()
             self.event.set()
 
-    def on_complete(self, t, ttft, ntok, tag, t_hdr=0.0):
+    def on_complete(self, t, ttft, ntok, tag, t_hdr=0.0, dur=0.0):
         self.records.append((t, ttft, ntok, tag, t_hdr))
+        if dur > 0:
+            self.durations.append(dur)
         if self.k is not None and self.t1 is None and tag == "poisson":
             self.phase_count += 1
             if self.phase_count >= self.k:
@@ -277,8 +280,10 @@ def loadgen_child_main():
             except (httpx.HTTPError, OSError):
                 return
             if ttft is not None:
-                print(f"DONE {time.monotonic():.4f} {ttft:.4f} "
-                      f"{cfg['output_len']} {tag} {t_hdr:.4f}", flush=True)
+                t1 = time.monotonic()
+                print(f"DONE {t1:.4f} {ttft:.4f} "
+                      f"{cfg['output_len']} {tag} {t_hdr:.4f} "
+                      f"{t1 - t0:.4f}", flush=True)
 
         async def ramp_client():
             while ramping["on"]:
@@ -443,6 +448,8 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
                     comp.on_complete(float(parts[1]), float(parts[2]),
                                      int(parts[3]), parts[4].decode(),
                                      float(parts[5]) if len(parts) > 5
+                                     else 0.0,
+                                     float(parts[6]) if len(parts) > 6
                                      else 0.0)
 
         from concurrent.futures import ThreadPoolExecutor
@@ -481,9 +488,17 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
             await asyncio.sleep(0.25)
 
         # ---- phase B: open-loop Poisson ---------------------------------
+        durs = sorted(comp.durations)
+        d_est = durs[len(durs) // 2] if durs else ramp_s
         comp.arm(args.warmup)
+        poisson_t0 = time.monotonic()
         cmd_all("poisson {}", [rate / n_lg] * n_lg)
         await comp.event.wait()          # W completions under Poisson load
+        # in-flight reaches steady ~R x D one request-duration after the
+        # open loop starts; timing earlier under-measures the steady rate
+        settle = poisson_t0 + 1.3 * d_est - time.monotonic()
+        if settle > 0:
+            await asyncio.sleep(settle)
 
         await barrier()                  # B2
         if use_gpu:
