@@ -71,6 +71,7 @@ class LLMEngine:
                  gpu_memory_utilization: float = 0.85,
                  max_kv_blocks: Optional[int] = None,
                  enable_prefix_caching: bool = True,
+                 prefill_hold_ms: float = 0.0,
                  enable_graphs: bool = True,
                  max_model_len: int = 4096,
                  swap_space_mb: int = 1024,
@@ -140,6 +141,7 @@ class LLMEngine:
         self.scheduler = EngineScheduler(
             self.block_manager, max_num_seqs=max_num_seqs,
             max_batched_tokens=max_batched_tokens,
+            prefill_hold_ms=prefill_hold_ms,
             swap_out=self._swap_out if has_swap else None,
             swap_in=self._swap_in if has_swap else None)
         if has_swap:

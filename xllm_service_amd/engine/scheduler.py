@@ -43,11 +43,18 @@ class EngineScheduler:
                  max_num_seqs: int = 256,
                  max_batched_tokens: int = 8192,
                  enable_chunked_prefill: bool = True,
+                 prefill_hold_ms: float = 0.0,
                  swap_out=None, swap_in=None):
         self.bm = block_manager
         self.max_num_seqs = max_num_seqs
         self.max_batched_tokens = max_batched_tokens
         self.enable_chunked_prefill = enable_chunked_prefill
+        # batch scattered arrivals into fewer (eager) prefill steps: hold
+        # new admissions up to this long while decodes keep the hipGraph
+        # fast path (serving arrivals otherwise pay one full eager step
+        # per prompt; TTFT cost is bounded by the hold)
+        self.prefill_hold_ms = prefill_hold_ms
+        self._last_admit = 0.0
         # swap_out(seq) -> cpu_blocks | None; swap_in(seq) -> None
         # (engine wires these to the host-DRAM KV tier; None = recompute)
         self.swap_out = swap_out
@@ -214,6 +221,13 @@ class EngineScheduler:
 
         # 3. admit waiting sequences (leaving the blocks this step's decodes
         # will take in append_slot untouched)
+        if (self.prefill_hold_ms > 0 and self.waiting and plan.decodes
+                and not plan.prefills and len(self.waiting) < 4):
+            import time as _t
+            now = _t.monotonic()
+            if (now - self._last_admit) * 1000.0 < self.prefill_hold_ms:
+                return plan              # hold: decode-only graph step
+            self._last_admit = now
         while self.waiting and budget > 0 and len(self.running) < self.max_num_seqs:
             seq = self.waiting[0]
             first_alloc = not seq.block_table

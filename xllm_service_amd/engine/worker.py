@@ -815,6 +815,10 @@ def main():
     ap.add_argument("--no-graphs", action="store_true")
     ap.add_argument("--ssd-swap-dir", default=None,
                     help="directory for the SSD KV swap tier (below DRAM)")
+    ap.add_argument("--prefill-hold-ms", type=float, default=25.0,
+                    help="batch scattered prompt arrivals into fewer eager "
+                         "prefill steps (decodes keep the graph path); "
+                         "bounded TTFT cost")
     ap.add_argument("--push-interval-ms", type=float, default=0.0,
                     help="coalesce plain token pushes to the master for up "
                          "to this long (first tokens/finishes always flush "
@@ -855,6 +859,7 @@ def main():
         push_interval_ms=args.push_interval_ms,
         engine_kwargs=dict(seed=args.seed, max_num_seqs=args.max_num_seqs,
                            max_batched_tokens=args.max_batched_tokens,
+                           prefill_hold_ms=args.prefill_hold_ms,
                            enable_graphs=not args.no_graphs,
                            ssd_swap_dir=args.ssd_swap_dir,
                            tp_size=args.tp,
