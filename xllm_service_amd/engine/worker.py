@@ -773,13 +773,20 @@ class Worker:
                         recent_max_tbt_ms=max(self._tbt_samples or [0.0])),
                     kv_stored=[bytes(h) for h in ev.stored],
                     kv_removed=[bytes(h) for h in ev.removed])
-                if self._ttft_samples and log.isEnabledFor(logging.INFO):
-                    ss = sorted(self._ttft_samples)
-                    log.info(
-                        "engine ttft ms p50=%.0f max=%.0f n=%d "
-                        "waiting=%d running=%d kv=%.2f",
-                        ss[len(ss) // 2], ss[-1], len(ss),
-                        st.num_waiting, st.num_running, st.kv_usage)
+                if log.isEnabledFor(logging.INFO):
+                    now_hb = time.monotonic()
+                    prev = getattr(self, "_hb_prev", None)
+                    self._hb_prev = (now_hb, st.steps, st.generated_tokens)
+                    ss = sorted(self._ttft_samples) or [0.0]
+                    if prev is not None:
+                        dt = now_hb - prev[0]
+                        log.info(
+                            "engine ttft ms p50=%.0f n=%d waiting=%d "
+                            "running=%d kv=%.2f | %.1f steps/s %.0f tok/s",
+                            ss[len(ss) // 2], len(self._ttft_samples),
+                            st.num_waiting, st.num_running, st.kv_usage,
+                            (st.steps - prev[1]) / dt,
+                            (st.generated_tokens - prev[2]) / dt)
                 self._ttft_samples.clear()
                 self._tbt_samples.clear()
                 # ship the first batch of profiling samples by re-PUTting
