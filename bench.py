@@ -471,34 +471,30 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
             await asyncio.sleep(0.25)
             if time.monotonic() > deadline:
                 raise TimeoutError("no tokens during ramp")
+        # calibrate on the LATE ramp window (the post-prefill-burst phase
+        # overstates steady capacity)
+        await asyncio.sleep(0.6 * ramp_s)
         c0, tA = await gen_tokens(), time.monotonic()
         await asyncio.sleep(ramp_s)
         cap_tok_s = (await gen_tokens() - c0) / (time.monotonic() - tA)
-        cmd_all("stopramp", )
 
         rate = args.arrival_rate or max(
             args.pace * cap_tok_s / args.output_len, 0.2)
 
-        # let the closed-loop burst drain completely so the open-loop
-        # phase builds its own steady state (in-flight reaches steady
-        # ~R x request-duration within one request duration)
-        drain_deadline = time.monotonic() + 300.0
-        while (await prom()).get("server_active_requests", 0.0) > 0 and \
-                time.monotonic() < drain_deadline:
-            await asyncio.sleep(0.25)
-
         # ---- phase B: open-loop Poisson ---------------------------------
-        durs = sorted(comp.durations)
-        d_est = durs[len(durs) // 2] if durs else ramp_s
-        comp.arm(args.warmup)
+        # hand over WITHOUT draining: ramp re-issues stop as Poisson
+        # arrivals start, so in-flight stays near steady (~R x D) instead
+        # of refilling from empty (a from-empty window under-measures the
+        # steady rate for its whole first request-duration)
+        cmd_all("stopramp", )
         poisson_t0 = time.monotonic()
         cmd_all("poisson {}", [rate / n_lg] * n_lg)
-        await comp.event.wait()          # W completions under Poisson load
-        # in-flight reaches steady ~R x D one request-duration after the
-        # open loop starts; timing earlier under-measures the steady rate
-        settle = poisson_t0 + 1.3 * d_est - time.monotonic()
+        d_est = args.output_len * conc / max(cap_tok_s, 1.0)
+        settle = poisson_t0 + 1.2 * d_est - time.monotonic()
         if settle > 0:
-            await asyncio.sleep(settle)
+            await asyncio.sleep(settle)   # ramp residuals gone, B steady
+        comp.arm(args.warmup)
+        await comp.event.wait()          # W completions at steady state
 
         await barrier()                  # B2
         if use_gpu:
