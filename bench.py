@@ -285,7 +285,8 @@ def loadgen_child_main():
                       f"{cfg['output_len']} {tag} {t_hdr:.4f} "
                       f"{t1 - t0:.4f}", flush=True)
 
-        async def ramp_client():
+        async def ramp_client(stagger):
+            await asyncio.sleep(rnd.uniform(0, stagger))
             while ramping["on"]:
                 await one_request("ramp")
 
@@ -304,8 +305,9 @@ def loadgen_child_main():
                 continue
             if parts[0] == "ramp":
                 ramping["on"] = True
+                stag = float(parts[2]) if len(parts) > 2 else 0.0
                 for _ in range(int(parts[1])):
-                    tasks.append(asyncio.create_task(ramp_client()))
+                    tasks.append(asyncio.create_task(ramp_client(stag)))
             elif parts[0] == "stopramp":
                 ramping["on"] = False
             elif parts[0] == "poisson":
@@ -461,19 +463,23 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
         ramp_s = args.ramp_s or (8.0 if use_gpu else 2.0)
         shares = [conc // n_lg + (1 if i < conc % n_lg else 0)
                   for i in range(n_lg)]
-        cmd_all("ramp {}", shares)
+        # stagger ramp starts so the first generation of closed-loop
+        # requests doesn't complete as one synchronized bunch (which
+        # makes the ramp->Poisson handover oscillate for many request
+        # durations)
+        cmd_all("ramp {} {}", shares, [ramp_s] * n_lg)
 
         async def gen_tokens():
             return (await prom()).get("generated_tokens_total", 0.0)
 
-        base = await gen_tokens()
-        while await gen_tokens() - base < conc:   # first tokens flowing
-            await asyncio.sleep(0.25)
+        async def active_now():
+            return (await prom()).get("server_active_requests", 0.0)
+
+        while await active_now() < 0.9 * conc:    # ramp fully started
+            await asyncio.sleep(0.5)
             if time.monotonic() > deadline:
-                raise TimeoutError("no tokens during ramp")
-        # calibrate on the LATE ramp window (the post-prefill-burst phase
-        # overstates steady capacity)
-        await asyncio.sleep(0.6 * ramp_s)
+                raise TimeoutError("ramp never filled")
+        await asyncio.sleep(2.0)
         c0, tA = await gen_tokens(), time.monotonic()
         await asyncio.sleep(ramp_s)
         cap_tok_s = (await gen_tokens() - c0) / (time.monotonic() - tA)
@@ -490,7 +496,7 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
         poisson_t0 = time.monotonic()
         cmd_all("poisson {}", [rate / n_lg] * n_lg)
         d_est = args.output_len * conc / max(cap_tok_s, 1.0)
-        settle = poisson_t0 + 1.2 * d_est - time.monotonic()
+        settle = poisson_t0 + 1.5 * d_est - time.monotonic()
         if settle > 0:
             await asyncio.sleep(settle)   # ramp residuals gone, B steady
         comp.arm(args.warmup)
