@@ -298,3 +298,73 @@ async def test_latency_samples_reach_master():
     finally:
         await w.stop()
         await master.stop()
+
+
+# ---------------------------------------------- migration overlap (PD)
+class _FakeEvent:
+    def __init__(self):
+        self.done = False
+
+    def query(self):
+        return self.done
+
+
+def test_decode_continues_during_migration_pull():
+    """Pending migrated-in requests must not stall the decode loop: steps
+    keep producing tokens while the (fake) copy event is in flight, and
+    the sequence activates only once it fires."""
+    eng = LLMEngine("llama-tiny", device="cpu", max_kv_blocks=128, seed=7,
+                    enable_prefix_caching=False)
+    cfg = get_config("llama-tiny")
+    torch.manual_seed(41)
+    eng.add_request("bg", torch.randint(0, cfg.vocab_size, (24,)).tolist(),
+                    SamplingParams(max_tokens=64, ignore_eos=True))
+
+    # prefill the migrating request's KV on a source engine, ship bytes
+    src = LLMEngine("llama-tiny", device="cpu", max_kv_blocks=64, seed=7,
+                    enable_prefix_caching=False)
+    prompt = torch.randint(0, cfg.vocab_size, (20,)).tolist()
+    src.add_request("mig", prompt, SamplingParams(max_tokens=1,
+                                                  ignore_eos=True),
+                    hold_blocks=True)
+    while src.has_work():
+        first = [o.new_token_ids[0] for o in src.step() if o.new_token_ids]
+    blocks_src = src.held_block_table("mig")
+
+    dst_blocks = eng.alloc_migration_blocks(len(blocks_src))
+    eng.import_block_bytes(dst_blocks, src.export_block_bytes(blocks_src))
+    ev = _FakeEvent()
+    eng.enqueue_migrated_request("mig", prompt, first, dst_blocks,
+                                 SamplingParams(max_tokens=6,
+                                                ignore_eos=True), event=ev)
+    bg_before = len(eng.seqs["bg"].output_token_ids)
+    for _ in range(4):
+        outs = eng.step()
+        assert all(o.request_id == "bg" for o in outs)
+    assert len(eng.seqs["bg"].output_token_ids) >= bg_before + 4
+    assert "mig" not in eng.seqs          # still pending
+
+    ev.done = True
+    for _ in range(3):
+        eng.step()
+    assert "mig" in eng.seqs              # activated after the event
+    # and it decodes to completion
+    while eng.seqs.get("mig") is not None and eng.has_work():
+        eng.step()
+
+
+def test_abort_during_migration_pull_frees_blocks():
+    eng = LLMEngine("llama-tiny", device="cpu", max_kv_blocks=64, seed=7,
+                    enable_prefix_caching=False)
+    free0 = eng.block_manager.num_free
+    blocks = eng.alloc_migration_blocks(4)
+    ev = _FakeEvent()
+    eng.enqueue_migrated_request("m2", [1, 2, 3], [4], blocks,
+                                 SamplingParams(max_tokens=8), event=ev)
+    assert eng.abort_request("m2") is True
+    eng.step()                            # copy still in flight: no free yet
+    assert eng.block_manager.num_free == free0 - 4
+    ev.done = True
+    eng.step()                            # event fired: blocks released
+    assert eng.block_manager.num_free == free0
+    assert "m2" not in eng.seqs

@@ -36,6 +36,25 @@ class StepOutput:
 
 
 @dataclass
+class PendingMigration:
+    """A migrated-in request whose KV blocks are still being pulled on the
+    side stream. The engine polls `event` each step and activates when the
+    copy lands; aborts mark it and the blocks are freed once the in-flight
+    copy finishes (freeing earlier would let the allocator hand the
+    destination blocks to another sequence mid-copy) — SURVEY.md hard
+    parts 2-3."""
+    request_id: str
+    prompt_token_ids: List[int]
+    first_token_ids: List[int]
+    block_ids: List[int]
+    params: "SamplingParams"
+    priority: int = 0
+    mrope_delta: int = 0
+    event: Optional[object] = None      # torch.cuda.Event | truthy .query()
+    aborted: bool = False
+
+
+@dataclass
 class EngineStats:
     num_waiting: int = 0
     num_running: int = 0
@@ -159,6 +178,7 @@ class LLMEngine:
             self.runner.capture_graphs()
         self.seqs: Dict[str, Sequence] = {}
         self.held: Dict[str, Sequence] = {}   # finished, blocks kept (PD)
+        self.pending_migrations: List[PendingMigration] = []
         self.stats = EngineStats()
         self.eos_token_id: Optional[int] = None  # set by tokenizer owner
 
@@ -359,7 +379,53 @@ class LLMEngine:
         self.scheduler.running.append(seq)
         return None
 
+    def enqueue_migrated_request(self, request_id: str,
+                                 prompt_token_ids: List[int],
+                                 first_token_ids: List[int],
+                                 block_ids: List[int],
+                                 params: Optional[SamplingParams] = None,
+                                 priority: int = 0, mrope_delta: int = 0,
+                                 event=None) -> None:
+        """Queue a migrated-in request for activation once its KV pull
+        event fires (None = data already resident). Decode steps continue
+        while the copy is in flight."""
+        self.pending_migrations.append(PendingMigration(
+            request_id=request_id,
+            prompt_token_ids=list(prompt_token_ids),
+            first_token_ids=list(first_token_ids),
+            block_ids=list(block_ids),
+            params=params or SamplingParams(),
+            priority=priority, mrope_delta=mrope_delta, event=event))
+
+    def _poll_pending_migrations(self) -> List[StepOutput]:
+        outs: List[StepOutput] = []
+        if not self.pending_migrations:
+            return outs
+        for pm in list(self.pending_migrations):
+            if pm.event is not None and not pm.event.query():
+                continue                       # copy still in flight
+            self.pending_migrations.remove(pm)
+            if pm.aborted:
+                self.free_blocks(pm.block_ids)
+                continue
+            fin = self.activate_migrated_request(
+                pm.request_id, pm.prompt_token_ids, pm.first_token_ids,
+                pm.block_ids, pm.params, priority=pm.priority,
+                mrope_delta=pm.mrope_delta)
+            if fin is not None:
+                self.free_blocks(pm.block_ids)
+                outs.append(StepOutput(
+                    request_id=pm.request_id, new_token_ids=[],
+                    finished=True, finish_reason=fin,
+                    num_prompt_tokens=len(pm.prompt_token_ids),
+                    num_output_tokens=len(pm.first_token_ids)))
+        return outs
+
     def abort_request(self, request_id: str) -> bool:
+        for pm in self.pending_migrations:
+            if pm.request_id == request_id and not pm.aborted:
+                pm.aborted = True              # blocks freed after the copy
+                return True
         seq = self.scheduler.abort(request_id)
         self.seqs.pop(request_id, None)
         if self.runner.graph_runner is not None:
@@ -367,13 +433,14 @@ class LLMEngine:
         return seq is not None
 
     def has_work(self) -> bool:
-        return self.scheduler.has_work()
+        return self.scheduler.has_work() or bool(self.pending_migrations)
 
     # ---- main loop ----------------------------------------------------------
     def step(self) -> List[StepOutput]:
+        pre = self._poll_pending_migrations()
         plan = self.scheduler.schedule()
         if plan.empty:
-            return []
+            return pre
         self.stats.last_prefill_tokens = sum(sp.chunk_len
                                              for sp in plan.prefills)
         self.stats.last_decodes = len(plan.decodes)
@@ -431,7 +498,7 @@ class LLMEngine:
         self.stats.num_waiting = self.scheduler.num_waiting
         self.stats.num_running = len(self.scheduler.running)
         self.stats.kv_usage = self.block_manager.usage()
-        return outputs
+        return pre + outputs
 
     def export_block_bytes(self, block_ids: List[int]) -> bytes:
         """Serialize KV blocks (all layers, k then v) — the DRAM/RPC

@@ -106,13 +106,15 @@ class MigrationManager:
     def has_peer(self, name: str) -> bool:
         return name in self.peers
 
-    def pull_blocks(self, src_name: str, src_blocks: List[int],
-                    dst_blocks: List[int]):
-        """Copy blocks from the peer cache into ours (all layers) on the side
-        stream; synchronizes before returning so the sequence can be
-        activated immediately after."""
+    def pull_blocks_async(self, src_name: str, src_blocks: List[int],
+                          dst_blocks: List[int]) -> "torch.cuda.Event":
+        """Start the block pull on the dedicated side stream and return the
+        completion event — the decode loop keeps stepping while xGMI copies
+        fly; the sequence activates only when the event fires (engine
+        pending-migration poll). Safe to call from any thread."""
         pc = self.peers[src_name]
         my_dev = self.engine.device.index or 0
+        ev = torch.cuda.Event()
         with torch.cuda.stream(self.stream):
             for layer, (kc, vc) in enumerate(self.engine.runner.kv_caches):
                 sk, sv = pc.views[layer]
@@ -120,4 +122,10 @@ class MigrationManager:
                                         src_blocks, dst_blocks)
                 ops.migrate_blocks_peer(vc, my_dev, sv, pc.src_device,
                                         src_blocks, dst_blocks)
-        self.stream.synchronize()
+            ev.record(self.stream)
+        return ev
+
+    def pull_blocks(self, src_name: str, src_blocks: List[int],
+                    dst_blocks: List[int]):
+        """Synchronous pull (tests / compatibility)."""
+        self.pull_blocks_async(src_name, src_blocks, dst_blocks).synchronize()

@@ -527,37 +527,36 @@ class Worker:
                 await self._run_on_engine(
                     lambda: self.mig.open_peer(src_name, exported))
 
-        def _recv():
-            blocks = self.engine.alloc_migration_blocks(n_blocks)
-            try:
-                if transport == "bytes":
-                    self.engine.import_block_bytes(blocks, data)
-                elif transport == "xgmi":
-                    self.mig.pull_blocks(src_name, src_blocks, blocks)
-                else:
-                    raise ValueError(f"unknown transport {transport}")
-            except Exception:
-                self.engine.free_blocks(blocks)
-                raise
-            fin = self.engine.activate_migrated_request(
+        # allocate destination blocks on the engine thread (block manager
+        # is engine-thread state), then pull OFF the engine thread so
+        # decode steps keep running during the copy (SURVEY hard-part 2)
+        blocks = await self._run_on_engine(
+            lambda: self.engine.alloc_migration_blocks(n_blocks))
+        event = None
+        try:
+            if transport == "bytes":
+                await self._loop.run_in_executor(
+                    None, lambda: self.engine.import_block_bytes(blocks,
+                                                                 data))
+            elif transport == "xgmi":
+                # copies fly on the migration side stream; the engine
+                # activates the sequence when the event fires
+                event = self.mig.pull_blocks_async(src_name, src_blocks,
+                                                   blocks)
+            else:
+                raise ValueError(f"unknown transport {transport}")
+        except Exception:
+            self._post_to_engine(lambda: self.engine.free_blocks(blocks))
+            raise
+        self._post_to_engine(
+            lambda: self.engine.enqueue_migrated_request(
                 service_request_id, prompt_token_ids, first_token_ids,
                 blocks, sp, priority=1 if offline else 0,
-                mrope_delta=mrope_delta)
-            if fin is not None:
-                self.engine.free_blocks(blocks)
-            return fin
-
-        fin = await self._run_on_engine(_recv)
-        if fin is not None:
-            # the prefill token(s) already exhausted the request's budget
-            # (or ended on EOS): report the finish through the normal push
-            # path (handles the relay topology too)
-            from .engine import StepOutput
-            self._out_q.put([StepOutput(
-                service_request_id, new_token_ids=[], finished=True,
-                finish_reason=fin,
-                num_prompt_tokens=len(prompt_token_ids),
-                num_output_tokens=len(first_token_ids))])
+                mrope_delta=mrope_delta, event=event))
+        if event is not None:
+            # don't answer the prefill peer until the copy has landed: it
+            # releases its held source blocks when this RPC returns
+            await self._loop.run_in_executor(None, event.synchronize)
         return True
 
     # ------------------------------------------------------------- push loop
