@@ -1,0 +1,98 @@
+"""Adaptive P:D role flipping, the REAL version (VERDICT round-1 item 7):
+the master's SLO flip must reach the worker — which adopts the new role,
+re-partitions its KV pool (drops the prefix-cache reserve when becoming a
+decode), and refreshes its registration meta — not just move a scheduling
+index. (reference flips scheduling only: instance_mgr.cpp:1023-1063)"""
+import asyncio
+
+import pytest
+
+from tests.test_service_integration import (http_client, make_master,
+                                            wait_for, worker_kwargs)
+from xllm_service_amd.engine.worker import Worker
+from xllm_service_amd.service.types import InstanceType
+
+
+@pytest.fixture
+def anyio_backend():
+    return "asyncio"
+
+
+@pytest.mark.anyio
+async def test_slo_flip_converts_worker_and_capacity():
+    master = make_master(policy="SLO_AWARE")
+    await master.start(serve_http=False)
+    workers = [Worker("p0", "PREFILL", **worker_kwargs(master)),
+               Worker("p1", "PREFILL", **worker_kwargs(master)),
+               Worker("d0", "DECODE", **worker_kwargs(master))]
+    try:
+        for w in workers:
+            await w.start()
+        await wait_for(lambda: len(master.instance_mgr.instances) == 3
+                       and master.scheduler.has_available_instances())
+        client = await http_client(master)
+
+        # run one request so p1 has cached prefix blocks to give up
+        r = await client.post("/v1/completions", json={
+            "model": "llama-tiny", "prompt": list(range(10, 60)),
+            "max_tokens": 4, "temperature": 0.0, "ignore_eos": True})
+        assert r.status_code == 200
+
+        pol = master.scheduler.policy
+        # load shift: decodes are saturated (slow observed TPOT), prefills
+        # idle (fast observed TTFT) -> SLO policy flips a prefill to decode
+        pol.tpot.clear()
+        pol.ttft.clear()
+        for _ in range(8):
+            pol.observe_tpot("d0", 8, 64, 500.0)
+            pol.observe_ttft("p0", 64, 5.0)
+            pol.observe_ttft("p1", 64, 5.0)
+        pair = pol.select_instances_pair(list(range(30)))
+        assert pair.ok
+
+        mgr = master.instance_mgr
+        await wait_for(lambda: "p1" in mgr.decode_index)
+        assert "p1" not in mgr.prefill_index
+
+        # the worker itself converted (not just the index)
+        w_p1 = workers[1]
+        await wait_for(lambda: w_p1.itype == InstanceType.DECODE)
+        # master-side meta reflects the new role
+        await wait_for(
+            lambda: mgr.get("p1").meta.itype == InstanceType.DECODE.value)
+        # prefix-cache reserve was dropped (pool re-partitioned)
+        assert len(w_p1.engine.block_manager.evictable) == 0
+
+        # measured capacity change: the decode side now schedules 2
+        # instances and a request decode-bound to p1 completes end to end
+        decode_names = {i.name for i in mgr.schedulable_decodes()}
+        assert decode_names == {"d0", "p1"}
+        for _ in range(6):
+            r = await client.post("/v1/completions", json={
+                "model": "llama-tiny", "prompt": list(range(40, 80)),
+                "max_tokens": 4, "temperature": 0.0, "ignore_eos": True})
+            assert r.status_code == 200, r.text
+        await client.aclose()
+    finally:
+        for w in workers:
+            await w.stop()
+        await master.stop()
+
+
+@pytest.mark.anyio
+async def test_flip_keeps_last_prefill():
+    """The scheduler must never flip away the last prefill-side instance."""
+    master = make_master(policy="SLO_AWARE")
+    await master.start(serve_http=False)
+    workers = [Worker("p0", "PREFILL", **worker_kwargs(master)),
+               Worker("d0", "DECODE", **worker_kwargs(master))]
+    try:
+        for w in workers:
+            await w.start()
+        await wait_for(lambda: len(master.instance_mgr.instances) == 2)
+        assert master.instance_mgr.flip_instance_role("p0", "decode") is False
+        assert master.instance_mgr.flip_instance_role("d0", "prefill") is False
+    finally:
+        for w in workers:
+            await w.stop()
+        await master.stop()

@@ -166,6 +166,21 @@ class BlockManager:
             self._decref(blk)
         seq.block_table = []
 
+    def reset_prefix_cache(self) -> int:
+        """Drop every evictable cached block (role flip: a decode role
+        wants its pool for long-lived decode KV, not prefill prefix reuse).
+        Returns the number of blocks released to the free list."""
+        n = len(self.evictable)
+        for blk in list(self.evictable):
+            h = self.block_hash[blk]
+            if h is not None:
+                self.hash_to_block.pop(h, None)
+                self.block_hash[blk] = None
+                self.events.removed.add(h)
+        self.free_blocks.extend(self.evictable)
+        self.evictable.clear()
+        return n
+
     # ---- migration support (PD disaggregation) -----------------------------
     def allocate_raw(self, n: int) -> List[int]:
         """Allocate n unhashed blocks (decode side of a KV migration)."""

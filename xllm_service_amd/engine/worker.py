@@ -492,6 +492,34 @@ class Worker:
                 pass
         log.warning("worker %s: relay forward to master failed", self.name)
 
+    async def on_role_change(self, conn, new_type: str):
+        """SLO-aware adaptive P:D flip (master -> worker): adopt the new
+        role for FUTURE requests (in-flight old-role work drains
+        naturally), re-partition the KV pool — a decode role reclaims the
+        prefix-cache reserve for long-lived decode KV — and refresh the
+        registration meta so GetInstanceInfo reflects the new type."""
+        old = self.itype
+        try:
+            self.itype = InstanceType(new_type)
+        except ValueError:
+            return
+        if self.engine is not None and self.itype == InstanceType.DECODE:
+            released = await self._run_on_engine(
+                lambda: self.engine.block_manager.reset_prefix_cache())
+            log.info("worker %s: role %s -> %s, released %d cached blocks",
+                     self.name, old.value, new_type, released)
+        else:
+            log.info("worker %s: role %s -> %s", self.name, old.value,
+                     new_type)
+        # same key + incarnation: the master treats this as a meta refresh
+        if self.registry is not None:
+            try:
+                await self.registry.put_json(self._regkey(),
+                                             self.meta().to_dict(),
+                                             lease_id=self._lease_id)
+            except Exception:
+                pass
+
     def on_abort_request(self, conn, service_request_id: str):
         self.pending_migration.pop(service_request_id, None)
         self._post_to_engine(

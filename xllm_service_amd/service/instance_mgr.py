@@ -434,21 +434,37 @@ class InstanceMgr:
     # ---- role flipping (SLO-aware adaptive P:D) ------------------------------
     def flip_instance_role(self, name: str, new_side: str) -> bool:
         """Move an instance between the prefill and decode scheduling sides
-        (MIX instances only are eligible; memory re-partitioning is handled
-        worker-side on its next idle point)."""
+        AND tell the worker: it re-partitions its KV pool for the new role
+        (drops the prefix-cache reserve when becoming a decode), updates
+        its registration meta, and keeps serving in-flight old-role work
+        until it drains (reference flips scheduling only,
+        instance_mgr.cpp:1023-1063 — SURVEY hard-part 4 asks for more)."""
         inst = self.instances.get(name)
         if inst is None:
             return False
+        flipped = False
         if new_side == "decode" and name in self.prefill_index:
             if len(self.prefill_index) <= 1:
                 return False
             self.prefill_index.remove(name)
             self.decode_index.append(name)
-            return True
-        if new_side == "prefill" and name in self.decode_index:
+            flipped = True
+        elif new_side == "prefill" and name in self.decode_index:
             if len(self.decode_index) <= 1:
                 return False
             self.decode_index.remove(name)
             self.prefill_index.append(name)
-            return True
-        return False
+            flipped = True
+        if flipped:
+            self._update_index_gauges()
+            new_type = (InstanceType.DECODE if new_side == "decode"
+                        else InstanceType.PREFILL)
+            inst.meta.itype = new_type.value
+            if inst.conn is not None:
+                try:
+                    asyncio.get_event_loop().create_task(
+                        inst.conn.notify("role_change",
+                                         new_type=new_type.value))
+                except RuntimeError:
+                    pass   # no running loop (unit tests)
+        return flipped
