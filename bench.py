@@ -80,8 +80,10 @@ def parse_args():
                     help="req/s override (0 = auto-calibrate)")
     ap.add_argument("--ramp-s", type=float, default=0.0,
                     help="capacity-calibration window (0 = auto)")
-    ap.add_argument("--push-interval-ms", type=float, default=25.0,
-                    help="worker->master token push coalescing window")
+    ap.add_argument("--push-interval-ms", type=float, default=0.0,
+                    help="worker->master token push coalescing window "
+                         "(0 = auto: scales with decode GPU count so the "
+                         "master's SSE fan-out stays off the hot path)")
     ap.add_argument("--max-kv-blocks", type=int, default=0)
     ap.add_argument("--startup-timeout", type=float, default=900.0)
     return ap.parse_args()
@@ -668,6 +670,11 @@ def main():
             args.output_len = 16
         if args.concurrency == 128:
             args.concurrency = 8
+
+    if args.push_interval_ms <= 0:
+        r = topology(max(world, 1))
+        nd = sum(1 for x in r if x != "PREFILL") or 1
+        args.push_interval_ms = 25.0 * max(1.0, nd / 2.0)
 
     if args.mode == "engine":
         device = args.device or (f"cuda:{local_rank}" if use_gpu else "cpu")

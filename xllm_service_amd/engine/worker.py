@@ -728,7 +728,20 @@ class Worker:
                     lambda: self.engine.export_block_bytes(blocks))
             else:
                 kwargs["src_blocks"] = blocks
-            await conn.call("migrate_in", timeout=60.0, **kwargs)
+            try:
+                await conn.call("migrate_in", timeout=60.0, **kwargs)
+            except Exception as e:
+                if transport != "xgmi":
+                    raise
+                # xGMI/IPC path failed on this topology: fall back to the
+                # serialized-bytes transport rather than killing the request
+                log.warning("xgmi migration of %s failed (%s); retrying "
+                            "with bytes transport", rid, e)
+                kwargs["transport"] = "bytes"
+                kwargs.pop("src_blocks", None)
+                kwargs["data"] = await self._run_on_engine(
+                    lambda: self.engine.export_block_bytes(blocks))
+                await conn.call("migrate_in", timeout=60.0, **kwargs)
         except Exception as e:
             log.warning("migration of %s to %s failed: %s", rid, decode_name, e)
             # tell the master the request died (client will retry)
