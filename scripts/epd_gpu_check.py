@@ -54,14 +54,18 @@ async def scenario(worker_specs):
     await master.start(serve_http=False)
     procs = []
     try:
+        os.makedirs("gpurun_out", exist_ok=True)
+        logs = []
         for name, itype in worker_specs:
+            lf = open(f"gpurun_out/epd_{name}.log", "wb")
+            logs.append(lf)
             procs.append(subprocess.Popen([
                 sys.executable, "-m", "xllm_service_amd.engine.worker",
                 "--name", name, "--type", itype, "--model", MODEL,
                 "--device", "cuda:0", "--registry-port",
                 str(master.opts.registry_port), "--max-kv-blocks", "256",
                 "--seed", "11", "--no-graphs"],
-                cwd=ROOT, stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+                cwd=ROOT, stdout=lf, stderr=subprocess.STDOUT))
         want_encode = any(t == "ENCODE" for _, t in worker_specs)
         for _ in range(600):
             ready = master.scheduler.has_available_instances() and (
@@ -71,7 +75,6 @@ async def scenario(worker_specs):
             await asyncio.sleep(0.5)
             for p in procs:
                 if p.poll() is not None:
-                    print(p.communicate()[0].decode()[-3000:])
                     raise RuntimeError("worker died during startup")
         else:
             raise TimeoutError("workers never became available")
