@@ -18,7 +18,7 @@ import sys
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, ROOT)
 
-MODEL = "qwen2-vl-tiny"
+MODEL = "qwen2-vl-tiny128"
 
 MESSAGES = [
     [{"role": "user", "content": [

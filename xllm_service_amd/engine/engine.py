@@ -33,6 +33,7 @@ class StepOutput:
     first_token: bool = False
     ttft_ms: Optional[float] = None   # set on the first emitted token
     tbt_ms: Optional[float] = None    # inter-token gap (decode steps)
+    error: Optional[str] = None       # engine-failure abort detail
 
 
 @dataclass

@@ -647,6 +647,8 @@ class Worker:
                     g["logprobs"] = (g["logprobs"] or []) + o.logprobs
                 g["finished"] = o.finished
                 g["finish_reason"] = o.finish_reason
+                if getattr(o, "error", None):
+                    g["error"] = o.error
                 g["finished_on_prefill"] = g["finished_on_prefill"] or (
                     o.first_token and
                     self.itype in (InstanceType.DEFAULT, InstanceType.MIX,

