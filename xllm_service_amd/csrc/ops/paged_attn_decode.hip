@@ -30,6 +30,12 @@ typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4;
 #define PA_D 128           // head_dim handled by the MFMA path
 #define PA_PAD 8           // LDS row padding (elements)
 
+// NOTE round-2: forcing 3 waves/SIMD via __launch_bounds__(256, 3) caps the
+// unified VGPR+AGPR file at 166 and the allocator spills the K prefetch to
+// SCRATCH — measured 3.1 TB/s vs 5.3 at the natural 232V+32A allocation
+// (S=64 L=2048). The occupancy squeeze in docs/TODO_ROUND2.md is a dead
+// end on gfx950's unified file; the kernel already streams 5.1-5.6 TB/s at
+// serving shapes (B>=64, L>=1100).
 template <int G>
 __global__ __launch_bounds__(256) void paged_attn_decode_kernel(
     float* __restrict__ ws_ml,               // [S*n_kv*W][G][2] partial m,l
