@@ -42,7 +42,8 @@ typedef __attribute__((__vector_size__(2 * sizeof(unsigned)))) unsigned u32x2;
 #define PF_PPAD 8          // pad elements per P row
 #define PF_BS 16           // cache block size
 
-__global__ __launch_bounds__(512) void paged_attn_prefill_kernel(
+__global__ __launch_bounds__(512) __attribute__((amdgpu_waves_per_eu(4)))
+void paged_attn_prefill_kernel(
     unsigned short* __restrict__ out,        // [total_q, n_qheads, D] bf16
     const unsigned short* __restrict__ q,    // [total_q, n_qheads, D] bf16
     const unsigned short* __restrict__ k_cache,  // [blocks, n_kv, 16, D]
@@ -78,12 +79,13 @@ __global__ __launch_bounds__(512) void paged_attn_prefill_kernel(
   const int wg_q_end = min(q0 + PF_WGQ, q_len);
   const int wg_kmax = ctx + wg_q_end;
 
-  // K/V tiles are DOUBLE-buffered (85 KB total — only one workgroup is
-  // resident at this register count, so the LDS is free): the stage write
-  // for tile t+1 lands in the other buffer, needing ONE barrier per tile
-  // instead of two (PMC: the 2-barrier shape parked 45% of wave-cycles)
-  __shared__ unsigned short k_lds[2][PF_KBLK][PF_D + PF_KPAD];
-  __shared__ unsigned short v_img[2][2][4096];  // tr16 image per 32-key half
+  // v4: SINGLE-buffered K/V staging (52 KB total) + registers capped at
+  // 128 so TWO 8-wave workgroups co-reside per CU. That costs a second
+  // barrier per tile, but with 2 WGs the other workgroup's MFMAs cover
+  // every barrier/stage stall — the round-1 single-WG profile parked 45%
+  // of wave-cycles at SQ_WAIT with nothing to switch to.
+  __shared__ unsigned short k_lds[PF_KBLK][PF_D + PF_KPAD];
+  __shared__ unsigned short v_img[2][4096];     // tr16 image per 32-key half
   __shared__ unsigned short p_lds[8][PF_QW][PF_KBLK + PF_PPAD];
 
   // ---- Q fragment: one m-tile, lane row frow -------------------------------
@@ -136,7 +138,7 @@ __global__ __launch_bounds__(512) void paged_attn_prefill_kernel(
     }
   };
 
-  auto stage_write = [&](int kv0, int buf) {
+  auto stage_write = [&](int kv0) {
 #pragma unroll
     for (int it = 0; it < 2; it++) {
       const int idx = sidx0 + it * 4096;
@@ -147,21 +149,20 @@ __global__ __launch_bounds__(512) void paged_attn_prefill_kernel(
 #pragma unroll
         for (int j = 0; j < 8; j++) { kvv.x[j] = 0; vvv.x[j] = 0; }
       }
-      *reinterpret_cast<ushort8_t*>(&k_lds[buf][kt][d]) = kvv;
+      *reinterpret_cast<ushort8_t*>(&k_lds[kt][d]) = kvv;
       const int k32 = kt & 31;
       const int koff =
           ((k32 >> 2) & 1) * 256 + (k32 >> 3) * 64 + (k32 & 3) * 16;
       *reinterpret_cast<ushort8_t*>(
-          &v_img[buf][kt >> 5][(d >> 4) * 512 + koff + (d & 15)]) = vvv;
+          &v_img[kt >> 5][(d >> 4) * 512 + koff + (d & 15)]) = vvv;
     }
   };
 
   stage_load(0);
-  stage_write(0, 0);
+  stage_write(0);
   __syncthreads();
 
-  int buf = 0;
-  for (int kv0 = 0; kv0 < wg_kmax; kv0 += PF_KBLK, buf ^= 1) {
+  for (int kv0 = 0; kv0 < wg_kmax; kv0 += PF_KBLK) {
     const bool has_next = kv0 + PF_KBLK < wg_kmax;
     if (has_next) stage_load(kv0 + PF_KBLK);  // overlaps this tile's compute
 
@@ -175,7 +176,7 @@ __global__ __launch_bounds__(512) void paged_attn_prefill_kernel(
 #pragma unroll
         for (int ks = 0; ks < 4; ks++) {
           bf16x8 bk = *reinterpret_cast<const bf16x8*>(
-              &k_lds[buf][n * 16 + frow][ks * 32 + fcol8]);
+              &k_lds[n * 16 + frow][ks * 32 + fcol8]);
           s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[ks], bk, s[n],
                                                          0, 0, 0);
         }
@@ -224,7 +225,7 @@ __global__ __launch_bounds__(512) void paged_attn_prefill_kernel(
       // + (l>>4)*128B + (l&15)*8B; each 32-key half needs TWO 8-read
       // batches — one per 4 d-tiles — reusing one 16-VGPR tr[] set)
       const unsigned vbase =
-          (unsigned)(unsigned long long)(&v_img[buf][0][0]) +
+          (unsigned)(unsigned long long)(&v_img[0][0]) +
           ((lane >> 4) * 128u + (lane & 15) * 8u);
       u32x2 tr[8];
 #define PF_TR8(OFF)                                                        \
@@ -274,10 +275,14 @@ __global__ __launch_bounds__(512) void paged_attn_prefill_kernel(
 #undef PF_TR8
 #undef PF_PV4
     }
-    // T14: the writes land AFTER compute so the loads' HBM latency hid
-    // under it — and into the OTHER buffer, so no pre-write barrier
-    if (has_next) stage_write(kv0 + PF_KBLK, buf ^ 1);
-    __syncthreads();   // all waves done with buf before it becomes t+2's
+    // single buffer: all waves must be done READING this tile before the
+    // next tile's write lands (the co-resident second workgroup hides
+    // both barriers)
+    __syncthreads();
+    if (has_next) {
+      stage_write(kv0 + PF_KBLK);
+      __syncthreads();
+    }
   }
 
   // ---- epilogue ------------------------------------------------------------
