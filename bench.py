@@ -661,8 +661,11 @@ def main():
         dist = dist_mod
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         # gloo for the benchmark's own barriers/reductions (host-side);
-        # the workers' data plane (RCCL / xGMI) is their own
-        dist.init_process_group(backend="gloo")
+        # the workers' data plane (RCCL / xGMI) is their own. Bounded
+        # timeout: a failing rank must not hang the others for 30 min.
+        import datetime
+        dist.init_process_group(backend="gloo",
+                                timeout=datetime.timedelta(seconds=900))
         if use_gpu:
             torch.cuda.set_device(local_rank)
 
