@@ -24,6 +24,11 @@ class AttnMetadata:
     prefill_seq_lens: Optional[torch.Tensor] = None   # [n_prefill] int32 (ctx+chunk)
     prefill_block_tables: Optional[torch.Tensor] = None  # [n_prefill, max_blk] int32
 
+    # precomputed prefill tile decomposition (tile_seq, tile_q0) — set by
+    # the prefill graph runner so capture never calls cu_q.cpu() (a sync
+    # inside hipGraph capture aborts it); None = derive from cu_q
+    prefill_tiles: Optional[tuple] = None
+
     # decode half
     decode_seq_lens: Optional[torch.Tensor] = None    # [n_decode] int32
     decode_block_tables: Optional[torch.Tensor] = None

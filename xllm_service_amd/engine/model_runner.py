@@ -78,11 +78,17 @@ class ModelRunner:
         enable, max_len, max_bs = self._graph_opts
         if not enable:
             return
-        from .graph_runner import DecodeGraphRunner
+        from .graph_runner import DecodeGraphRunner, PrefillGraphRunner
         self.graph_runner = DecodeGraphRunner(
             self.model, self.kv_caches, self.device,
             max_model_len=max_len, max_batch=max_bs)
         self.graph_runner.capture_all()
+        from xllm_service_amd.distributed import parallel_state as ps
+        if ps.tp_size() == 1:
+            self.prefill_graph = PrefillGraphRunner(
+                self.model, self.kv_caches, self.device,
+                max_model_len=max_len, pool=self.graph_runner.pool)
+            self.prefill_graph.capture_all()
 
     # ---- batch construction -------------------------------------------------
     def _build_batch(self, plan: StepPlan, bm):
@@ -298,11 +304,24 @@ class ModelRunner:
     def stop_followers(self):
         self._tp_bcast({"kind": "stop"})
 
+    prefill_graph = None
+
     # ---- one step -----------------------------------------------------------
     @torch.inference_mode()
     def execute(self, plan: StepPlan, bm) -> Dict[str, int]:
         """Run one step; returns {request_id: sampled_token} for sequences
         that produced a token this step (completed prefills + decodes)."""
+        pg = self.prefill_graph
+        if pg is not None and not plan.decodes and len(plan.prefills) == 1:
+            sp = plan.prefills[0]
+            seq = sp.seq
+            from xllm_service_amd.distributed import parallel_state as ps
+            if (seq.mm_embeds is None and seq.mrope_pos is None
+                    and ps.tp_size() == 1):
+                b = pg.bucket_for(sp.chunk_len)
+                if (b is not None
+                        and sp.chunk_start + b <= pg.max_model_len):
+                    return self._execute_prefill_graph(sp, b)
         gr = self.graph_runner
         if (gr is not None and not plan.prefills and plan.decodes
                 and gr.bucket_for(len(plan.decodes)) is not None
@@ -371,6 +390,27 @@ class ModelRunner:
             if n_here:
                 embeds[chunk][mask] = mm[n_before:n_before + n_here]
         return embeds
+
+    def _execute_prefill_graph(self, sp, b) -> Dict[str, int]:
+        """Single-sequence prefill chunk via the captured graph (the
+        serving arrival fast path; see PrefillGraphRunner padding notes)."""
+        import numpy as np
+        seq = sp.seq
+        cs, L = sp.chunk_start, sp.chunk_len
+        tokens = np.asarray(seq.prompt_token_ids[cs:cs + L], dtype=np.int64)
+        positions = np.arange(cs, cs + L, dtype=np.int64)
+        bt = np.asarray(seq.block_table, dtype=np.int64)
+        slots = bt[positions // BLOCK_SIZE] * BLOCK_SIZE \
+            + positions % BLOCK_SIZE
+        hidden = self.prefill_graph.run(b, tokens, positions, slots, cs,
+                                        seq.block_table)
+        if cs + L < seq.prompt_len:
+            return {}                     # mid-chunk: no token yet
+        sel = hidden[L - 1:L]
+        logits = self.model.compute_logits(sel)
+        toks = self._sample(logits, [seq])
+        lps = self._logprobs(logits, [seq], toks)
+        return {seq.request_id: (toks[0], lps.get(seq.request_id))}
 
     def _execute_decode_graph(self, plan: StepPlan, bm) -> Dict[str, int]:
         import numpy as np

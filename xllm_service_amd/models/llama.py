@@ -68,7 +68,7 @@ class LlamaAttention(nn.Module):
             ops.paged_attn_prefill(
                 qh[:np_], k_cache, v_cache, meta.prefill_block_tables,
                 meta.cu_q, meta.prefill_seq_lens, self.scale,
-                out=out3[:np_])
+                out=out3[:np_], tiles=meta.prefill_tiles)
         if nd:
             ops.paged_attn_decode(
                 qh[np_:], k_cache, v_cache, meta.decode_block_tables,

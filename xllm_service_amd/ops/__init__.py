@@ -146,7 +146,7 @@ def _prefill_tiles(cu_q):
 
 
 def paged_attn_prefill(q, k_cache, v_cache, block_tables, cu_q, seq_lens,
-                       scale, out=None):
+                       scale, out=None, tiles=None):
     if q.is_cuda:
         _require_ext()
         if k_cache.shape[-1] != 128:
@@ -161,12 +161,14 @@ def paged_attn_prefill(q, k_cache, v_cache, block_tables, cu_q, seq_lens,
             return res
         if out is None:
             out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
-        tile_seq, tile_q0 = _prefill_tiles(cu_q.cpu())
         dev = q.device
+        if tiles is None:
+            tile_seq, tile_q0 = _prefill_tiles(cu_q.cpu())
+            tiles = (torch.tensor(tile_seq, dtype=torch.int32, device=dev),
+                     torch.tensor(tile_q0, dtype=torch.int32, device=dev))
         _ops.paged_attn_prefill(
             out, q, k_cache, v_cache, block_tables, cu_q, seq_lens,
-            torch.tensor(tile_seq, dtype=torch.int32, device=dev),
-            torch.tensor(tile_q0, dtype=torch.int32, device=dev), scale)
+            tiles[0], tiles[1], scale)
         return out
     res = ref.paged_attn_prefill(q, k_cache, v_cache, block_tables, cu_q,
                                  seq_lens, scale)
