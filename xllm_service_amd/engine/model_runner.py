@@ -84,7 +84,16 @@ class ModelRunner:
             max_model_len=max_len, max_batch=max_bs)
         self.graph_runner.capture_all()
         from xllm_service_amd.distributed import parallel_state as ps
-        if ps.tp_size() == 1:
+        import os
+        # OPT-IN (XLLM_PREFILL_GRAPHS=1): replaying padded single-seq
+        # prefill chunks aborts with a device fault on the debug-model
+        # chunked-prefill shapes (L < bucket; tests/test_gpu_engine.py
+        # test_gpu_prefix_cache_and_chunked_prefill) while the full-length
+        # serving shape (L == 1024 == bucket) runs and saves ~2.5 ms per
+        # arrival. Until the padded-replay fault is isolated this stays
+        # off the default path — docs/TODO_ROUND3.md.
+        if (ps.tp_size() == 1
+                and os.environ.get("XLLM_PREFILL_GRAPHS") == "1"):
             self.prefill_graph = PrefillGraphRunner(
                 self.model, self.kv_caches, self.device,
                 max_model_len=max_len, pool=self.graph_runner.pool)
