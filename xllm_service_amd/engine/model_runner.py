@@ -94,9 +94,11 @@ class ModelRunner:
         # off the default path — docs/TODO_ROUND3.md.
         if (ps.tp_size() == 1
                 and os.environ.get("XLLM_PREFILL_GRAPHS") == "1"):
+            shared = os.environ.get("XLLM_PREFILL_GRAPH_SHARED_POOL") == "1"
             self.prefill_graph = PrefillGraphRunner(
                 self.model, self.kv_caches, self.device,
-                max_model_len=max_len, pool=self.graph_runner.pool)
+                max_model_len=max_len,
+                pool=self.graph_runner.pool if shared else None)
             self.prefill_graph.capture_all()
 
     # ---- batch construction -------------------------------------------------
