@@ -368,3 +368,36 @@ def test_abort_during_migration_pull_frees_blocks():
     eng.step()                            # event fired: blocks released
     assert eng.block_manager.num_free == free0
     assert "m2" not in eng.seqs
+
+
+@pytest.mark.anyio
+async def test_xgmi_migration_falls_back_to_bytes():
+    """If the xGMI/IPC migration path fails (unproven topologies), the
+    prefill worker must retry with the serialized-bytes transport instead
+    of erroring the request."""
+    from tests.test_service_integration import (http_client, make_master,
+                                                wait_for, worker_kwargs)
+    from xllm_service_amd.engine.worker import Worker
+
+    master = make_master(policy="RR")
+    await master.start(serve_http=False)
+    p0 = Worker("p0", "PREFILL", **worker_kwargs(master))
+    d0 = Worker("d0", "DECODE", **worker_kwargs(master))
+    try:
+        await p0.start()
+        await d0.start()
+        await wait_for(lambda: master.scheduler.has_available_instances())
+        # force the prefill side to attempt the xgmi transport on CPU —
+        # the decode side's IPC open fails, the fallback must kick in
+        p0._pick_transport = lambda peer: "xgmi"
+        client = await http_client(master)
+        r = await client.post("/v1/completions", json={
+            "model": "llama-tiny", "prompt": list(range(50, 90)),
+            "max_tokens": 6, "temperature": 0.0, "ignore_eos": True})
+        assert r.status_code == 200, r.text
+        assert r.json()["usage"]["completion_tokens"] == 6
+        await client.aclose()
+    finally:
+        await p0.stop()
+        await d0.stop()
+        await master.stop()
