@@ -565,6 +565,8 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
             "p50_headers_ms": round(sorted(
                 r[4] for r in window)[len(window) // 2] * 1000, 1)
             if window else 0.0,
+            "p50_request_s": round(sorted(comp.durations)[
+                len(comp.durations) // 2], 2) if comp.durations else 0.0,
             "calibrated_capacity_tok_s": round(cap_tok_s, 1),
             "requests_timed": len(window),
             "config": {
