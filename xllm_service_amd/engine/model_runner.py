@@ -417,7 +417,10 @@ class ModelRunner:
                                         seq.block_table)
         if cs + L < seq.prompt_len:
             return {}                     # mid-chunk: no token yet
-        sel = hidden[L - 1:L]
+        # clone the row OUT of the graph's private pool before the eager
+        # lm_head GEMM reads it (hipBLASLt reads past a tightly-packed
+        # pool allocation were the suspected device fault)
+        sel = hidden[L - 1:L].clone()
         logits = self.model.compute_logits(sel)
         toks = self._sample(logits, [seq])
         lps = self._logprobs(logits, [seq], toks)
