@@ -515,7 +515,7 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
             torch.cuda.synchronize()
         await barrier()                  # B3
 
-        elapsed = t1 - t0
+        elapsed = max(t1 - t0, 1e-3)   # K completions can share a tick
         if world > 1:
             te = torch.tensor([elapsed], dtype=torch.float64)
             await asyncio.get_event_loop().run_in_executor(
@@ -583,6 +583,9 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
             },
         }))
         cmd_all("quit", )
+        for rt in readers:
+            rt.cancel()
+        await asyncio.gather(*readers, return_exceptions=True)
         # debugging artifacts: per-request completions + final master metrics
         try:
             with open("gpurun_out/bench_requests.csv", "w") as f:
