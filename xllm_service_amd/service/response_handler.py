@@ -20,10 +20,14 @@ def _sse(obj) -> str:
     return f"data: {json.dumps(obj, ensure_ascii=False)}\n\n"
 
 
-async def _next_delta(queue) -> GenerationDelta:
+async def _next_delta(queue, hold: list) -> GenerationDelta:
     """Await one delta, then greedily coalesce any backlog of plain token
     deltas into it (one SSE write then carries the merged text). Deltas
-    carrying errors, text overrides or logprobs are never merged across."""
+    carrying errors, text overrides or logprobs are never merged across —
+    they are parked in `hold` (a per-stream 1-slot list) and delivered on
+    the next call."""
+    if hold:
+        return hold.pop()
     gen: GenerationDelta = await asyncio.wait_for(queue.get(),
                                                   STREAM_TIMEOUT_S)
     if gen.error or gen.finished or gen.text is not None or gen.logprobs:
@@ -34,8 +38,7 @@ async def _next_delta(queue) -> GenerationDelta:
         except asyncio.QueueEmpty:
             break
         if nxt.error or nxt.text is not None or nxt.logprobs:
-            # cannot merge: re-deliver it right after this one
-            queue._queue.appendleft(nxt)  # asyncio.Queue: deque internally
+            hold.append(nxt)    # deliver right after the merged delta
             break
         gen.token_ids = list(gen.token_ids) + list(nxt.token_ids)
         if nxt.finished:
@@ -85,9 +88,10 @@ class ResponseHandler:
                          for tt, vv in (lp.get("top") or {}).items()]}
                     for t, lp in zip(gen.token_ids, gen.logprobs)]
 
+        held: list = []
         try:
             while True:
-                gen = await _next_delta(req.output_queue)
+                gen = await _next_delta(req.output_queue, held)
                 if gen.error:
                     yield _sse({"error": {"message": gen.error,
                                           "type": "server_error"}})
@@ -200,9 +204,10 @@ class ResponseHandler:
                         "created": created, "model": req.model,
                         "choices": [{"index": 0, "text": echo_text,
                                      "finish_reason": None}]})
+        held: list = []
         try:
             while True:
-                gen = await _next_delta(req.output_queue)
+                gen = await _next_delta(req.output_queue, held)
                 if gen.error:
                     yield _sse({"error": {"message": gen.error,
                                           "type": "server_error"}})
