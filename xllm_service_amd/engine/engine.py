@@ -390,6 +390,9 @@ class LLMEngine:
         """Queue a migrated-in request for activation once its KV pull
         event fires (None = data already resident). Decode steps continue
         while the copy is in flight."""
+        for pm in self.pending_migrations:
+            if pm.request_id == request_id:
+                pm.aborted = True      # superseded (transport retry)
         self.pending_migrations.append(PendingMigration(
             request_id=request_id,
             prompt_token_ids=list(prompt_token_ids),
@@ -406,7 +409,7 @@ class LLMEngine:
             if pm.event is not None and not pm.event.query():
                 continue                       # copy still in flight
             self.pending_migrations.remove(pm)
-            if pm.aborted:
+            if pm.aborted or pm.request_id in self.seqs:
                 self.free_blocks(pm.block_ids)
                 continue
             fin = self.activate_migrated_request(
