@@ -277,3 +277,42 @@ def test_parser_families(model, family, rtag, ttags):
     if srp is not None:
         assert reason_acc == "thinking..."
     assert len(out_calls) == 1 and out_calls[0].name == "f"
+
+
+def test_sse_delta_coalescing():
+    """_next_delta merges backlogged plain deltas, stops at finishes, and
+    parks unmergeable deltas in the hold slot."""
+    import asyncio
+
+    from xllm_service_amd.service.request import GenerationDelta
+    from xllm_service_amd.service.response_handler import _next_delta
+
+    async def run():
+        q = asyncio.Queue()
+        hold = []
+        q.put_nowait(GenerationDelta(token_ids=[1]))
+        q.put_nowait(GenerationDelta(token_ids=[2]))
+        q.put_nowait(GenerationDelta(token_ids=[3], finished=True,
+                                     finish_reason="stop",
+                                     usage_prompt_tokens=5,
+                                     usage_completion_tokens=3))
+        g = await _next_delta(q, hold)
+        assert g.token_ids == [1, 2, 3] and g.finished
+        assert g.usage_completion_tokens == 3
+
+        # unmergeable (text override) delta is held and delivered next
+        q.put_nowait(GenerationDelta(token_ids=[4]))
+        q.put_nowait(GenerationDelta(token_ids=[], text="tail",
+                                     finished=True, finish_reason="stop"))
+        g1 = await _next_delta(q, hold)
+        assert g1.token_ids == [4] and not g1.finished
+        g2 = await _next_delta(q, hold)
+        assert g2.text == "tail" and g2.finished
+
+        # logprob-carrying deltas are never merged
+        q.put_nowait(GenerationDelta(token_ids=[5],
+                                     logprobs=[{"token_logprob": -0.5}]))
+        q.put_nowait(GenerationDelta(token_ids=[6]))
+        g3 = await _next_delta(q, hold)
+        assert g3.token_ids == [5]
+    asyncio.run(run())
