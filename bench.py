@@ -326,6 +326,25 @@ def loadgen_child_main():
     asyncio.run(run())
 
 
+def summarize_window(records, t0, t1, slo_s, rate, output_len):
+    """Pure window math for the serving record (unit-tested): SLO-gated
+    goodput capped at the offered load, raw window rate, TTFT percentiles.
+    records: (t_end, ttft_s, out_tokens, tag, hdr_s)."""
+    window = [r for r in records if t0 <= r[0] <= t1 and r[3] == "poisson"]
+    elapsed = max(t1 - t0, 1e-3)
+    offered = rate * output_len
+    good = sum(r[2] for r in window if r[1] <= slo_s) / elapsed
+    alltok = sum(r[2] for r in window) / elapsed
+    ttfts = sorted(r[1] for r in window)
+    p50 = ttfts[len(ttfts) // 2] * 1000 if ttfts else 0.0
+    p99 = ttfts[int(len(ttfts) * 0.99)] * 1000 if ttfts else 0.0
+    hdrs = sorted(r[4] for r in window)
+    p50h = hdrs[len(hdrs) // 2] * 1000 if hdrs else 0.0
+    return dict(value=min(good, offered), total=min(alltok, offered),
+                window_raw=alltok, offered=offered, p50_ttft_ms=p50,
+                p99_ttft_ms=p99, p50_headers_ms=p50h, n=len(window))
+
+
 def _parse_prom(text: str) -> dict:
     vals = {}
     for line in text.splitlines():
@@ -522,20 +541,9 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
                 None, dist.all_reduce, te, dist.ReduceOp.MAX)
             elapsed = float(te.item())
 
-        window = [r for r in comp.records
-                  if t0 <= r[0] <= t1 and r[3] == "poisson"]
-        slo_s = args.slo_ttft_ms / 1000.0
-        good_tokens = sum(r[2] for r in window if r[1] <= slo_s)
-        all_tokens = sum(r[2] for r in window)
-        # steady-state goodput cannot exceed the offered load; a short
-        # window catching a Poisson completion cluster would otherwise
-        # over-report (raw window rate stays in window_tok_per_s)
-        offered = rate * args.output_len
-        raw_good = good_tokens / elapsed
-        raw_all = all_tokens / elapsed
-        ttfts = sorted(r[1] for r in window)
-        p50 = ttfts[len(ttfts) // 2] * 1000 if ttfts else 0.0
-        p99 = ttfts[int(len(ttfts) * 0.99)] * 1000 if ttfts else 0.0
+        sm = summarize_window(comp.records, t0, t0 + elapsed,
+                              args.slo_ttft_ms / 1000.0, rate,
+                              args.output_len)
 
         n_p = sum(1 for r in roles if r == "PREFILL")
         n_d = len(roles) - n_p
@@ -544,7 +552,7 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
         print(json.dumps({
             "metric": "SLO-goodput (out tok/s under p50 TTFT SLO), "
                       "Llama-3-8B PD-disagg on 8 MI355X",
-            "value": round(min(raw_good, offered), 2),
+            "value": round(sm["value"], 2),
             "unit": "tok/s",
             "n_gpus": n_total,
             "steps": args.steps,
@@ -555,20 +563,18 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
             "vs_baseline": None,
             "dtype": "bf16" if use_gpu else "fp32",
             "data": "synthetic",
-            "total_tok_per_s": round(min(raw_all, offered), 2),
-            "window_tok_per_s": round(raw_all, 2),
-            "offered_tok_per_s": round(offered, 2),
-            "p50_ttft_ms": round(p50, 1),
-            "p99_ttft_ms": round(p99, 1),
+            "total_tok_per_s": round(sm["total"], 2),
+            "window_tok_per_s": round(sm["window_raw"], 2),
+            "offered_tok_per_s": round(sm["offered"], 2),
+            "p50_ttft_ms": round(sm["p50_ttft_ms"], 1),
+            "p99_ttft_ms": round(sm["p99_ttft_ms"], 1),
             "slo_ttft_ms": args.slo_ttft_ms,
             "arrival_rate_req_s": round(rate, 2),
-            "p50_headers_ms": round(sorted(
-                r[4] for r in window)[len(window) // 2] * 1000, 1)
-            if window else 0.0,
+            "p50_headers_ms": round(sm["p50_headers_ms"], 1),
             "p50_request_s": round(sorted(comp.durations)[
                 len(comp.durations) // 2], 2) if comp.durations else 0.0,
             "calibrated_capacity_tok_s": round(cap_tok_s, 1),
-            "requests_timed": len(window),
+            "requests_timed": sm["n"],
             "config": {
                 "model": model_name,
                 "global_batch": conc,

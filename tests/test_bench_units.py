@@ -43,3 +43,26 @@ def test_completions_exact_k():
         c.arm(0)                                   # k=0 fires immediately
         assert c.event.is_set()
     asyncio.run(run())
+
+
+def test_summarize_window_gating_and_cap():
+    recs = [
+        (10.0, 0.1, 100, "poisson", 0.002),   # good
+        (11.0, 2.0, 100, "poisson", 0.002),   # misses the 1s TTFT SLO
+        (12.0, 0.2, 100, "poisson", 0.004),   # good
+        (12.5, 0.2, 100, "ramp", 0.002),      # ramp never counted
+        (99.0, 0.1, 100, "poisson", 0.002),   # outside window
+    ]
+    sm = bench.summarize_window(recs, t0=9.0, t1=13.0, slo_s=1.0,
+                                rate=1.0, output_len=100)
+    assert sm["n"] == 3
+    assert sm["window_raw"] == 300 / 4.0
+    assert sm["value"] == 200 / 4.0          # SLO-gated
+    assert sm["offered"] == 100.0
+    # over-offered window is capped
+    burst = [(10.0 + i * 0.01, 0.1, 100, "poisson", 0.001)
+             for i in range(50)]
+    sm2 = bench.summarize_window(burst, 10.0, 10.5, 1.0, rate=2.0,
+                                 output_len=100)
+    assert sm2["value"] == 200.0             # capped at offered
+    assert sm2["window_raw"] > 200.0
