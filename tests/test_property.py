@@ -24,7 +24,12 @@ def test_windowed_decoder_matches_full_decode(text, chunk_sizes):
         out += dec.push(ids[i:i + c])
         i += c
     out += dec.push(ids[i:])
-    assert out == tk.decode(ids) == text
+    full = tk.decode(ids)
+    # contract: a trailing U+FFFD may be withheld at stream end (it looks
+    # like a partial UTF-8 sequence); everything before it must match
+    assert full == text
+    assert out == full or (full.startswith(out)
+                           and set(full[len(out):]) == {"\ufffd"})
 
 
 # ------------------------------------------------------------ stop scanner
@@ -108,3 +113,43 @@ def test_pack_gemm_weight_is_a_permutation(shape):
     back = (p.view(n // 16, k // 32, 4, 16, 8)
              .permute(0, 3, 1, 2, 4).reshape(n, k))
     assert torch.equal(back, w)
+
+
+@settings(max_examples=8, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(st.lists(st.tuples(st.sampled_from(["add", "addoff", "abort", "step"]),
+                          st.integers(min_value=0, max_value=9)),
+                min_size=8, max_size=26),
+       st.integers(min_value=0, max_value=500))
+def test_engine_fuzz_with_swap_and_priorities(ops_list, seed):
+    """Same invariant with the DRAM swap tier enabled and mixed
+    online/offline priorities: preemption may swap or recompute; all
+    HBM and DRAM blocks return to their pools after draining."""
+    from xllm_service_amd.engine.engine import LLMEngine
+    from xllm_service_amd.engine.sampling import SamplingParams
+    eng = LLMEngine("llama-tiny", device="cpu", max_kv_blocks=20, seed=3,
+                    swap_space_mb=1, max_num_seqs=6)
+    free0 = eng.block_manager.num_free
+    cfree0 = eng.cpu_block_manager.num_free if eng.cpu_block_manager else 0
+    torch.manual_seed(seed)
+    nid = 0
+    live = []
+    for op, arg in ops_list:
+        if op in ("add", "addoff"):
+            n = 4 + (arg * 7) % 36
+            prompt = torch.randint(0, eng.cfg.vocab_size, (n,)).tolist()
+            eng.add_request(f"s{nid}", prompt,
+                            SamplingParams(max_tokens=1 + arg % 6,
+                                           ignore_eos=True),
+                            priority=1 if op == "addoff" else 0)
+            live.append(f"s{nid}")
+            nid += 1
+        elif op == "abort" and live:
+            eng.abort_request(live.pop(arg % len(live)))
+        elif op == "step":
+            eng.step()
+    while eng.has_work():
+        eng.step()
+    assert eng.block_manager.num_free == free0
+    if eng.cpu_block_manager:
+        assert eng.cpu_block_manager.num_free == cfree0

@@ -161,7 +161,12 @@ class IncrementalDecoder:
     Decodes only a sliding tail window (prefix/read offsets) so per-push
     cost is O(new tokens), not O(all tokens so far) — at 1024-token
     streams the full-redecode scheme is O(n^2) per request and dominates
-    the master's token fan-out path."""
+    the master's token fan-out path.
+
+    Contract: a trailing U+FFFD is withheld until more text arrives (it is
+    indistinguishable from a partial UTF-8 sequence mid-stream); if the
+    stream ends on one, it is dropped. A replacement char FOLLOWED by more
+    text is emitted normally on the next push."""
 
     def __init__(self, tokenizer: Tokenizer):
         self.tk = tokenizer
