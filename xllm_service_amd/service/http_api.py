@@ -237,10 +237,23 @@ def build_app(master) -> FastAPI:
                            "type": "server_error"}}, status_code=503)
         return None
 
+    def _bad_params(body) -> Optional[JSONResponse]:
+        mt = getattr(body, "max_tokens", None)
+        if mt is None:
+            mt = getattr(body, "max_completion_tokens", None)
+        if mt is not None and mt < 1:
+            # the old params plumbing turned 0 into the default (128):
+            # reject like OpenAI instead of silently generating
+            return JSONResponse(
+                {"error": {"message": "max_tokens must be at least 1",
+                           "type": "invalid_request_error"}},
+                status_code=400)
+        return None
+
     @app.post("/v1/completions")
     async def completions(body: CompletionBody, request: Request):
         metrics.REQUEST_IN_TOTAL.labels(method="completion").inc()
-        na = _not_ready()
+        na = _not_ready() or _bad_params(body)
         if na is not None:
             return na
         if isinstance(body.prompt, list) and body.prompt \
@@ -263,7 +276,7 @@ def build_app(master) -> FastAPI:
     @app.post("/v1/chat/completions")
     async def chat_completions(body: ChatBody, request: Request):
         metrics.REQUEST_IN_TOTAL.labels(method="chat").inc()
-        na = _not_ready()
+        na = _not_ready() or _bad_params(body)
         if na is not None:
             return na
         try:
