@@ -157,7 +157,9 @@ def build_app(master) -> FastAPI:
                         d = await _aio.wait_for(req.output_queue.get(), 600.0)
                         if d.error:
                             break
-                        text = dec.push(d.token_ids) if d.token_ids else ""
+                        text = (d.text if d.text is not None
+                                else dec.push(d.token_ids)
+                                if d.token_ids else "")
                         if text:
                             yield ("event: content_block_delta\ndata: " +
                                    _json.dumps({
