@@ -350,3 +350,32 @@ def test_max_tokens_zero_rejected():
             await w.stop()
             await master.stop()
     asyncio.run(run())
+
+
+def test_unsupported_shapes_rejected():
+    import asyncio
+
+    from tests.test_service_integration import (http_client, make_master,
+                                                wait_for, worker_kwargs)
+    from xllm_service_amd.engine.worker import Worker
+
+    async def run():
+        master = make_master(policy="RR")
+        await master.start(serve_http=False)
+        w = Worker("w0", "DEFAULT", **worker_kwargs(master))
+        try:
+            await w.start()
+            await wait_for(
+                lambda: master.scheduler.has_available_instances())
+            client = await http_client(master)
+            r = await client.post("/v1/completions", json={
+                "model": "llama-tiny", "prompt": [1, 2, 3], "n": 2})
+            assert r.status_code == 400
+            r = await client.post("/v1/completions", json={
+                "model": "llama-tiny", "prompt": ["two", "prompts"]})
+            assert r.status_code == 400
+            await client.aclose()
+        finally:
+            await w.stop()
+            await master.stop()
+    asyncio.run(run())

@@ -250,6 +250,12 @@ def build_app(master) -> FastAPI:
                 {"error": {"message": "max_tokens must be at least 1",
                            "type": "invalid_request_error"}},
                 status_code=400)
+        if getattr(body, "n", 1) != 1:
+            # better a clear 400 than silently returning one choice
+            return JSONResponse(
+                {"error": {"message": "n != 1 is not supported",
+                           "type": "invalid_request_error"}},
+                status_code=400)
         return None
 
     @app.post("/v1/completions")
@@ -262,6 +268,16 @@ def build_app(master) -> FastAPI:
                 and isinstance(body.prompt[0], int):
             token_ids = list(body.prompt)
             prompt_text = ""
+        elif isinstance(body.prompt, list) and len(body.prompt) > 1 \
+                and isinstance(body.prompt[0], str):
+            # OpenAI batches string lists into multiple choices; a silent
+            # concatenation would serve the wrong completion
+            return JSONResponse(
+                {"error": {"message": "batched prompt lists are not "
+                                       "supported; send one request per "
+                                       "prompt",
+                           "type": "invalid_request_error"}},
+                status_code=400)
         else:
             prompt_text = body.prompt if isinstance(body.prompt, str) \
                 else "".join(body.prompt or [])
