@@ -379,3 +379,52 @@ def test_unsupported_shapes_rejected():
             await w.stop()
             await master.stop()
     asyncio.run(run())
+
+
+def test_engine_failure_surfaces_error_to_client():
+    """A poisoned engine step aborts running sequences WITH an error
+    message — clients must see a 500, not an empty 200 (round-2 fix)."""
+    import asyncio
+
+    from tests.test_service_integration import (http_client, make_master,
+                                                wait_for, worker_kwargs)
+    from xllm_service_amd.engine.worker import Worker
+
+    async def run():
+        master = make_master(policy="RR")
+        await master.start(serve_http=False)
+        w = Worker("w0", "DEFAULT", **worker_kwargs(master))
+        try:
+            await w.start()
+            await wait_for(
+                lambda: master.scheduler.has_available_instances())
+
+            # poison the NEXT engine step only
+            real_step = w.engine.step
+            state = {"armed": False, "fired": False}
+
+            def boom():
+                if state["armed"] and not state["fired"]:
+                    state["fired"] = True
+                    raise RuntimeError("injected device fault")
+                return real_step()
+            w.engine.step = boom
+
+            client = await http_client(master)
+            state["armed"] = True
+            r = await client.post("/v1/completions", json={
+                "model": "llama-tiny", "prompt": list(range(20, 40)),
+                "max_tokens": 4, "temperature": 0.0, "ignore_eos": True})
+            assert r.status_code == 500, r.text
+            assert "injected device fault" in r.json()["error"]["message"]
+
+            # worker keeps serving afterwards
+            r2 = await client.post("/v1/completions", json={
+                "model": "llama-tiny", "prompt": list(range(20, 40)),
+                "max_tokens": 4, "temperature": 0.0, "ignore_eos": True})
+            assert r2.status_code == 200, r2.text
+            await client.aclose()
+        finally:
+            await w.stop()
+            await master.stop()
+    asyncio.run(run())

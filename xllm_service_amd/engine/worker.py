@@ -297,6 +297,8 @@ class Worker:
                     # running sequences with an abort so the master errors
                     # them out instead of hanging the clients
                     log.exception("engine step failed; aborting running seqs")
+                    import traceback
+                    err = traceback.format_exc().strip().splitlines()[-1]
                     from .engine import StepOutput
                     bad = [s.request_id
                            for s in list(self.engine.scheduler.running)
@@ -307,7 +309,8 @@ class Worker:
                         self.engine.abort_request(rid)
                         outs.append(StepOutput(
                             request_id=rid, new_token_ids=[], finished=True,
-                            finish_reason="abort"))
+                            finish_reason="abort",
+                            error=f"engine step failed: {err}"))
                 else:
                     ms = (time.monotonic() - t_step) * 1000.0
                     st = self.engine.stats
