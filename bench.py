@@ -521,14 +521,28 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
         if settle > 0:
             await asyncio.sleep(settle)   # ramp residuals gone, B steady
         comp.arm(args.warmup)
-        await comp.event.wait()          # W completions at steady state
+
+        async def wait_or_die(ev):
+            while not ev.is_set():
+                if any(ch.poll() is not None for ch in children):
+                    raise RuntimeError("a load-generator child died")
+                if mproc.poll() is not None:
+                    raise RuntimeError("master process died")
+                if proc.poll() is not None:
+                    raise RuntimeError("worker process died")
+                try:
+                    await asyncio.wait_for(asyncio.shield(ev.wait()), 5.0)
+                except asyncio.TimeoutError:
+                    pass
+
+        await wait_or_die(comp.event)    # W completions at steady state
 
         await barrier()                  # B2
         if use_gpu:
             torch.cuda.synchronize()
         comp.arm(args.steps)
         t0 = time.monotonic()
-        await comp.event.wait()
+        await wait_or_die(comp.event)
         t1 = comp.t1
         if use_gpu:
             torch.cuda.synchronize()
