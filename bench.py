@@ -531,7 +531,7 @@ async def run_serving_rank0(args, world, dist, use_gpu, model_name,
                 if proc.poll() is not None:
                     raise RuntimeError("worker process died")
                 try:
-                    await asyncio.wait_for(asyncio.shield(ev.wait()), 5.0)
+                    await asyncio.wait_for(ev.wait(), 5.0)
                 except asyncio.TimeoutError:
                     pass
 
