@@ -1,6 +1,12 @@
 #!/usr/bin/env python3
 """Full-stack serving benchmark: HTTP -> master -> RPC -> workers.
 
+NOTE (round 2): the driver-contract bench (bench.py, default serving mode)
+supersedes this script for headline measurements — it adds multi-process
+load generation, ramp-calibrated open-loop arrivals, steady-state window
+gating and the offered-load cap. This script remains useful for ad-hoc
+topologies on one box (arbitrary TYPE:device worker specs).
+
 Measures SLO-goodput (output tok/s from requests whose TTFT meets the SLO)
 through the complete control plane, on any topology:
 
