@@ -96,3 +96,37 @@ async def test_flip_keeps_last_prefill():
         for w in workers:
             await w.stop()
         await master.stop()
+
+
+@pytest.mark.anyio
+async def test_flip_back_decode_to_prefill():
+    """The reverse direction: a drained decode flips to the prefill side
+    and the worker adopts PREFILL (no cache reset needed that way)."""
+    master = make_master(policy="SLO_AWARE")
+    await master.start(serve_http=False)
+    workers = [Worker("p0", "PREFILL", **worker_kwargs(master)),
+               Worker("d0", "DECODE", **worker_kwargs(master)),
+               Worker("d1", "DECODE", **worker_kwargs(master))]
+    try:
+        for w in workers:
+            await w.start()
+        await wait_for(lambda: len(master.instance_mgr.instances) == 3)
+        mgr = master.instance_mgr
+        assert mgr.flip_instance_role("d1", "prefill") is True
+        assert "d1" in mgr.prefill_index and "d1" not in mgr.decode_index
+        await wait_for(lambda: workers[2].itype == InstanceType.PREFILL)
+        await wait_for(
+            lambda: mgr.get("d1").meta.itype == InstanceType.PREFILL.value)
+        # prefill side now has two instances; routing still works
+        from tests.test_service_integration import http_client
+        client = await http_client(master)
+        for _ in range(3):
+            r = await client.post("/v1/completions", json={
+                "model": "llama-tiny", "prompt": list(range(30, 60)),
+                "max_tokens": 3, "temperature": 0.0, "ignore_eos": True})
+            assert r.status_code == 200, r.text
+        await client.aclose()
+    finally:
+        for w in workers:
+            await w.stop()
+        await master.stop()
